@@ -1,0 +1,72 @@
+"""Simulator correctness on hand-solvable traces + event-log format."""
+import json
+
+from gpuschedule_amd import EventLog, JobSpec, make_policy, make_trace, simulate
+
+
+def J(jid, t=0.0, n=1, iters=10, hint=1.0, model="resnet18"):
+    return JobSpec(job_id=jid, submit_time=t, n_gpus=n, total_iters=iters,
+                   model=model, iter_time_hint=hint)
+
+
+def test_single_job_jct():
+    # 10 iters x 1s + 0.5s cold start ~= 10.5s JCT
+    log, s = simulate([J("a", iters=10)], make_policy("fifo"), n_gpus=1)
+    assert abs(s["avg_jct"] - 10.5) < 0.3
+    assert abs(s["makespan"] - s["avg_jct"]) < 1e-6
+
+
+def test_two_sequential_jobs_makespan():
+    specs = [J("a", t=0.0, iters=10), J("b", t=0.0, iters=10)]
+    log, s = simulate(specs, make_policy("fifo"), n_gpus=1)
+    # a: ~10.5, b waits then runs ~10.5 => makespan ~21, avg ~15.75
+    assert 20.0 < s["makespan"] < 22.5
+    assert 14.5 < s["avg_jct"] < 17.0
+
+
+def test_parallel_jobs_independent():
+    specs = [J("a", iters=10), J("b", iters=10)]
+    log, s = simulate(specs, make_policy("fifo"), n_gpus=2)
+    assert s["makespan"] < 12.0
+
+
+def test_gang_allreduce_overhead():
+    one, _ = simulate([J("a", n=1, iters=50, hint=0.0)],
+                      make_policy("fifo"), n_gpus=8)
+    four, _ = simulate([J("a", n=4, iters=50, hint=0.0)],
+                       make_policy("fifo"), n_gpus=8)
+    t1 = one.summary()["makespan"]
+    t4 = four.summary()["makespan"]
+    assert t4 > t1  # gang pays all-reduce overhead (weak scaling)
+    assert t4 < 1.3 * t1
+
+
+def test_event_log_jsonl_roundtrip(tmp_path):
+    specs = make_trace(8, seed=2, mean_interarrival=2.0,
+                       iters_range=(5, 20))
+    log, s = simulate(specs, make_policy("srtf"), n_gpus=2)
+    p = tmp_path / "events.jsonl"
+    log.save(str(p))
+    with open(p) as f:
+        for line in f:
+            rec = json.loads(line)
+            assert "t" in rec and "event" in rec
+    loaded = EventLog.load(str(p))
+    assert loaded.summary() == s
+
+
+def test_all_jobs_complete_all_policies():
+    specs = make_trace(24, mix="philly-slice", seed=11,
+                       mean_interarrival=1.0, iters_range=(5, 60))
+    for pol in ["fifo", "srtf", "las", "gandiva", "elastic"]:
+        log, s = simulate(specs, make_policy(pol), n_gpus=4)
+        assert int(s["n_complete"]) == 24, f"{pol} lost jobs"
+
+
+def test_contention_separates_policies():
+    # Heavy contention: SRTF should beat FIFO on avg JCT.
+    specs = [J("big", t=0.0, iters=200)] + \
+        [J(f"s{i}", t=1.0 + i, iters=5) for i in range(8)]
+    _, fifo = simulate(specs, make_policy("fifo"), n_gpus=1)
+    _, srtf = simulate(specs, make_policy("srtf"), n_gpus=1)
+    assert srtf["avg_jct"] < 0.5 * fifo["avg_jct"]
