@@ -73,6 +73,18 @@ class Scheduler:
         self.log.emit(job.submit_t, "submit", spec.job_id,
                       n_gpus=spec.n_gpus, model=spec.model,
                       iters=spec.total_iters)
+        # Reject unsatisfiable jobs at admission (they would otherwise
+        # starve in the queue forever).
+        width = spec.min_gpus if spec.elastic else spec.n_gpus
+        mem = catalog.job_mem_bytes(spec)
+        too_wide = width > min(self.cluster.n_gpus,
+                               self.cluster.gpus_per_node)
+        too_big = mem > max(g.hbm_bytes for g in self.cluster.gpus)
+        if too_wide or too_big:
+            reason = ("gang wider than cluster/node" if too_wide
+                      else "per-GPU memory exceeds HBM")
+            self.on_fail(spec.job_id, f"unsatisfiable: {reason}",
+                         job.submit_t)
 
     def next_arrival(self) -> Optional[float]:
         return self._pending[0].submit_time if self._pending else None

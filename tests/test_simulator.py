@@ -44,7 +44,8 @@ def test_gang_allreduce_overhead():
 def test_event_log_jsonl_roundtrip(tmp_path):
     specs = make_trace(8, seed=2, mean_interarrival=2.0,
                        iters_range=(5, 20))
-    log, s = simulate(specs, make_policy("srtf"), n_gpus=2)
+    log, s = simulate(specs, make_policy("srtf"), n_gpus=4)
+    assert int(s["n_complete"]) == 8
     p = tmp_path / "events.jsonl"
     log.save(str(p))
     with open(p) as f:
@@ -70,3 +71,11 @@ def test_contention_separates_policies():
     _, fifo = simulate(specs, make_policy("fifo"), n_gpus=1)
     _, srtf = simulate(specs, make_policy("srtf"), n_gpus=1)
     assert srtf["avg_jct"] < 0.5 * fifo["avg_jct"]
+
+
+def test_unsatisfiable_job_fails_at_admission():
+    specs = [J("wide", n=4, iters=5), J("ok", n=1, iters=5)]
+    log, s = simulate(specs, make_policy("fifo"), n_gpus=2)
+    fails = [e for e in log.events if e["event"] == "fail"]
+    assert len(fails) == 1 and fails[0]["job"] == "wide"
+    assert int(s["n_complete"]) == 1
