@@ -1,0 +1,83 @@
+"""BERT-base encoder (MLM objective) on the gfx950 op set.
+
+Hot path per layer: fused QKV MFMALinear, flash attention, output
+projection, FusedLayerNorm (post-LN), GELU FFN on MFMALinear.  Vocab
+is padded to a multiple of 64 for the MFMA LM head.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.attention import flash_attention
+from ..ops.linear import MFMALinear
+from ..ops.norms import FusedLayerNorm
+
+
+@dataclass
+class BertConfig:
+    vocab_size: int = 30528        # 30522 padded to %64
+    hidden: int = 768
+    layers: int = 12
+    heads: int = 12
+    ffn: int = 3072
+    max_seq: int = 512
+
+
+class BertLayer(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.heads = cfg.heads
+        self.hd = cfg.hidden // cfg.heads
+        self.qkv = MFMALinear(cfg.hidden, 3 * cfg.hidden)
+        self.proj = MFMALinear(cfg.hidden, cfg.hidden)
+        self.ln1 = FusedLayerNorm(cfg.hidden)
+        self.fc1 = MFMALinear(cfg.hidden, cfg.ffn)
+        self.fc2 = MFMALinear(cfg.ffn, cfg.hidden)
+        self.ln2 = FusedLayerNorm(cfg.hidden)
+
+    def forward(self, x):
+        B, S, Hd = x.shape
+        qkv = self.qkv(x).view(B, S, 3, self.heads, self.hd)
+        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+        att = flash_attention(q, k, v, causal=False)
+        att = att.transpose(1, 2).reshape(B, S, Hd)
+        x = self.ln1(x + self.proj(att))
+        h = F.gelu(self.fc1(x), approximate="tanh")
+        x = self.ln2(x + self.fc2(h))
+        return x
+
+
+class BertMLM(nn.Module):
+    def __init__(self, cfg: BertConfig = BertConfig()):
+        super().__init__()
+        self.cfg = cfg
+        self.tok = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.pos = nn.Embedding(cfg.max_seq, cfg.hidden)
+        self.ln = FusedLayerNorm(cfg.hidden)
+        self.layers = nn.ModuleList(BertLayer(cfg)
+                                    for _ in range(cfg.layers))
+        self.head = MFMALinear(cfg.hidden, cfg.vocab_size)
+        self.apply(self._init)
+
+    @staticmethod
+    def _init(m):
+        if isinstance(m, (nn.Embedding,)):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, tokens):
+        B, S = tokens.shape
+        pos = torch.arange(S, device=tokens.device)
+        x = self.ln(self.tok(tokens) + self.pos(pos)[None])
+        for layer in self.layers:
+            x = layer(x)
+        return self.head(x)
+
+    def loss(self, tokens, labels):
+        logits = self(tokens)
+        return F.cross_entropy(logits.view(-1, self.cfg.vocab_size).float(),
+                               labels.view(-1))

@@ -1,0 +1,7 @@
+from ._ext import get_ext, have_ext
+from .norms import FusedLayerNorm, FusedRMSNorm
+from .linear import MFMALinear
+from .optim import FusedAdamW, FusedSGD
+
+__all__ = ["get_ext", "have_ext", "FusedLayerNorm", "FusedRMSNorm",
+           "MFMALinear", "FusedAdamW", "FusedSGD"]

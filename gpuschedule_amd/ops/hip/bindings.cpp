@@ -1,0 +1,27 @@
+// Python bindings for the gfx950 kernel library.
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> norm_fwd(torch::Tensor x, torch::Tensor gamma,
+                                    c10::optional<torch::Tensor> beta,
+                                    double eps, bool rms);
+std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                    c10::optional<torch::Tensor> mean,
+                                    torch::Tensor rstd, torch::Tensor gamma,
+                                    bool rms);
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> bias, bool relu);
+torch::Tensor transpose2d(torch::Tensor x);
+void adamw_step(torch::Tensor p_bf16, torch::Tensor p32, torch::Tensor grad,
+                torch::Tensor m, torch::Tensor v, double lr, double beta1,
+                double beta2, double eps, double wd, long step);
+void sgd_step(torch::Tensor p32, torch::Tensor p_bf16, torch::Tensor grad,
+              torch::Tensor buf, double lr, double momentum, double wd);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("norm_fwd", &norm_fwd, "fused LayerNorm/RMSNorm forward (bf16)");
+  m.def("norm_bwd", &norm_bwd, "fused LayerNorm/RMSNorm backward (bf16)");
+  m.def("gemm_nt", &gemm_nt, "MFMA bf16 GEMM C = A @ B^T (+bias, +relu)");
+  m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
+  m.def("adamw_step", &adamw_step, "fused AdamW step");
+  m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
+}

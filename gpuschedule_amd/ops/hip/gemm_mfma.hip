@@ -1,0 +1,267 @@
+// MFMA bf16 GEMM for gfx950 (MI355X): C[M,N] = A[M,K] . B[N,K]^T (+bias)
+//
+// Design (CDNA4-first, per the canonical MFMA-GEMM anatomy):
+//  * v_mfma_f32_16x16x32_bf16 tiles, fp32 accumulation in AGPRs.
+//  * 128x128 block tile, BK=64, 256 threads = 4 waves in a 2x2 grid of
+//    64x64 wave tiles (4x4 fragments of 16x16 each).
+//  * Both operands staged [out-dim][K] in LDS (the NT layout makes A
+//    and B staging symmetric and fully coalesced along K).
+//  * XOR bank swizzle (byte ^= (row&7)<<4) on the LDS images so the
+//    per-fragment ds_read_b128 (16 lanes reading 16 different rows at
+//    one 16-B slot) spreads across 8 slots instead of 8-way
+//    conflicting.
+//  * Double-buffered LDS; per K-tile: issue next-tile global loads
+//    early, MFMA on the current buffer, write the staged registers
+//    late (async-stage split, T14), one barrier per tile.
+//  * XCD-aware bijective blockIdx swizzle: MI355X has 8 XCDs with
+//    private L2s; consecutive blocks share B-panels, so giving each
+//    XCD a contiguous grid chunk turns neighbor-tile re-reads into
+//    L2 hits.
+//  * M/N tails predicated; K must be a multiple of 64 (all shipped
+//    model shapes are).
+#include <torch/extension.h>
+#include <c10/cuda/CUDAStream.h>
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int NTHREADS = 256;
+
+// LDS row stride in bytes (BK bf16) and the bank swizzle.
+constexpr int ROWB = BK * 2;  // 128 B
+__device__ __forceinline__ int swz(int row, int byte_in_row) {
+  return row * ROWB + (byte_in_row ^ ((row & 7) << 4));
+}
+
+// Each thread stages 4 x 16B pieces per operand tile:
+// piece p covers row r = p*32 + tid/8, 16B chunk c = tid%8.
+struct StageRegs { short8 v[4]; };
+
+template <bool PRED>
+__device__ __forceinline__ void stage_load(
+    const short* __restrict__ src, long ld, int row0, int rows,
+    int k0, StageRegs& R) {
+  const int tid = threadIdx.x;
+  const int c = (tid & 7) * 8;       // bf16 offset in row
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int r = p * 32 + (tid >> 3);
+    int gr = row0 + r;
+    if (!PRED || gr < rows) {
+      R.v[p] = *reinterpret_cast<const short8*>(src + (long)gr * ld + k0 + c);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) R.v[p][j] = 0;
+    }
+  }
+}
+
+__device__ __forceinline__ void stage_write(char* lds, const StageRegs& R) {
+  const int tid = threadIdx.x;
+  const int cb = (tid & 7) * 16;     // byte offset in row
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int r = p * 32 + (tid >> 3);
+    *reinterpret_cast<short8*>(lds + swz(r, cb)) = R.v[p];
+  }
+}
+
+__global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  // XCD-aware bijective block swizzle (8 XCDs).
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN - 1) / BN;
+  const int bm0 = (wg / nbn) * BM;
+  const int bn0 = (wg % nbn) * BN;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM * BK * 2];
+  // Explicit double-buffer pointers (no runtime-indexed array: rule 20)
+  char* Ac = lds;
+  char* Bc = lds + BM * ROWB;
+  char* An = lds + 2 * BM * ROWB;
+  char* Bn = lds + 3 * BM * ROWB;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = (wid >> 1) * 64;    // wave row offset in tile
+  const int wc = (wid & 1) * 64;     // wave col offset
+
+  const bool mtail = bm0 + BM > M;
+  const bool ntail = bn0 + BN > N;
+
+  float4v acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  StageRegs ra, rb;
+  if (mtail) stage_load<true>(A, K, bm0, M, 0, ra);
+  else       stage_load<false>(A, K, bm0, M, 0, ra);
+  if (ntail) stage_load<true>(B, K, bn0, N, 0, rb);
+  else       stage_load<false>(B, K, bn0, N, 0, rb);
+  stage_write(Ac, ra);
+  stage_write(Bc, rb);
+  __syncthreads();
+
+  const int nk = K / BK;
+  // Fragment LDS byte offsets (row varies per lane: swizzle applies).
+  const int arow = wr + (lane & 15);
+  const int brow = wc + (lane & 15);
+  const int kb = (lane >> 4) * 16;   // 8 bf16 = 16 B per k-subtile slot
+
+  for (int t = 0; t < nk; ++t) {
+    if (t + 1 < nk) {
+      if (mtail) stage_load<true>(A, K, bm0, M, (t + 1) * BK, ra);
+      else       stage_load<false>(A, K, bm0, M, (t + 1) * BK, ra);
+      if (ntail) stage_load<true>(B, K, bn0, N, (t + 1) * BK, rb);
+      else       stage_load<false>(B, K, bn0, N, (t + 1) * BK, rb);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bf[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        af[f] = *reinterpret_cast<const bf16x8*>(
+            Ac + swz(arow + f * 16, kk * 64 + kb));
+        bf[f] = *reinterpret_cast<const bf16x8*>(
+            Bc + swz(brow + f * 16, kk * 64 + kb));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    if (t + 1 < nk) {
+      stage_write(An, ra);
+      stage_write(Bn, rb);
+      char* tmp = Ac; Ac = An; An = tmp;
+      tmp = Bc; Bc = Bn; Bn = tmp;
+    }
+    __syncthreads();
+  }
+
+  // Epilogue: C row = bm0+wr+i*16+(lane>>4)*4+r, col = bn0+wc+j*16+(lane&15)
+  const int crow0 = bm0 + wr + (lane >> 4) * 4;
+  const int ccol0 = bn0 + wc + (lane & 15);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = ccol0 + j * 16;
+      if (ntail && col >= N) continue;
+      float bv = bias ? bf16_to_f32(bias[col]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + i * 16 + r;
+        if (mtail && row >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (relu) v = fmaxf(v, 0.f);
+        C[(long)row * N + col] = f32_to_bf16(v);
+      }
+    }
+  }
+}
+
+// ---- bf16 2-D transpose (LDS-tiled, vectorized loads) --------------
+constexpr int TT = 64;  // tile
+__global__ __launch_bounds__(256) void transpose_kernel(
+    const short* __restrict__ in, short* __restrict__ out,
+    int R, int Cc) {
+  __shared__ short tile[TT][TT + 8];
+  int tr0 = blockIdx.y * TT, tc0 = blockIdx.x * TT;
+  int tid = threadIdx.x;
+  // load: 256 threads x 8 cols = 2048 elems/pass; 2 passes for 64x64
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int r = p * 32 + (tid >> 3);
+    int c = (tid & 7) * 8;
+    int gr = tr0 + r, gc = tc0 + c;
+    if (gr < R) {
+      if (gc + 7 < Cc) {
+        short8 v = *reinterpret_cast<const short8*>(in + (long)gr * Cc + gc);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) tile[r][c + j] = v[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          tile[r][c + j] = (gc + j < Cc) ? in[(long)gr * Cc + gc + j] : 0;
+      }
+    }
+  }
+  __syncthreads();
+  // store: out[c][r]; thread writes 8 contiguous out-cols (= in-rows)
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int c = p * 32 + (tid >> 3);   // in col = out row
+    int r = (tid & 7) * 8;         // in row = out col
+    int gor = tc0 + c, goc = tr0 + r;
+    if (gor < Cc) {
+      if (goc + 7 < R) {
+        short8 v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = tile[r + j][c];
+        *reinterpret_cast<short8*>(out + (long)gor * R + goc) = v;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (goc + j < R) out[(long)gor * R + goc + j] = tile[r + j][c];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> bias, bool relu) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16, "gemm_nt: bf16 only");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm_nt: 2-D only");
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
+  TORCH_CHECK(K % BK == 0, "gemm_nt: K must be a multiple of ", BK);
+  auto c = torch::empty({M, N}, a.options());
+  const short* bias_p = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kBFloat16 &&
+                bias->numel() == N);
+    bias_p = (const short*)bias->data_ptr();
+  }
+  int grid = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gemm_nt_kernel, dim3(grid), dim3(NTHREADS), 0,
+                     stream.stream(), (const short*)a.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                     bias_p, M, N, K, relu ? 1 : 0);
+  HIP_CHECK_LAST();
+  return c;
+}
+
+torch::Tensor transpose2d(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  const int R = x.size(0), C = x.size(1);
+  auto out = torch::empty({C, R}, x.options());
+  dim3 grid((C + TT - 1) / TT, (R + TT - 1) / TT);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(transpose_kernel, grid, dim3(256), 0, stream.stream(),
+                     (const short*)x.data_ptr(), (short*)out.data_ptr(),
+                     R, C);
+  HIP_CHECK_LAST();
+  return out;
+}

@@ -1,0 +1,295 @@
+// Fused LayerNorm / RMSNorm forward+backward for gfx950 (MI355X).
+//
+// Memory-bound kernels: the design goal is the HBM roofline, so all
+// bf16 traffic is vectorized as short8 (16 B/lane) per guideline G13,
+// rows are block-resident (one 256-thread block per row, chunks held
+// in registers with compile-time indexing), and reductions are 64-wide
+// wave shuffles + one LDS hop (no atomics in the row path).
+// Grad wrt gamma/beta uses per-block partials reduced by a second tiny
+// kernel (guideline 12: no global atomic contention).
+#include <torch/extension.h>
+#include <c10/cuda/CUDAStream.h>
+#include "common.h"
+
+namespace {
+
+constexpr int NT = 256;    // threads per block (4 waves)
+constexpr int VEC = 8;     // bf16 per vector load
+constexpr int CHUNK = NT * VEC;  // elements covered per chunk pass
+
+// ---------------- forward ------------------------------------------
+// one block per row; NCHUNK = ceil(D / 2048) template-dispatched so the
+// row stays in registers (runtime-indexed arrays go to scratch).
+template <int NCHUNK, bool RMS>
+__global__ __launch_bounds__(NT) void norm_fwd_kernel(
+    const short* __restrict__ x, const short* __restrict__ gamma,
+    const short* __restrict__ beta, short* __restrict__ y,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    int n_rows, int D, float eps) {
+  __shared__ float scratch[16];
+  for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const short* xr = x + (long)row * D;
+    short8 xv[NCHUNK];
+    float sum = 0.f, sq = 0.f;
+#pragma unroll
+    for (int c = 0; c < NCHUNK; ++c) {
+      int i = c * CHUNK + threadIdx.x * VEC;
+      if (i < D) {
+        xv[c] = *reinterpret_cast<const short8*>(xr + i);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float f = bf16_to_f32(xv[c][j]);
+          sum += f;
+          sq += f * f;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) xv[c][j] = 0;
+      }
+    }
+    float mean = 0.f;
+    if (!RMS) {
+      mean = block_reduce_sum<NT / WAVE>(sum, scratch) / D;
+      __syncthreads();
+    }
+    float ms = block_reduce_sum<NT / WAVE>(sq, scratch) / D;
+    float rstd = rsqrtf((RMS ? ms : ms - mean * mean) + eps);
+    if (threadIdx.x == 0) {
+      if (!RMS && mean_out) mean_out[row] = mean;
+      if (rstd_out) rstd_out[row] = rstd;
+    }
+    short* yr = y + (long)row * D;
+#pragma unroll
+    for (int c = 0; c < NCHUNK; ++c) {
+      int i = c * CHUNK + threadIdx.x * VEC;
+      if (i < D) {
+        short8 gv = *reinterpret_cast<const short8*>(gamma + i);
+        short8 out;
+        if (!RMS) {
+          short8 bv = *reinterpret_cast<const short8*>(beta + i);
+#pragma unroll
+          for (int j = 0; j < VEC; ++j) {
+            float f = (bf16_to_f32(xv[c][j]) - mean) * rstd;
+            out[j] = f32_to_bf16(f * bf16_to_f32(gv[j]) +
+                                 bf16_to_f32(bv[j]));
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < VEC; ++j) {
+            float f = bf16_to_f32(xv[c][j]) * rstd;
+            out[j] = f32_to_bf16(f * bf16_to_f32(gv[j]));
+          }
+        }
+        *reinterpret_cast<short8*>(yr + i) = out;
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------- backward (dx + per-block dgamma/dbeta partials) ---
+template <int NCHUNK, bool RMS>
+__global__ __launch_bounds__(NT) void norm_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const short* __restrict__ gamma, short* __restrict__ dx,
+    float* __restrict__ part_dg, float* __restrict__ part_db,
+    int n_rows, int D) {
+  __shared__ float scratch[16];
+  float acc_dg[NCHUNK][VEC];
+  float acc_db[NCHUNK][VEC];
+#pragma unroll
+  for (int c = 0; c < NCHUNK; ++c)
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) { acc_dg[c][j] = 0.f; acc_db[c][j] = 0.f; }
+
+  for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const short* dyr = dy + (long)row * D;
+    const short* xr = x + (long)row * D;
+    const float mu = RMS ? 0.f : mean[row];
+    const float rs = rstd[row];
+    short8 dyv[NCHUNK], xv[NCHUNK], gv[NCHUNK];
+    float c1 = 0.f, c2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < NCHUNK; ++c) {
+      int i = c * CHUNK + threadIdx.x * VEC;
+      if (i < D) {
+        dyv[c] = *reinterpret_cast<const short8*>(dyr + i);
+        xv[c] = *reinterpret_cast<const short8*>(xr + i);
+        gv[c] = *reinterpret_cast<const short8*>(gamma + i);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float h = bf16_to_f32(dyv[c][j]) * bf16_to_f32(gv[c][j]);
+          float xn = (bf16_to_f32(xv[c][j]) - mu) * rs;
+          c1 += h;
+          c2 += h * xn;
+        }
+      }
+    }
+    if (!RMS) {
+      c1 = block_reduce_sum<NT / WAVE>(c1, scratch) / D;
+      __syncthreads();
+    }
+    c2 = block_reduce_sum<NT / WAVE>(c2, scratch) / D;
+    short* dxr = dx + (long)row * D;
+#pragma unroll
+    for (int c = 0; c < NCHUNK; ++c) {
+      int i = c * CHUNK + threadIdx.x * VEC;
+      if (i < D) {
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float d = bf16_to_f32(dyv[c][j]);
+          float h = d * bf16_to_f32(gv[c][j]);
+          float xn = (bf16_to_f32(xv[c][j]) - mu) * rs;
+          float v = RMS ? rs * (h - xn * c2)
+                        : rs * (h - c1 - xn * c2);
+          out[j] = f32_to_bf16(v);
+          acc_dg[c][j] += d * xn;
+          acc_db[c][j] += d;
+        }
+        *reinterpret_cast<short8*>(dxr + i) = out;
+      }
+    }
+    __syncthreads();
+  }
+  // per-block partials (block owns slice blockIdx.x)
+#pragma unroll
+  for (int c = 0; c < NCHUNK; ++c) {
+    int i = c * CHUNK + threadIdx.x * VEC;
+    if (i < D) {
+      float* pg = part_dg + (long)blockIdx.x * D + i;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) pg[j] = acc_dg[c][j];
+      if (part_db) {
+        float* pb = part_db + (long)blockIdx.x * D + i;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) pb[j] = acc_db[c][j];
+      }
+    }
+  }
+}
+
+// partials [P][D] f32 -> grad bf16 [D]
+__global__ void reduce_partials_kernel(const float* __restrict__ part,
+                                       short* __restrict__ out,
+                                       int P, int D) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= D) return;
+  float s = 0.f;
+  for (int p = 0; p < P; ++p) s += part[(long)p * D + i];
+  out[i] = f32_to_bf16(s);
+}
+
+int pick_nchunk(int D) {
+  int need = (D + CHUNK - 1) / CHUNK;
+  for (int c : {1, 2, 4, 8})
+    if (need <= c) return c;
+  TORCH_CHECK(false, "norm: D too large (max ", 8 * CHUNK, "), got ", D);
+  return -1;
+}
+
+#define DISPATCH_NCHUNK(NC, RMS, ...)                 \
+  switch (NC) {                                       \
+    case 1: { constexpr int kNC = 1; __VA_ARGS__; break; } \
+    case 2: { constexpr int kNC = 2; __VA_ARGS__; break; } \
+    case 4: { constexpr int kNC = 4; __VA_ARGS__; break; } \
+    case 8: { constexpr int kNC = 8; __VA_ARGS__; break; } \
+  }
+
+void check_norm_inputs(const torch::Tensor& x, const torch::Tensor& gamma) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous CUDA");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  TORCH_CHECK(gamma.scalar_type() == torch::kBFloat16, "gamma must be bf16");
+  TORCH_CHECK(x.size(-1) % VEC == 0, "D must be a multiple of 8");
+}
+
+}  // namespace
+
+// y, mean, rstd = layernorm_fwd(x, gamma, beta, eps); RMS: mean unused.
+std::vector<torch::Tensor> norm_fwd(torch::Tensor x, torch::Tensor gamma,
+                                    c10::optional<torch::Tensor> beta,
+                                    double eps, bool rms) {
+  check_norm_inputs(x, gamma);
+  const int D = x.size(-1);
+  const long n_rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto mean = rms ? torch::Tensor() : torch::empty({n_rows}, fopt);
+  auto rstd = torch::empty({n_rows}, fopt);
+  int nc = pick_nchunk(D);
+  int grid = (int)std::min<long>(n_rows, 8192);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  const short* beta_p = nullptr;
+  if (!rms) {
+    TORCH_CHECK(beta.has_value(), "layernorm needs beta");
+    beta_p = (const short*)beta->data_ptr();
+  }
+  DISPATCH_NCHUNK(nc, rms, {
+    if (rms) {
+      hipLaunchKernelGGL((norm_fwd_kernel<kNC, true>), dim3(grid), dim3(NT),
+                         0, stream.stream(), (const short*)x.data_ptr(),
+                         (const short*)gamma.data_ptr(), nullptr,
+                         (short*)y.data_ptr(), nullptr,
+                         rstd.data_ptr<float>(), (int)n_rows, D, (float)eps);
+    } else {
+      hipLaunchKernelGGL((norm_fwd_kernel<kNC, false>), dim3(grid), dim3(NT),
+                         0, stream.stream(), (const short*)x.data_ptr(),
+                         (const short*)gamma.data_ptr(), beta_p,
+                         (short*)y.data_ptr(), mean.data_ptr<float>(),
+                         rstd.data_ptr<float>(), (int)n_rows, D, (float)eps);
+    }
+  });
+  HIP_CHECK_LAST();
+  return {y, mean.defined() ? mean : rstd, rstd};
+}
+
+std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                    c10::optional<torch::Tensor> mean,
+                                    torch::Tensor rstd, torch::Tensor gamma,
+                                    bool rms) {
+  check_norm_inputs(x, gamma);
+  TORCH_CHECK(dy.is_contiguous() && dy.scalar_type() == torch::kBFloat16);
+  const int D = x.size(-1);
+  const long n_rows = x.numel() / D;
+  auto dx = torch::empty_like(x);
+  int nc = pick_nchunk(D);
+  int grid = (int)std::min<long>(n_rows, 512);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto part_dg = torch::empty({grid, D}, fopt);
+  auto part_db = rms ? torch::Tensor() : torch::empty({grid, D}, fopt);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  DISPATCH_NCHUNK(nc, rms, {
+    if (rms) {
+      hipLaunchKernelGGL((norm_bwd_kernel<kNC, true>), dim3(grid), dim3(NT),
+                         0, stream.stream(), (const short*)dy.data_ptr(),
+                         (const short*)x.data_ptr(), nullptr,
+                         rstd.data_ptr<float>(),
+                         (const short*)gamma.data_ptr(),
+                         (short*)dx.data_ptr(), part_dg.data_ptr<float>(),
+                         nullptr, (int)n_rows, D);
+    } else {
+      hipLaunchKernelGGL((norm_bwd_kernel<kNC, false>), dim3(grid), dim3(NT),
+                         0, stream.stream(), (const short*)dy.data_ptr(),
+                         (const short*)x.data_ptr(),
+                         mean->data_ptr<float>(), rstd.data_ptr<float>(),
+                         (const short*)gamma.data_ptr(),
+                         (short*)dx.data_ptr(), part_dg.data_ptr<float>(),
+                         part_db.data_ptr<float>(), (int)n_rows, D);
+    }
+  });
+  auto dgamma = torch::empty({D}, x.options());
+  int rgrid = (D + NT - 1) / NT;
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rgrid), dim3(NT), 0,
+                     stream.stream(), part_dg.data_ptr<float>(),
+                     (short*)dgamma.data_ptr(), grid, D);
+  torch::Tensor dbeta;
+  if (!rms) {
+    dbeta = torch::empty({D}, x.options());
+    hipLaunchKernelGGL(reduce_partials_kernel, dim3(rgrid), dim3(NT), 0,
+                       stream.stream(), part_db.data_ptr<float>(),
+                       (short*)dbeta.data_ptr(), grid, D);
+  }
+  HIP_CHECK_LAST();
+  return {dx, dgamma, rms ? dgamma : dbeta};
+}

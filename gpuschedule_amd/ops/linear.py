@@ -1,0 +1,93 @@
+"""MFMA-backed Linear layer.
+
+Forward: C = x @ W^T (+bias) on the hand-written bf16 MFMA GEMM
+(ops/hip/gemm_mfma.hip, NT layout).  Backward reuses the same NT kernel
+with fast LDS-tiled transposes:
+
+    dx[M,K] = dy[M,N] @ W[N,K]        = gemm_nt(dy, W^T)
+    dW[N,K] = dy^T[N,M] @ x[M,K]      = gemm_nt(dy^T, x^T)
+    db      = dy.sum over rows
+
+The reduction dim of every gemm_nt call must be a multiple of 64; the
+wrapper zero-pads the token dim when needed (zero rows contribute
+nothing to dW / db).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ._ext import get_ext
+
+
+def _pad_rows64(t: torch.Tensor) -> torch.Tensor:
+    m = t.shape[0]
+    pad = (-m) % 64
+    if pad == 0:
+        return t
+    return F.pad(t, (0, 0, 0, pad))
+
+
+class _MFMALinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, weight, bias, relu):
+        ext = get_ext(required=True)
+        y = ext.gemm_nt(x2d, weight, bias, relu)
+        ctx.save_for_backward(x2d, weight, y if relu else None)
+        ctx.relu = relu
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        x2d, weight, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        if ctx.relu:
+            dy = dy * (y > 0).to(dy.dtype)
+        wt = ext.transpose2d(weight)            # [K, N]
+        dx = ext.gemm_nt(dy, wt, None, False)   # [M, K]
+        dyp = _pad_rows64(dy)
+        xp = _pad_rows64(x2d)
+        dyt = ext.transpose2d(dyp)              # [N, Mp]
+        xt = ext.transpose2d(xp)                # [K, Mp]
+        dw = ext.gemm_nt(dyt, xt, None, False)  # [N, K]
+        db = dy.sum(dim=0) if ctx.has_bias else None
+        return dx, dw, db, None
+
+
+class MFMALinear(nn.Module):
+    """Drop-in nn.Linear running on the gfx950 MFMA GEMM.
+
+    ``relu=True`` fuses the activation into the GEMM epilogue (and its
+    mask into backward).
+    """
+
+    def __init__(self, in_features: int, out_features: int,
+                 bias: bool = True, relu: bool = False):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        self.relu = relu
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
+
+    def forward(self, x):
+        shape = x.shape
+        x2d = x.reshape(-1, shape[-1])
+        if x.is_cuda:
+            x2d = x2d.contiguous()
+            y = _MFMALinearFn.apply(x2d, self.weight, self.bias, self.relu)
+        else:
+            y = F.linear(x2d, self.weight, self.bias)
+            if self.relu:
+                y = F.relu(y)
+        return y.reshape(*shape[:-1], self.out_features)
+
+    def extra_repr(self):
+        return (f"in={self.in_features}, out={self.out_features}, "
+                f"bias={self.bias is not None}, relu={self.relu}")
