@@ -1,0 +1,107 @@
+"""Per-rank job runner: owns a job's replica and executes iterations.
+
+One JobRunner per (job, rank).  The model lives on this rank's GPU (or
+CPU in tests); gang members share a pre-created RCCL group and reduce
+grads through parallel/ddp.GradReducer.  Weights are seeded by job_id
+so every gang member constructs an identical replica without any
+broadcast; data is seeded by (job, rank, iteration) so DP shards
+differ and a migrated/restored job replays a deterministic stream.
+"""
+from __future__ import annotations
+
+import time
+import zlib
+from typing import Optional
+
+import torch
+
+from ..models import build_model, compute_loss
+from ..ops.optim import FusedAdamW, FusedSGD
+from ..parallel.ddp import GradReducer
+
+
+def _job_seed(job_id: str) -> int:
+    return zlib.crc32(job_id.encode()) & 0x7FFFFFFF
+
+
+class JobRunner:
+    def __init__(self, job_id: str, model_name: str, batch_size: int,
+                 device, dtype, gang_group, gang_rank: int,
+                 gang_world: int, done_iters: int = 0,
+                 state: Optional[dict] = None):
+        self.job_id = job_id
+        self.model_name = model_name
+        self.batch_size = batch_size
+        self.device = device
+        self.dtype = dtype
+        self.gang_group = gang_group
+        self.gang_rank = gang_rank
+        self.gang_world = gang_world
+        self.done_iters = done_iters
+
+        self.model, self.make_batch, default_bs = build_model(
+            model_name, device=device, dtype=dtype,
+            seed=_job_seed(job_id))
+        if device == "cpu" or str(device) == "cpu":
+            self.model = self.model.float()
+        self.batch_size = batch_size or default_bs
+        if model_name.startswith("resnet"):
+            self.optim = FusedSGD(self.model.parameters(), lr=0.05,
+                                  momentum=0.9, weight_decay=5e-4)
+        else:
+            self.optim = FusedAdamW(self.model.parameters(), lr=1e-4,
+                                    weight_decay=0.01)
+        if state is not None:
+            self.load_state(state)
+        self.reducer = GradReducer(self.model, gang_group)
+        self.last_loss = float("nan")
+
+    # ---- state movement -----------------------------------------
+
+    def state_dict(self) -> dict:
+        return {"model": self.model.state_dict(),
+                "optim": self.optim.state_dict(),
+                "done_iters": self.done_iters}
+
+    def load_state(self, state: dict) -> None:
+        self.model.load_state_dict(state["model"])
+        self.optim.load_state_dict(state["optim"])
+        self.done_iters = state.get("done_iters", self.done_iters)
+
+    # ---- execution ----------------------------------------------
+
+    def _batch(self, it: int):
+        seed = (_job_seed(self.job_id) * 1000003 + it * 8191 +
+                self.gang_rank) & 0x7FFFFFFF
+        dev = torch.device(self.device)
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        # generate on CPU (deterministic across devices), move over
+        x, y = self.make_batch(self.batch_size, "cpu", gen)
+        in_dtype = self.dtype if x.is_floating_point() else None
+        nb = str(dev) != "cpu"
+        x = x.to(dev, dtype=in_dtype, non_blocking=nb)
+        y = y.to(dev, non_blocking=nb)
+        return x, y
+
+    def step(self, n_iters: int = 1) -> tuple:
+        """Run up to n_iters training iterations; returns
+        (iters_done, wall_seconds)."""
+        t0 = time.perf_counter()
+        for _ in range(n_iters):
+            x, y = self._batch(self.done_iters)
+            self.optim.zero_grad(set_to_none=False)
+            loss = compute_loss(self.model_name, self.model, x, y)
+            loss.backward()
+            self.reducer.finalize()
+            self.optim.step()
+            self.last_loss = float(loss.detach())
+            self.done_iters += 1
+        if str(self.device) != "cpu":
+            torch.cuda.synchronize(self.device)
+        return n_iters, time.perf_counter() - t0
+
+    def release(self) -> None:
+        """Drop GPU state (eviction or completion)."""
+        self.reducer.detach()
+        self.model = None
+        self.optim = None

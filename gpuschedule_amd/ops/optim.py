@@ -14,7 +14,27 @@ import torch
 from ._ext import get_ext
 
 
-class FusedAdamW(torch.optim.Optimizer):
+class _MasterStateMixin:
+    """Keep fp32 master/state fp32 across load_state_dict (torch's
+    default casts floating state to the param dtype, which would
+    quantize the masters to bf16 on restore)."""
+
+    _FP32_KEYS = ("p32", "m", "v", "buf")
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        for group in self.param_groups:
+            for p in group["params"]:
+                st = self.state.get(p)
+                if not st:
+                    continue
+                for k in self._FP32_KEYS:
+                    if k in st and isinstance(st[k], torch.Tensor):
+                        st[k] = st[k].to(device=p.device,
+                                         dtype=torch.float32)
+
+
+class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=0.01):
         defaults = dict(lr=lr, betas=betas, eps=eps,
@@ -57,7 +77,7 @@ class FusedAdamW(torch.optim.Optimizer):
         return loss
 
 
-class FusedSGD(torch.optim.Optimizer):
+class FusedSGD(_MasterStateMixin, torch.optim.Optimizer):
     def __init__(self, params, lr=0.1, momentum=0.9, weight_decay=5e-4):
         defaults = dict(lr=lr, momentum=momentum,
                         weight_decay=weight_decay)

@@ -76,7 +76,11 @@ class Scheduler:
         # Reject unsatisfiable jobs at admission (they would otherwise
         # starve in the queue forever).
         width = spec.min_gpus if spec.elastic else spec.n_gpus
-        mem = catalog.job_mem_bytes(spec)
+        try:
+            mem = catalog.job_mem_bytes(spec)
+        except ValueError as e:  # unknown model family
+            self.on_fail(spec.job_id, str(e), job.submit_t)
+            return
         too_wide = width > min(self.cluster.n_gpus,
                                self.cluster.gpus_per_node)
         too_big = mem > max(g.hbm_bytes for g in self.cluster.gpus)
@@ -351,6 +355,8 @@ class Scheduler:
 
     def on_fail(self, job_id: str, err: str, now: float) -> None:
         job = self.jobs[job_id]
+        if not job.is_active:
+            return
         self.cluster.release(job)
         job.state = JobState.FAILED
         job.finish_t = now

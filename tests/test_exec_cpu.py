@@ -1,0 +1,158 @@
+"""Execution-backend tests on CPU: single process and gloo world-2."""
+import os
+import sys
+
+import pytest
+import torch
+
+from gpuschedule_amd import Cluster, JobSpec, Scheduler, make_policy
+from gpuschedule_amd.exec import CheckpointStore, JobRunner, TorchBackend
+
+
+def J(jid, t=0.0, n=1, iters=4, model="llama3-tiny", batch=2, **kw):
+    return JobSpec(job_id=jid, submit_time=t, n_gpus=n, total_iters=iters,
+                   model=model, batch_size=batch, **kw)
+
+
+def make_backend(specs, policy_name="fifo", rank=0, world=1, **kw):
+    sched = None
+    if rank == 0:
+        sched = Scheduler(Cluster(world, world), make_policy(policy_name))
+        sched.submit(specs)
+    return TorchBackend(sched, rank=rank, world=world, device_type="cpu",
+                        iters_per_tick=2, **kw)
+
+
+# ---- single process -----------------------------------------------
+
+def test_single_rank_runs_trace(tmp_path):
+    specs = [J("a", iters=4), J("b", t=0.0, iters=3, model="resnet18",
+               batch=4)]
+    be = make_backend(specs, ckpt_root=str(tmp_path / "ck"))
+    log = be.run(max_seconds=300)
+    s = log.summary()
+    assert int(s["n_complete"]) == 2
+    assert be.sched.jobs["a"].done_iters == 4
+
+
+def test_preemption_resume_single_rank(tmp_path):
+    # SRTF: short job preempts the long one mid-flight; both finish.
+    specs = [J("long", iters=12, iter_time_hint=1.0),
+             J("short", t=0.5, iters=2, iter_time_hint=0.001)]
+    be = make_backend(specs, "srtf", ckpt_root=str(tmp_path / "ck"))
+    log = be.run(max_seconds=300)
+    s = log.summary()
+    assert int(s["n_complete"]) == 2
+    assert s["preemptions"] >= 1
+
+
+def test_unknown_model_fails_at_admission(tmp_path):
+    specs = [J("bad", iters=3, model="nosuchmodel"), J("ok", iters=2)]
+    be = make_backend(specs, ckpt_root=str(tmp_path / "ck"))
+    log = be.run(max_seconds=300)
+    events = {e["job"]: e["event"] for e in log.events
+              if e["event"] in ("complete", "fail")}
+    assert events["bad"] == "fail" and events["ok"] == "complete"
+
+
+def test_runtime_failure_isolated(tmp_path):
+    # Inject a model family that crashes mid-training; the other job
+    # must still complete and the cluster must stay healthy.
+    from gpuschedule_amd.models import registry
+    from gpuschedule_amd import catalog as cat
+
+    class Exploding(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.lin = torch.nn.Linear(8, 8)
+            self.n = 0
+
+        def forward(self, x):
+            self.n += 1
+            if self.n >= 2:
+                raise RuntimeError("boom")
+            return self.lin(x)
+
+    registry._REGISTRY["crash"] = (
+        Exploding, registry._img_batch((8,), 8))
+    cat.CATALOG["crash"] = cat.ModelMeta("crash", 4, 1 << 30, 0.01, 0.1)
+    try:
+        specs = [J("bad", iters=5, model="crash", batch=4),
+                 J("ok", iters=2)]
+        be = make_backend(specs, "gandiva", ckpt_root=str(tmp_path / "ck"))
+        log = be.run(max_seconds=300)
+        events = {e["job"]: e["event"] for e in log.events
+                  if e["event"] in ("complete", "fail")}
+        assert events["bad"] == "fail" and events["ok"] == "complete"
+    finally:
+        registry._REGISTRY.pop("crash")
+        cat.CATALOG.pop("crash")
+
+
+def test_runner_checkpoint_determinism(tmp_path):
+    ck = CheckpointStore(str(tmp_path / "ck"))
+    r1 = JobRunner("j", "llama3-tiny", 2, "cpu", torch.float32, None, 0, 1)
+    r1.step(3)
+    ck.save("j", r1.model, r1.optim, r1.done_iters)
+    r1.step(2)
+
+    state = ck.load("j")
+    r2 = JobRunner("j", "llama3-tiny", 2, "cpu", torch.float32, None, 0, 1,
+                   state=state)
+    assert r2.done_iters == 3
+    r2.step(2)
+    for (n1, p1), (n2, p2) in zip(r1.model.named_parameters(),
+                                  r2.model.named_parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5), \
+            f"restore diverged at {n1}"
+
+
+# ---- gloo world-2 ---------------------------------------------------
+
+def _worker(rank, world, port, scenario, out_file):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        if scenario == "gang":
+            specs = [J("gang", n=2, iters=4)]
+            pol = "fifo"
+        elif scenario == "mix":
+            specs = [J("g", n=2, iters=3), J("s1", t=0.2, iters=2),
+                     J("s2", t=0.2, iters=2)]
+            pol = "srtf"
+        else:  # elastic grow
+            specs = [JobSpec(job_id="el", submit_time=0.0, n_gpus=1,
+                             total_iters=6, model="llama3-tiny",
+                             batch_size=2, elastic=True, min_gpus=1,
+                             max_gpus=2)]
+            pol = "elastic"
+        ck = f"/tmp/gpuschedule_test_ck_{port}"
+        be = make_backend(specs, pol, rank=rank, world=world, ckpt_root=ck)
+        log = be.run(max_seconds=300)
+        if rank == 0:
+            import json
+            with open(out_file, "w") as f:
+                json.dump({"summary": log.summary(),
+                           "events": log.events}, f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("scenario", ["gang", "mix", "elastic"])
+def test_world2_scenarios(tmp_path, scenario):
+    import torch.multiprocessing as mp
+    import json
+    port = 29310 + {"gang": 0, "mix": 1, "elastic": 2}[scenario]
+    out = str(tmp_path / "out.json")
+    mp.spawn(_worker, args=(2, port, scenario, out), nprocs=2,
+             join=True)
+    with open(out) as f:
+        res = json.load(f)
+    s = res["summary"]
+    assert s["n_complete"] == s["n_jobs"], res["events"]
+    if scenario == "elastic":
+        assert any(e["event"] == "resize" for e in res["events"]) or \
+            any(e["event"] == "start" and len(e.get("gpus", [])) == 2
+                for e in res["events"])
