@@ -1,0 +1,183 @@
+"""GPU numerics: every HIP kernel vs a plain-torch fp32 reference.
+
+All tolerances are scaled for bf16 I/O with fp32 accumulation.
+Transpose-detecting inputs (asymmetric random) throughout.
+"""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def ext():
+    from gpuschedule_amd.ops import get_ext
+    return get_ext(required=True)
+
+
+def mx(err, ref):
+    return (err.float() - ref.float()).abs().max().item()
+
+
+# ---- norms ---------------------------------------------------------
+
+@pytest.mark.parametrize("shape", [(2048, 768), (512, 4096), (33, 1024),
+                                   (4096, 5120)])
+def test_layernorm_fwd_bwd(shape):
+    torch.manual_seed(0)
+    N, D = shape
+    x = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+
+    y, mean, rstd = ext().norm_fwd(x, g, b, 1e-5, False)
+    xf = x.float().requires_grad_(True)
+    gf = g.float().requires_grad_(True)
+    bf = b.float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(xf, (D,), gf, bf, 1e-5)
+    assert mx(y, ref) < 3e-2
+    ref.backward(dy.float())
+    dx, dgm, dbt = ext().norm_bwd(dy, x, mean, rstd, g, False)
+    assert mx(dx, xf.grad) < 5e-2
+    gs = gf.grad.abs().max().item() + 1.0
+    assert mx(dgm, gf.grad) / gs < 2e-2
+    bs = bf.grad.abs().max().item() + 1.0
+    assert mx(dbt, bf.grad) / bs < 2e-2
+
+
+@pytest.mark.parametrize("shape", [(2048, 768), (512, 4096)])
+def test_rmsnorm_fwd_bwd(shape):
+    torch.manual_seed(1)
+    N, D = shape
+    x = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+    y, _, rstd = ext().norm_fwd(x, g, None, 1e-6, True)
+    xf = x.float().requires_grad_(True)
+    ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) \
+        * g.float()
+    assert mx(y, ref) < 3e-2
+    ref.backward(dy.float())
+    dx, dgm, _ = ext().norm_bwd(dy, x, None, rstd, g, True)
+    assert mx(dx, xf.grad) < 5e-2
+
+
+# ---- GEMM ----------------------------------------------------------
+
+@pytest.mark.parametrize("mnk", [(256, 128, 64), (128, 128, 128),
+                                 (512, 768, 768), (4096, 3072, 768),
+                                 (100, 60, 64), (1000, 130, 192),
+                                 (2048, 2048, 4096)])
+def test_gemm_nt(mnk):
+    torch.manual_seed(2)
+    M, N, K = mnk
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    c = ext().gemm_nt(a, b, None, False)
+    ref = a.float() @ b.float().t()
+    err = (c.float() - ref).abs()
+    tol = 0.02 * K ** 0.5  # fp32-accum bf16 roundoff grows ~sqrt(K)
+    assert err.max().item() < tol, f"max err {err.max().item()} tol {tol}"
+
+
+def test_gemm_identity_transpose_detecting():
+    # A = I with ASYMMETRIC B: catches row/col-swapped C writes.
+    K = 128
+    a = torch.eye(K, device="cuda", dtype=torch.bfloat16)
+    b = torch.arange(K * K, device="cuda",
+                     dtype=torch.float32).reshape(K, K) % 37
+    b = b.to(torch.bfloat16)
+    c = ext().gemm_nt(a, b, None, False)
+    assert torch.equal(c.float(), b.float().t())
+
+
+def test_gemm_bias_relu():
+    torch.manual_seed(3)
+    a = torch.randn(256, 256, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(192, 256, device="cuda", dtype=torch.bfloat16)
+    bias = torch.randn(192, device="cuda", dtype=torch.bfloat16)
+    c = ext().gemm_nt(a, b, bias, True)
+    ref = torch.relu(a.float() @ b.float().t() + bias.float())
+    assert mx(c, ref) < 0.4
+    assert (c.float() >= 0).all()
+
+
+def test_transpose2d():
+    torch.manual_seed(4)
+    for shape in [(128, 256), (100, 60), (4096, 768), (130, 1000)]:
+        x = torch.randn(*shape, device="cuda", dtype=torch.bfloat16)
+        assert torch.equal(ext().transpose2d(x), x.t().contiguous())
+
+
+def test_mfma_linear_autograd():
+    from gpuschedule_amd.ops import MFMALinear
+    torch.manual_seed(5)
+    lin = MFMALinear(768, 512).cuda().bfloat16()
+    x = torch.randn(300, 768, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = lin(x)
+    loss = y.float().square().mean()
+    loss.backward()
+
+    xf = x.detach().float().requires_grad_(True)
+    wf = lin.weight.detach().float().requires_grad_(True)
+    bf = lin.bias.detach().float().requires_grad_(True)
+    reff = torch.nn.functional.linear(xf, wf, bf)
+    reff.float().square().mean().backward()
+    assert mx(y, reff) < 0.2
+    scale = xf.grad.abs().max().item() + 1e-6
+    assert mx(x.grad, xf.grad) / scale < 0.05
+    scale = wf.grad.abs().max().item() + 1e-6
+    assert mx(lin.weight.grad, wf.grad) / scale < 0.05
+
+
+# ---- optimizers ----------------------------------------------------
+
+def test_fused_adamw_matches_reference():
+    torch.manual_seed(6)
+    n = 4099
+    p32 = torch.randn(n, device="cuda")
+    p = p32.bfloat16()
+    g = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    pr, mr, vr = p32.clone(), m.clone(), v.clone()
+    for t in (1, 2, 3):
+        ext().adamw_step(p, p32, g, m, v, 1e-2, 0.9, 0.999, 1e-8, 0.01, t)
+        gf = g.float()
+        mr.mul_(0.9).add_(gf, alpha=0.1)
+        vr.mul_(0.999).addcmul_(gf, gf, value=0.001)
+        mh = mr / (1 - 0.9 ** t)
+        vh = vr / (1 - 0.999 ** t)
+        pr.add_(mh / (vh.sqrt() + 1e-8) + 0.01 * pr, alpha=-1e-2)
+    assert mx(p32, pr) < 1e-5
+    assert mx(p, pr.bfloat16()) == 0
+
+
+def test_fused_sgd_matches_reference():
+    torch.manual_seed(7)
+    n = 1027
+    p32 = torch.randn(n, device="cuda")
+    p = p32.bfloat16()
+    g = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    buf = torch.zeros(n, device="cuda")
+    pr, br = p32.clone(), buf.clone()
+    for _ in range(3):
+        ext().sgd_step(p32, p, g, buf, 0.1, 0.9, 5e-4)
+        gf = g.float() + 5e-4 * pr
+        br.mul_(0.9).add_(gf)
+        pr.add_(br, alpha=-0.1)
+    assert mx(p32, pr) < 1e-5
+
+
+# ---- end-to-end single job on GPU ---------------------------------
+
+def test_resnet18_job_step_on_gpu():
+    from gpuschedule_amd.exec import JobRunner
+    r = JobRunner("gputest", "resnet18", 32, "cuda:0", torch.bfloat16,
+                  None, 0, 1)
+    losses = []
+    for _ in range(6):
+        r.step(1)
+        losses.append(r.last_loss)
+    assert all(torch.isfinite(torch.tensor(losses)))
