@@ -31,11 +31,22 @@ def _pad_rows64(t: torch.Tensor) -> torch.Tensor:
     return F.pad(t, (0, 0, 0, pad))
 
 
+def _gemm_nt(ext, a, b, bias=None, relu=False):
+    """gemm_nt with zero-padding of the reduction dim when K % 64 != 0
+    (zero columns contribute nothing to the product)."""
+    k = a.shape[1]
+    pad = (-k) % 64
+    if pad:
+        a = F.pad(a, (0, pad))
+        b = F.pad(b, (0, pad))
+    return ext.gemm_nt(a.contiguous(), b.contiguous(), bias, relu)
+
+
 class _MFMALinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, weight, bias, relu):
         ext = get_ext(required=True)
-        y = ext.gemm_nt(x2d, weight, bias, relu)
+        y = _gemm_nt(ext, x2d, weight, bias, relu)
         ctx.save_for_backward(x2d, weight, y if relu else None)
         ctx.relu = relu
         ctx.has_bias = bias is not None
@@ -49,12 +60,12 @@ class _MFMALinearFn(torch.autograd.Function):
         if ctx.relu:
             dy = dy * (y > 0).to(dy.dtype)
         wt = ext.transpose2d(weight)            # [K, N]
-        dx = ext.gemm_nt(dy, wt, None, False)   # [M, K]
+        dx = _gemm_nt(ext, dy, wt)              # [M, K]
         dyp = _pad_rows64(dy)
         xp = _pad_rows64(x2d)
         dyt = ext.transpose2d(dyp)              # [N, Mp]
         xt = ext.transpose2d(xp)                # [K, Mp]
-        dw = ext.gemm_nt(dyt, xt, None, False)  # [N, K]
+        dw = _gemm_nt(ext, dyt, xt)             # [N, K]
         db = dy.sum(dim=0) if ctx.has_bias else None
         return dx, dw, db, None
 

@@ -111,6 +111,11 @@ class Scheduler:
             self.policy._total_gpus = self.cluster.n_gpus
         self.policy.on_tick(now, self.jobs)
         requests = self.policy.rank(now, self.jobs)
+        # Best-effort class ranks strictly below normal jobs (stable).
+        requests = ([r for r in requests
+                     if not self.jobs[r[0]].spec.background] +
+                    [r for r in requests
+                     if self.jobs[r[0]].spec.background])
 
         alloc, evictions = self._allocate(requests)
         plan = self._diff(alloc, evictions, now)

@@ -50,6 +50,10 @@ class JobSpec:
     # Optional oracle iteration time (seconds) for the simulator; the
     # real backend measures it instead.
     iter_time_hint: float = 0.0
+    # Best-effort class: scheduled only onto GPUs no normal job wants
+    # this tick (the scheduler core ranks background jobs last under
+    # every policy).  Used for fill/background workloads.
+    background: bool = False
 
     def __post_init__(self) -> None:
         if self.n_gpus < 1:

@@ -21,7 +21,7 @@ from .job import JobSpec
 
 REQUIRED = ("job_id", "submit_time", "num_gpus", "iterations", "model")
 OPTIONAL = ("batch_size", "mem_bytes", "elastic", "min_gpus", "max_gpus",
-            "iter_time_hint")
+            "iter_time_hint", "background")
 
 
 def _parse_bool(s: str) -> bool:
@@ -57,6 +57,7 @@ def load_trace(src: Union[str, io.TextIOBase]) -> List[JobSpec]:
                 min_gpus=int(row.get("min_gpus") or 1),
                 max_gpus=int(row.get("max_gpus") or 8),
                 iter_time_hint=float(row.get("iter_time_hint") or 0.0),
+                background=_parse_bool(row.get("background") or ""),
             ))
         specs.sort(key=lambda s: (s.submit_time, s.job_id))
         ids = [s.job_id for s in specs]
@@ -81,7 +82,7 @@ def save_trace(specs: Iterable[JobSpec], dst: Union[str, io.TextIOBase]) -> None
             w.writerow([s.job_id, s.submit_time, s.n_gpus, s.total_iters,
                         s.model, s.batch_size, s.mem_bytes,
                         int(s.elastic), s.min_gpus, s.max_gpus,
-                        s.iter_time_hint])
+                        s.iter_time_hint, int(s.background)])
     finally:
         if close:
             dst.close()

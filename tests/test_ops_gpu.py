@@ -35,10 +35,11 @@ def test_layernorm_fwd_bwd(shape):
     gf = g.float().requires_grad_(True)
     bf = b.float().requires_grad_(True)
     ref = torch.nn.functional.layer_norm(xf, (D,), gf, bf, 1e-5)
-    assert mx(y, ref) < 3e-2
+    # bf16 output quantization: 2^-8 relative of the output range
+    assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 8e-3
     ref.backward(dy.float())
     dx, dgm, dbt = ext().norm_bwd(dy, x, mean, rstd, g, False)
-    assert mx(dx, xf.grad) < 5e-2
+    assert mx(dx, xf.grad) / (xf.grad.abs().max().item() + 1e-3) < 1e-2
     gs = gf.grad.abs().max().item() + 1.0
     assert mx(dgm, gf.grad) / gs < 2e-2
     bs = bf.grad.abs().max().item() + 1.0
@@ -56,10 +57,10 @@ def test_rmsnorm_fwd_bwd(shape):
     xf = x.float().requires_grad_(True)
     ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) \
         * g.float()
-    assert mx(y, ref) < 3e-2
+    assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 8e-3
     ref.backward(dy.float())
     dx, dgm, _ = ext().norm_bwd(dy, x, None, rstd, g, True)
-    assert mx(dx, xf.grad) < 5e-2
+    assert mx(dx, xf.grad) / (xf.grad.abs().max().item() + 1e-3) < 1e-2
 
 
 # ---- GEMM ----------------------------------------------------------
