@@ -16,6 +16,8 @@ void adamw_step(torch::Tensor p_bf16, torch::Tensor p32, torch::Tensor grad,
                 double beta2, double eps, double wd, long step);
 void sgd_step(torch::Tensor p32, torch::Tensor p_bf16, torch::Tensor grad,
               torch::Tensor buf, double lr, double momentum, double wd);
+torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                       bool causal);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "fused LayerNorm/RMSNorm forward (bf16)");
@@ -24,4 +26,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
+  m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax)");
 }
