@@ -1,0 +1,123 @@
+"""Command-line interface.
+
+    python -m gpuschedule_amd simulate --policy las --trace t.csv --gpus 8
+    python -m gpuschedule_amd simulate --config configs/c1_fifo_sim.yaml
+    python -m gpuschedule_amd replay   --policy srtf --trace t.csv
+    python -m gpuschedule_amd gen-trace --jobs 64 --mix philly-slice -o t.csv
+
+``simulate`` is the pure-CPU discrete-event path; ``replay`` executes
+real training on GPUs (single process, or under torchrun for N>1).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+import yaml
+
+from . import (Cluster, Scheduler, load_trace, make_policy, make_trace,
+               save_trace, simulate)
+from .trace.synth import make_elastic_trace
+
+
+def _load_config(path):
+    with open(path) as f:
+        return yaml.safe_load(f)
+
+
+def _policy_from_args(args, cfg):
+    name = cfg.get("policy", args.policy)
+    kw = cfg.get("policy_args", {})
+    return make_policy(name, **kw)
+
+
+def _specs_from_args(args, cfg):
+    if cfg.get("trace_file") or args.trace:
+        return load_trace(cfg.get("trace_file") or args.trace)
+    t = cfg.get("synthetic", {})
+    if t.get("kind") == "elastic":
+        return make_elastic_trace(seed=t.get("seed", 0),
+                                  total_iters=t.get("iters", 400))
+    return make_trace(t.get("jobs", args.jobs), mix=t.get("mix", args.mix),
+                      seed=t.get("seed", 0),
+                      mean_interarrival=t.get("mean_interarrival", 10.0),
+                      iters_range=tuple(t.get("iters_range", (50, 400))),
+                      gang_sizes=t.get("gang_sizes"))
+
+
+def cmd_simulate(args):
+    cfg = _load_config(args.config) if args.config else {}
+    pol = _policy_from_args(args, cfg)
+    specs = _specs_from_args(args, cfg)
+    n_gpus = cfg.get("gpus", args.gpus)
+    log, summary = simulate(specs, pol, n_gpus=n_gpus)
+    if args.log:
+        log.save(args.log)
+    print(json.dumps({"policy": pol.name, "gpus": n_gpus,
+                      **{k: round(v, 3) for k, v in summary.items()}}))
+
+
+def cmd_replay(args):
+    import torch
+    from .exec import TorchBackend
+    from .exec.backend import init_distributed
+    cfg = _load_config(args.config) if args.config else {}
+    rank, world = init_distributed("gloo")
+    n_gpus = max(cfg.get("gpus", args.gpus), world)
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    sched = None
+    if rank == 0:
+        sched = Scheduler(Cluster(n_gpus, max(n_gpus, 8)),
+                          _policy_from_args(args, cfg))
+        sched.submit(_specs_from_args(args, cfg))
+    be = TorchBackend(sched, rank=rank, world=world, device_type=device,
+                      iters_per_tick=cfg.get("iters_per_tick", 2),
+                      gang_backend="nccl" if device == "cuda" else "gloo")
+    log = be.run(max_seconds=args.max_seconds)
+    if rank == 0:
+        if args.log:
+            log.save(args.log)
+        print(json.dumps({k: round(v, 3)
+                          for k, v in log.summary().items()}))
+
+
+def cmd_gen_trace(args):
+    specs = make_trace(args.jobs, mix=args.mix, seed=args.seed,
+                       mean_interarrival=args.interarrival)
+    save_trace(specs, args.output)
+    print(f"wrote {len(specs)} jobs -> {args.output}")
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser(prog="gpuschedule_amd")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    common = argparse.ArgumentParser(add_help=False)
+    common.add_argument("--policy", default="fifo")
+    common.add_argument("--trace", default=None)
+    common.add_argument("--config", default=None)
+    common.add_argument("--gpus", type=int, default=8)
+    common.add_argument("--jobs", type=int, default=16)
+    common.add_argument("--mix", default="mixed-cv-nlp")
+    common.add_argument("--log", default=None, help="event log JSONL out")
+
+    p = sub.add_parser("simulate", parents=[common])
+    p.set_defaults(fn=cmd_simulate)
+    p = sub.add_parser("replay", parents=[common])
+    p.add_argument("--max-seconds", type=float, default=None)
+    p.set_defaults(fn=cmd_replay)
+    p = sub.add_parser("gen-trace")
+    p.add_argument("--jobs", type=int, default=64)
+    p.add_argument("--mix", default="philly-slice")
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--interarrival", type=float, default=10.0)
+    p.add_argument("-o", "--output", required=True)
+    p.set_defaults(fn=cmd_gen_trace)
+
+    args = ap.parse_args(argv)
+    args.fn(args)
+
+
+if __name__ == "__main__":
+    main()
