@@ -39,23 +39,30 @@ from gpuschedule_amd.exec.backend import init_distributed
 
 
 def build_trace(n_jobs: int, steps: int, n_gpus: int, iters_per_tick: int,
-                seed: int = 1234):
+                tick_s: float, seed: int = 1234):
     """Single-GPU job mix whose total tick-slot demand fits 70% of the
-    K x N slot budget (identical trace at every N for strong scaling)."""
+    K x N slot budget (identical job set at every N for strong
+    scaling).  Arrivals are Poisson over the first 25% of the expected
+    window, using the warmup-calibrated tick time ``tick_s`` so the
+    arrival process lands inside the K-tick measurement regardless of
+    hardware speed."""
     rng = random.Random(seed)
     budget = int(0.70 * steps * n_gpus) * iters_per_tick
     models = ["resnet50", "bert-base"]
     # log-uniform weights, normalized to the budget
     raw = [10 ** rng.uniform(0, 1.2) for _ in range(n_jobs)]
     scale = budget / sum(raw)
+    arrival_window = 0.25 * steps * tick_s
+    mean_ia = arrival_window / max(1, n_jobs - 1)
     specs = []
     t = 0.0
     for i, r in enumerate(raw):
         iters = max(iters_per_tick, int(r * scale))
         specs.append(JobSpec(
-            job_id=f"job{i:03d}", submit_time=round(t, 3), n_gpus=1,
+            job_id=f"job{i:03d}", submit_time=round(min(t, arrival_window),
+                                                    4), n_gpus=1,
             total_iters=iters, model=models[i % 2]))
-        t += rng.expovariate(1.0 / 2.0)  # ~2s mean interarrival
+        t += rng.expovariate(1.0 / mean_ia) if mean_ia > 0 else 0.0
     # per-GPU fill jobs: best-effort, never complete inside the window
     for g in range(n_gpus):
         specs.append(JobSpec(
@@ -93,11 +100,11 @@ def main():
         torch.cuda.set_device(rank)
     n_jobs = args.jobs or (6 + 2 * n_gpus)
 
-    def make_backend(specs, policy):
+    def make_backend(specs, policy, policy_kw=None):
         sched = None
         if rank == 0:
             sched = Scheduler(Cluster(n_gpus, max(n_gpus, 8)),
-                              make_policy(policy))
+                              make_policy(policy, **(policy_kw or {})))
             sched.submit(specs)
         return TorchBackend(
             sched, rank=rank, world=world, device_type=device_type,
@@ -116,7 +123,16 @@ def main():
                           model=("bert-base" if g % 2 else "resnet50"),
                           background=True) for g in range(n_gpus)]
     wb = make_backend(warm_specs, "fifo")
+    tw0 = time.perf_counter()
     run_phase(wb, max(args.warmup, 1))
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    # calibrated tick time (drop tick 1: model-build cold start)
+    tick_s = (time.perf_counter() - tw0) / max(args.warmup, 1)
+    if world > 1:
+        t = torch.tensor([tick_s], dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        tick_s = float(t.item())
     for r in list(wb.runners.values()):
         r.release()
     wb.runners.clear()
@@ -124,8 +140,17 @@ def main():
         torch.cuda.empty_cache()
 
     # ---- timed region: EXACTLY K ticks of the real trace
-    specs = build_trace(n_jobs, args.steps, n_gpus, args.iters_per_tick)
-    be = make_backend(specs, args.policy)
+    specs = build_trace(n_jobs, args.steps, n_gpus, args.iters_per_tick,
+                        tick_s)
+    policy_kw = {}
+    if args.policy == "las":
+        # LAS thresholds in GPU-seconds, scaled to the workload window
+        horizon = args.steps * tick_s
+        policy_kw = {"thresholds": (horizon / 8, horizon / 2),
+                     "starve_limit": 4.0}
+    elif args.policy == "gandiva":
+        policy_kw = {"quantum": max(4 * tick_s, 1e-3)}
+    be = make_backend(specs, args.policy, policy_kw)
     barrier_sync()
     t0 = time.perf_counter()
     run_phase(be, args.steps)

@@ -40,3 +40,20 @@ def test_simulate_writes_event_log(tmp_path):
             "--log", log)
     lines = [json.loads(x) for x in open(log) if x.strip()]
     assert any(e["event"] == "complete" for e in lines)
+
+
+def test_bench_cpu_contract(tmp_path):
+    """bench.py emits the driver-contract JSON line (CPU debug mode)."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "6", "--warmup", "1",
+         "--device", "cpu", "--jobs", "3"],
+        cwd=REPO, capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stderr
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in res
+    assert res["steps"] == 6 and res["n_gpus"] == 1
+    assert res["higher_is_better"] is False
