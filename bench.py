@@ -39,40 +39,41 @@ from gpuschedule_amd.exec.backend import init_distributed
 
 
 def build_trace(n_jobs: int, steps: int, n_gpus: int, iters_per_tick: int,
-                tick_s: float, seed: int = 1234):
+                seed: int = 1234):
     """Single-GPU job mix whose total tick-slot demand fits 70% of the
     K x N slot budget (identical job set at every N for strong
-    scaling).  Arrivals are Poisson over the first 25% of the expected
-    window, using the warmup-calibrated tick time ``tick_s`` so the
-    arrival process lands inside the K-tick measurement regardless of
-    hardware speed."""
+    scaling).  Arrivals are TICK-INDEXED (injected by the bench loop
+    over the first 25% of the K ticks) so the arrival process lands
+    inside the measurement window regardless of hardware speed.
+
+    Returns (arrivals, fill_specs): arrivals is a list of
+    (arrival_tick, JobSpec)."""
     rng = random.Random(seed)
     budget = int(0.70 * steps * n_gpus) * iters_per_tick
     models = ["resnet50", "bert-base"]
     # log-uniform weights, normalized to the budget
     raw = [10 ** rng.uniform(0, 1.2) for _ in range(n_jobs)]
     scale = budget / sum(raw)
-    arrival_window = 0.25 * steps * tick_s
-    mean_ia = arrival_window / max(1, n_jobs - 1)
-    specs = []
-    t = 0.0
+    arrivals = []
     for i, r in enumerate(raw):
         iters = max(iters_per_tick, int(r * scale))
-        specs.append(JobSpec(
-            job_id=f"job{i:03d}", submit_time=round(min(t, arrival_window),
-                                                    4), n_gpus=1,
-            total_iters=iters, model=models[i % 2]))
-        t += rng.expovariate(1.0 / mean_ia) if mean_ia > 0 else 0.0
+        tick = int(i * (0.25 * steps) / max(1, n_jobs - 1))
+        arrivals.append((tick, JobSpec(
+            job_id=f"job{i:03d}", submit_time=0.0, n_gpus=1,
+            total_iters=iters, model=models[i % 2])))
     # per-GPU fill jobs: best-effort, never complete inside the window
-    for g in range(n_gpus):
-        specs.append(JobSpec(
-            job_id=f"bg{g}", submit_time=0.0, n_gpus=1,
-            total_iters=10**9, model="resnet50", background=True))
-    return specs
+    fill = [JobSpec(job_id=f"bg{g}", submit_time=0.0, n_gpus=1,
+                    total_iters=10**9, model="resnet50", background=True)
+            for g in range(n_gpus)]
+    return arrivals, fill
 
 
-def run_phase(backend, ticks: int):
-    for _ in range(ticks):
+def run_phase(backend, ticks: int, arrivals=None, rank: int = 0):
+    for t in range(ticks):
+        if rank == 0 and arrivals:
+            for at, spec in arrivals:
+                if at == t:
+                    backend.sched.submit_now(spec, backend.now())
         if backend.tick_once():
             break
 
@@ -140,8 +141,8 @@ def main():
         torch.cuda.empty_cache()
 
     # ---- timed region: EXACTLY K ticks of the real trace
-    specs = build_trace(n_jobs, args.steps, n_gpus, args.iters_per_tick,
-                        tick_s)
+    arrivals, fill = build_trace(n_jobs, args.steps, n_gpus,
+                                 args.iters_per_tick)
     policy_kw = {}
     if args.policy == "las":
         # LAS thresholds in GPU-seconds, scaled to the workload window
@@ -150,10 +151,10 @@ def main():
                      "starve_limit": 4.0}
     elif args.policy == "gandiva":
         policy_kw = {"quantum": max(4 * tick_s, 1e-3)}
-    be = make_backend(specs, args.policy, policy_kw)
+    be = make_backend(fill, args.policy, policy_kw)
     barrier_sync()
     t0 = time.perf_counter()
-    run_phase(be, args.steps)
+    run_phase(be, args.steps, arrivals, rank)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -165,7 +166,7 @@ def main():
 
     if rank == 0:
         log = be.sched.log
-        trace_ids = {s.job_id for s in specs if not s.background}
+        trace_ids = {spec.job_id for _, spec in arrivals}
         completes = [e for e in log.events if e["event"] == "complete"
                      and e["job"] in trace_ids]
         submits = {e["job"]: e["t"] for e in log.events

@@ -57,12 +57,21 @@ SAMPLE_TOKENS = {"resnet18": 1, "resnet50": 1, "bert-base": 128,
 
 def build_model(name: str, device="cpu",
                 dtype: torch.dtype = torch.bfloat16, seed: int = 0):
-    """Returns (model, make_batch, per_gpu_batch)."""
+    """Returns (model, make_batch, per_gpu_batch).
+
+    Init happens DIRECTLY on the target device (seconds vs tens of
+    seconds for big models on CPU + transfer); gang replicas stay
+    identical because every rank seeds the same and MI355X GPUs
+    generate identical sequences for identical seeds.
+    """
     ctor, make_batch = _REGISTRY[name]
     torch.manual_seed(seed)
-    model = ctor()
     if str(device) != "cpu":
-        model = model.to(device=device, dtype=dtype)
+        with torch.device(device):
+            model = ctor()
+        model = model.to(dtype=dtype)
+    else:
+        model = ctor()
     return model, make_batch, catalog.model_meta(name).default_batch
 
 

@@ -8,12 +8,15 @@ profiles/.
 """
 import argparse
 import json
-import math
+import os
+import sys
 import time
 
 import torch
 
-from gpuschedule_amd.ops import get_ext
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+from gpuschedule_amd.ops import get_ext  # noqa: E402
 
 
 def timeit(fn, warmup=10, iters=50):
