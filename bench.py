@@ -63,7 +63,9 @@ def build_trace(n_jobs: int, steps: int, n_gpus: int, iters_per_tick: int,
             total_iters=iters, model=models[i % 2])))
     # per-GPU fill jobs: best-effort, never complete inside the window
     fill = [JobSpec(job_id=f"bg{g}", submit_time=0.0, n_gpus=1,
-                    total_iters=10**9, model="resnet50", background=True)
+                    total_iters=10**9,
+                    model=("bert-base" if g % 2 == 0 else "resnet50"),
+                    background=True)
             for g in range(n_gpus)]
     return arrivals, fill
 

@@ -80,6 +80,8 @@ class JobRunner:
         in_dtype = self.dtype if x.is_floating_point() else None
         nb = str(dev) != "cpu"
         x = x.to(dev, dtype=in_dtype, non_blocking=nb)
+        if nb and x.dim() == 4:
+            x = x.contiguous(memory_format=torch.channels_last)
         y = y.to(dev, non_blocking=nb)
         return x, y
 

@@ -70,6 +70,10 @@ def build_model(name: str, device="cpu",
         with torch.device(device):
             model = ctor()
         model = model.to(dtype=dtype)
+        if name.startswith("resnet"):
+            # NHWC: MIOpen's native layout — avoids the per-conv
+            # batched_transpose kernels that dominate NCHW profiles
+            model = model.to(memory_format=torch.channels_last)
     else:
         model = ctor()
     return model, make_batch, catalog.model_meta(name).default_batch

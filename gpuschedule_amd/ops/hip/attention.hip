@@ -257,6 +257,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
 
 }  // namespace
 
+torch::Tensor transpose2d(torch::Tensor x);  // gemm_mfma.hip (batched)
+
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
@@ -267,9 +269,11 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
               "attn_fwd: expand KV heads before calling (GQA)");
   TORCH_CHECK(D == 64 || D == 128, "attn_fwd: D must be 64 or 128");
   TORCH_CHECK(S % 8 == 0, "attn_fwd: S must be a multiple of 8");
-  // pre-transpose V: [B,H,S,D] -> [B,H,D,S] (one memory-bound pass;
-  // lets the PV B-fragments read 8 bf16 contiguous along kv)
-  auto vt = v.transpose(2, 3).contiguous();
+  // pre-transpose V: [B,H,S,D] -> [B,H,D,S] (one memory-bound pass on
+  // the batched LDS-tiled transpose; lets the PV B-fragments read
+  // 8 bf16 contiguous along kv)
+  auto vt = transpose2d(v.reshape({(long)B * H, S, D}))
+                .reshape({B, H, D, S});
   auto o = torch::empty_like(q);
   float scale = 1.0f / sqrtf((float)D);
   int grid = B * H * ((S + 63) / 64);

@@ -177,10 +177,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
 }
 
 // ---- bf16 2-D transpose (LDS-tiled, vectorized loads) --------------
+// grid.z = batch index (stride R*C per matrix) for batched use.
 constexpr int TT = 64;  // tile
 __global__ __launch_bounds__(256) void transpose_kernel(
-    const short* __restrict__ in, short* __restrict__ out,
+    const short* __restrict__ in0, short* __restrict__ out0,
     int R, int Cc) {
+  const short* in = in0 + (long)blockIdx.z * R * Cc;
+  short* out = out0 + (long)blockIdx.z * R * Cc;
   __shared__ short tile[TT][TT + 8];
   int tr0 = blockIdx.y * TT, tc0 = blockIdx.x * TT;
   int tid = threadIdx.x;
@@ -253,11 +256,16 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
 }
 
 torch::Tensor transpose2d(torch::Tensor x) {
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.dim() == 2 || x.dim() == 3,
+              "transpose2d: 2-D or batched 3-D");
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
-  const int R = x.size(0), C = x.size(1);
-  auto out = torch::empty({C, R}, x.options());
-  dim3 grid((C + TT - 1) / TT, (R + TT - 1) / TT);
+  const int batched = x.dim() == 3;
+  const long Bn = batched ? x.size(0) : 1;
+  const int R = x.size(batched ? 1 : 0), C = x.size(batched ? 2 : 1);
+  auto out = batched ? torch::empty({Bn, C, R}, x.options())
+                     : torch::empty({C, R}, x.options());
+  dim3 grid((C + TT - 1) / TT, (R + TT - 1) / TT, Bn);
   auto stream = c10::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(transpose_kernel, grid, dim3(256), 0, stream.stream(),
                      (const short*)x.data_ptr(), (short*)out.data_ptr(),

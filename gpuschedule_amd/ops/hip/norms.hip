@@ -170,15 +170,27 @@ __global__ __launch_bounds__(NT) void norm_bwd_kernel(
   }
 }
 
-// partials [P][D] f32 -> grad bf16 [D]
+// partials [P][D] f32 -> f32 accumulator [D] via device-scope atomics
+// (grid covers D x P/PCHUNK: enough blocks to fill the chip; the old
+// single-pass D/256-block version left 250+ CUs idle and was ~100x
+// slower than the bwd kernel it served).
+constexpr int PCHUNK = 16;
 __global__ void reduce_partials_kernel(const float* __restrict__ part,
-                                       short* __restrict__ out,
+                                       float* __restrict__ acc,
                                        int P, int D) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= D) return;
+  int p0 = blockIdx.y * PCHUNK;
+  int p1 = min(p0 + PCHUNK, P);
   float s = 0.f;
-  for (int p = 0; p < P; ++p) s += part[(long)p * D + i];
-  out[i] = f32_to_bf16(s);
+  for (int p = p0; p < p1; ++p) s += part[(long)p * D + i];
+  atomicAdd(acc + i, s);
+}
+
+__global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
+                                     short* __restrict__ out, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = f32_to_bf16(in[i]);
 }
 
 int pick_nchunk(int D) {
@@ -279,16 +291,25 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
     }
   });
   auto dgamma = torch::empty({D}, x.options());
-  int rgrid = (D + NT - 1) / NT;
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rgrid), dim3(NT), 0,
+  auto acc_g = torch::zeros({D}, fopt);
+  dim3 rgrid((D + NT - 1) / NT, (grid + PCHUNK - 1) / PCHUNK);
+  int cgrid = (D + NT - 1) / NT;
+  hipLaunchKernelGGL(reduce_partials_kernel, rgrid, dim3(NT), 0,
                      stream.stream(), part_dg.data_ptr<float>(),
-                     (short*)dgamma.data_ptr(), grid, D);
+                     acc_g.data_ptr<float>(), grid, D);
+  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(cgrid), dim3(NT), 0,
+                     stream.stream(), acc_g.data_ptr<float>(),
+                     (short*)dgamma.data_ptr(), D);
   torch::Tensor dbeta;
   if (!rms) {
     dbeta = torch::empty({D}, x.options());
-    hipLaunchKernelGGL(reduce_partials_kernel, dim3(rgrid), dim3(NT), 0,
+    auto acc_b = torch::zeros({D}, fopt);
+    hipLaunchKernelGGL(reduce_partials_kernel, rgrid, dim3(NT), 0,
                        stream.stream(), part_db.data_ptr<float>(),
-                       (short*)dbeta.data_ptr(), grid, D);
+                       acc_b.data_ptr<float>(), grid, D);
+    hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(cgrid), dim3(NT), 0,
+                       stream.stream(), acc_b.data_ptr<float>(),
+                       (short*)dbeta.data_ptr(), D);
   }
   HIP_CHECK_LAST();
   return {dx, dgamma, rms ? dgamma : dbeta};
