@@ -36,36 +36,32 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return row * ROWB + (byte_in_row ^ ((row & 7) << 4));
 }
 
-// Each thread stages 4 x 16B pieces per operand tile:
-// piece p covers row r = p*32 + tid/8, 16B chunk c = tid%8.
-struct StageRegs { short8 v[4]; };
-
-template <bool PRED>
-__device__ __forceinline__ void stage_load(
-    const short* __restrict__ src, long ld, int row0, int rows,
-    int k0, StageRegs& R) {
-  const int tid = threadIdx.x;
-  const int c = (tid & 7) * 8;       // bf16 offset in row
+// Tile staging via global_load_lds (direct HBM->LDS DMA, 16 B/lane —
+// the compiler never auto-emits it; width 4->16 alone is worth ~1.7x
+// on this structure).  The LDS destination is wave-uniform base +
+// lane*16 (lane-linear), so the XOR bank swizzle moves to the SOURCE
+// address: lane i loads the swizzle-inverse column of its row, making
+// the linear LDS image identical to the swizzled image the ds_read
+// fragments expect (both-sides-or-neither rule).
+// Wave w stages rows {p*32 + w*8 .. +8} of the [128][BK] tile.
+// Tail rows are CLAMPED (memory-safe; their products only reach C
+// rows/cols that the predicated epilogue never stores).
+__device__ __forceinline__ void glds_stage_tile(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int k0, char* lds_buf, int wid, int lane) {
+  const int rl = lane >> 3;                       // row within stripe
+  const int gcolb = ((lane & 7) * 16) ^ (rl << 4);  // swizzle-inv col
 #pragma unroll
   for (int p = 0; p < 4; ++p) {
-    int r = p * 32 + (tid >> 3);
-    int gr = row0 + r;
-    if (!PRED || gr < rows) {
-      R.v[p] = *reinterpret_cast<const short8*>(src + (long)gr * ld + k0 + c);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) R.v[p][j] = 0;
-    }
-  }
-}
-
-__device__ __forceinline__ void stage_write(char* lds, const StageRegs& R) {
-  const int tid = threadIdx.x;
-  const int cb = (tid & 7) * 16;     // byte offset in row
-#pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    int r = p * 32 + (tid >> 3);
-    *reinterpret_cast<short8*>(lds + swz(r, cb)) = R.v[p];
+    int rp = p * 32 + wid * 8;
+    int gr = row0 + rp + rl;
+    if (gr >= rows_max) gr = rows_max - 1;
+    const void* g = src + (long)gr * ld + k0 + gcolb / 2;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (lds_buf + rp * ROWB),
+        16, 0, 0);
   }
 }
 
@@ -106,14 +102,9 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  StageRegs ra, rb;
-  if (mtail) stage_load<true>(A, K, bm0, M, 0, ra);
-  else       stage_load<false>(A, K, bm0, M, 0, ra);
-  if (ntail) stage_load<true>(B, K, bn0, N, 0, rb);
-  else       stage_load<false>(B, K, bn0, N, 0, rb);
-  stage_write(Ac, ra);
-  stage_write(Bc, rb);
-  __syncthreads();
+  glds_stage_tile(A, K, bm0, M, 0, Ac, wid, lane);
+  glds_stage_tile(B, K, bn0, N, 0, Bc, wid, lane);
+  __syncthreads();  // drains the in-flight LDS-DMA (vmcnt 0) + barrier
 
   const int nk = K / BK;
   // Fragment LDS byte offsets (row varies per lane: swizzle applies).
@@ -123,10 +114,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
 
   for (int t = 0; t < nk; ++t) {
     if (t + 1 < nk) {
-      if (mtail) stage_load<true>(A, K, bm0, M, (t + 1) * BK, ra);
-      else       stage_load<false>(A, K, bm0, M, (t + 1) * BK, ra);
-      if (ntail) stage_load<true>(B, K, bn0, N, (t + 1) * BK, rb);
-      else       stage_load<false>(B, K, bn0, N, (t + 1) * BK, rb);
+      // issue next tile's LDS-DMA before compute; the end-of-iter
+      // __syncthreads() drains it right when the next tile is needed
+      glds_stage_tile(A, K, bm0, M, (t + 1) * BK, An, wid, lane);
+      glds_stage_tile(B, K, bn0, N, (t + 1) * BK, Bn, wid, lane);
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -146,8 +137,6 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
               af[i], bf[j], acc[i][j], 0, 0, 0);
     }
     if (t + 1 < nk) {
-      stage_write(An, ra);
-      stage_write(Bn, rb);
       char* tmp = Ac; Ac = An; An = tmp;
       tmp = Bc; Bc = Bn; Bn = tmp;
     }
