@@ -70,6 +70,12 @@ def build_model(name: str, device="cpu",
         with torch.device(device):
             model = ctor()
         model = model.to(dtype=dtype)
+        # BN statistics/affine params stay fp32 (standard bf16-training
+        # practice; the fused BN kernels expect fp32 gamma/beta)
+        from ..ops.bn import FusedBN2d
+        for m in model.modules():
+            if isinstance(m, FusedBN2d):
+                m.float()
         if name.startswith("resnet"):
             # NHWC: MIOpen's native layout — avoids the per-conv
             # batched_transpose kernels that dominate NCHW profiles

@@ -1,9 +1,10 @@
 """ResNet-18/50 implemented in-package (no torchvision in this env).
 
-Convolutions/BatchNorm go through torch (MIOpen on ROCm); the final
-classifier is an MFMALinear so even CV jobs exercise the hand-written
-GEMM.  Two stem variants: CIFAR (3x3, no max-pool; config-2 jobs) and
-ImageNet (7x7/2 + max-pool; config-3 ResNet-50 jobs).
+Convolutions go through torch (MIOpen on ROCm, channels_last/NHWC);
+BatchNorm (+fused ReLU) and the classifier run on the gfx950 kernels
+(ops/bn.py FusedBN2d, ops/linear.py MFMALinear).  Two stem variants:
+CIFAR (3x3, no max-pool; config-2 jobs) and ImageNet (7x7/2 +
+max-pool; config-3 ResNet-50 jobs).
 """
 from __future__ import annotations
 
@@ -11,6 +12,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.bn import FusedBN2d
 from ..ops.linear import MFMALinear
 
 
@@ -24,18 +26,18 @@ class BasicBlock(nn.Module):
     def __init__(self, cin, cout, stride=1):
         super().__init__()
         self.conv1 = conv3x3(cin, cout, stride)
-        self.bn1 = nn.BatchNorm2d(cout)
+        self.bn1 = FusedBN2d(cout, relu=True)
         self.conv2 = conv3x3(cout, cout)
-        self.bn2 = nn.BatchNorm2d(cout)
+        self.bn2 = FusedBN2d(cout)
         self.down = None
         if stride != 1 or cin != cout:
             self.down = nn.Sequential(
                 nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(cout))
+                FusedBN2d(cout))
 
     def forward(self, x):
         idt = x if self.down is None else self.down(x)
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         return F.relu(out + idt)
 
@@ -47,21 +49,21 @@ class Bottleneck(nn.Module):
         super().__init__()
         cout = cmid * self.expansion
         self.conv1 = nn.Conv2d(cin, cmid, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(cmid)
+        self.bn1 = FusedBN2d(cmid, relu=True)
         self.conv2 = conv3x3(cmid, cmid, stride)
-        self.bn2 = nn.BatchNorm2d(cmid)
+        self.bn2 = FusedBN2d(cmid, relu=True)
         self.conv3 = nn.Conv2d(cmid, cout, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(cout)
+        self.bn3 = FusedBN2d(cout)
         self.down = None
         if stride != 1 or cin != cout:
             self.down = nn.Sequential(
                 nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(cout))
+                FusedBN2d(cout))
 
     def forward(self, x):
         idt = x if self.down is None else self.down(x)
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = F.relu(self.bn2(self.conv2(out)))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         return F.relu(out + idt)
 
@@ -71,12 +73,12 @@ class ResNet(nn.Module):
         super().__init__()
         self.cifar = cifar
         if cifar:
-            self.stem = nn.Sequential(conv3x3(3, 64), nn.BatchNorm2d(64),
-                                      nn.ReLU(inplace=True))
+            self.stem = nn.Sequential(conv3x3(3, 64),
+                                      FusedBN2d(64, relu=True))
         else:
             self.stem = nn.Sequential(
                 nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False),
-                nn.BatchNorm2d(64), nn.ReLU(inplace=True),
+                FusedBN2d(64, relu=True),
                 nn.MaxPool2d(3, stride=2, padding=1))
         self.cin = 64
         stages = []

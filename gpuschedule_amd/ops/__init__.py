@@ -2,6 +2,7 @@ from ._ext import get_ext, have_ext
 from .norms import FusedLayerNorm, FusedRMSNorm
 from .linear import MFMALinear
 from .optim import FusedAdamW, FusedSGD
+from .bn import FusedBN2d
 
 __all__ = ["get_ext", "have_ext", "FusedLayerNorm", "FusedRMSNorm",
-           "MFMALinear", "FusedAdamW", "FusedSGD"]
+           "MFMALinear", "FusedAdamW", "FusedSGD", "FusedBN2d"]

@@ -1,0 +1,85 @@
+"""Fused BatchNorm2d (+ReLU) module on the gfx950 kernels.
+
+GPU path expects channels_last activations ([N,H,W,C] memory) — the
+registry puts ResNet models in channels_last.  Parameters and running
+stats stay fp32 (standard practice for BN in bf16 training).  The ReLU
+(and its backward mask) is fused into the apply kernels.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ._ext import get_ext
+
+
+class _BNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, weight, bias, mean, rstd, eps, relu, training):
+        ext = get_ext(required=True)
+        if training:
+            y, mean, rstd = ext.bn_fwd(x2d, weight, bias, None, None,
+                                       eps, relu)
+        else:
+            y, _, _ = ext.bn_fwd(x2d, weight, bias, mean, rstd, eps,
+                                 relu)
+        ctx.save_for_backward(x2d, y if relu else None, mean, rstd,
+                              weight)
+        ctx.relu = relu
+        return y, mean, rstd
+
+    @staticmethod
+    def backward(ctx, dy, _dm, _dr):
+        ext = get_ext(required=True)
+        x2d, y, mean, rstd, weight = ctx.saved_tensors
+        dx, dbeta, dgamma = ext.bn_bwd(dy.contiguous(), x2d, y, mean,
+                                       rstd, weight, ctx.relu)
+        return dx, dgamma, dbeta, None, None, None, None, None
+
+
+class FusedBN2d(nn.Module):
+    def __init__(self, channels: int, eps: float = 1e-5,
+                 momentum: float = 0.1, relu: bool = False):
+        super().__init__()
+        self.channels = channels
+        self.eps = eps
+        self.momentum = momentum
+        self.relu = relu
+        self.weight = nn.Parameter(torch.ones(channels))
+        self.bias = nn.Parameter(torch.zeros(channels))
+        self.register_buffer("running_mean", torch.zeros(channels))
+        self.register_buffer("running_var", torch.ones(channels))
+
+    def _cpu(self, x):
+        y = F.batch_norm(x.float(), self.running_mean, self.running_var,
+                         self.weight, self.bias, self.training,
+                         self.momentum, self.eps).to(x.dtype)
+        return F.relu(y) if self.relu else y
+
+    def forward(self, x):
+        if not x.is_cuda:
+            return self._cpu(x)
+        N, C, H, W = x.shape
+        x2d = x.contiguous(memory_format=torch.channels_last) \
+               .permute(0, 2, 3, 1).reshape(-1, C)
+        if self.training:
+            mean = rstd = self.running_mean  # placeholders (unused)
+        else:
+            mean = self.running_mean
+            rstd = (self.running_var + self.eps).rsqrt()
+        y2d, mean, rstd = _BNFn.apply(x2d, self.weight, self.bias, mean,
+                                      rstd, self.eps, self.relu,
+                                      self.training)
+        if self.training:
+            with torch.no_grad():
+                n = x2d.shape[0]
+                var = (1.0 / (rstd * rstd) - self.eps) * (n / max(1, n - 1))
+                self.running_mean.mul_(1 - self.momentum) \
+                    .add_(mean, alpha=self.momentum)
+                self.running_var.mul_(1 - self.momentum) \
+                    .add_(var, alpha=self.momentum)
+        return y2d.view(N, H, W, C).permute(0, 3, 1, 2)
+
+    def extra_repr(self):
+        return f"{self.channels}, relu={self.relu}"

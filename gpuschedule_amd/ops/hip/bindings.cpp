@@ -18,6 +18,15 @@ void sgd_step(torch::Tensor p32, torch::Tensor p_bf16, torch::Tensor grad,
               torch::Tensor buf, double lr, double momentum, double wd);
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        bool causal);
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta,
+                                  c10::optional<torch::Tensor> use_mean,
+                                  c10::optional<torch::Tensor> use_rstd,
+                                  double eps, bool relu);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  c10::optional<torch::Tensor> y,
+                                  torch::Tensor mean, torch::Tensor rstd,
+                                  torch::Tensor gamma, bool relu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "fused LayerNorm/RMSNorm forward (bf16)");
@@ -27,4 +36,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
   m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax)");
+  m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");
+  m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+ReLU) backward, channels_last");
 }

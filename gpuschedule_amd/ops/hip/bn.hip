@@ -1,0 +1,285 @@
+// Fused BatchNorm2d (training) for gfx950, NHWC/channels_last layout.
+//
+// torch's native batch_norm kernels were 32% of the whole trace-replay
+// bench (collect_statistics + transform_input + backward as separate
+// unfused passes).  Here: channels are the INNERMOST dim
+// (channels_last), so every 16-B vector load covers 8 consecutive
+// channels; per-channel fp32 statistics accumulate in REGISTERS
+// (each thread owns a fixed 8-channel group because the grid stride
+// 2048 is a multiple of C for every power-of-2 C), merge once per
+// block through LDS, and hit global memory with one atomicAdd per
+// channel per block.  Apply kernels stage the per-channel scale/shift
+// tables in LDS and fuse the ReLU (forward) / ReLU mask (backward).
+//
+// Memory passes: fwd = read x (stats) + read x/write y (apply);
+// bwd = read dy,x (stats) + read dy,x / write dx (apply) — each pass
+// at the vectorized-streaming roofline.
+#include <torch/extension.h>
+#include <c10/cuda/CUDAStream.h>
+#include "common.h"
+
+namespace {
+
+constexpr int NT = 256;
+constexpr int PASS = NT * 8;  // elements per block pass (2048)
+
+// ---- per-channel sum/sumsq (or sum_dy/sum_dy_xhat) ----------------
+// REGPATH requires PASS % C == 0 (all ResNet widths); BWD computes
+// backward sums instead; RELU masks dy by (y > 0).
+template <bool BWD, bool RELU>
+__global__ __launch_bounds__(NT) void bn_stats_kernel(
+    const short* __restrict__ x, const short* __restrict__ dy,
+    const short* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ rstd, float* __restrict__ out_a,
+    float* __restrict__ out_b, long E, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* lds_a = (float*)smem;          // [C]
+  float* lds_b = lds_a + C;             // [C]
+  for (int i = threadIdx.x; i < 2 * C; i += NT) lds_a[i] = 0.f;
+  __syncthreads();
+
+  const int t8 = threadIdx.x * 8;
+  const int ch0 = t8 % C;               // fixed: PASS % C == 0
+  float acc_a[8], acc_b[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) { acc_a[j] = 0.f; acc_b[j] = 0.f; }
+  float mu[8], rs[8];
+  if (BWD) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[ch0 + j];
+      rs[j] = rstd[ch0 + j];
+    }
+  }
+
+  long base = (long)blockIdx.x * PASS + t8;
+  long stride = (long)gridDim.x * PASS;
+  for (long e = base; e + 7 < E; e += stride) {
+    short8 xv = *reinterpret_cast<const short8*>(x + e);
+    if (!BWD) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_to_f32(xv[j]);
+        acc_a[j] += f;
+        acc_b[j] += f * f;
+      }
+    } else {
+      short8 dv = *reinterpret_cast<const short8*>(dy + e);
+      short8 yv;
+      if (RELU) yv = *reinterpret_cast<const short8*>(y + e);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float d = bf16_to_f32(dv[j]);
+        if (RELU && bf16_to_f32(yv[j]) <= 0.f) d = 0.f;
+        float xh = (bf16_to_f32(xv[j]) - mu[j]) * rs[j];
+        acc_a[j] += d;
+        acc_b[j] += d * xh;
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(lds_a + ch0 + j, acc_a[j]);
+    atomicAdd(lds_b + ch0 + j, acc_b[j]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < C; i += NT) {
+    atomicAdd(out_a + i, lds_a[i]);
+    atomicAdd(out_b + i, lds_b[i]);
+  }
+}
+
+// ---- apply: y = x * scale[c] + shift[c] (+relu) -------------------
+template <bool RELU>
+__global__ __launch_bounds__(NT) void bn_apply_kernel(
+    const short* __restrict__ x, short* __restrict__ y,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    long E, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s_sc = (float*)smem;
+  float* s_sh = s_sc + C;
+  for (int i = threadIdx.x; i < C; i += NT) {
+    s_sc[i] = scale[i];
+    s_sh[i] = shift[i];
+  }
+  __syncthreads();
+  const int t8 = threadIdx.x * 8;
+  const int ch0 = t8 % C;
+  long base = (long)blockIdx.x * PASS + t8;
+  long stride = (long)gridDim.x * PASS;
+  for (long e = base; e + 7 < E; e += stride) {
+    short8 xv = *reinterpret_cast<const short8*>(x + e);
+    short8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf16_to_f32(xv[j]) * s_sc[ch0 + j] + s_sh[ch0 + j];
+      if (RELU) f = fmaxf(f, 0.f);
+      out[j] = f32_to_bf16(f);
+    }
+    *reinterpret_cast<short8*>(y + e) = out;
+  }
+}
+
+// ---- backward apply:
+// dx = g*rstd * (dy' - sum_dy/Nc - xhat * sum_dyxhat/Nc)
+template <bool RELU>
+__global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const short* __restrict__ y, short* __restrict__ dx,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ sum_dy,
+    const float* __restrict__ sum_dyxh, long E, int C, float inv_cnt) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // per-channel constants: k1 = g*rstd, c1 = sum_dy/Nc,
+  // c2 = sum_dyxh/Nc, mu, rs
+  float* s_k1 = (float*)smem;
+  float* s_c1 = s_k1 + C;
+  float* s_c2 = s_c1 + C;
+  float* s_mu = s_c2 + C;
+  float* s_rs = s_mu + C;
+  for (int i = threadIdx.x; i < C; i += NT) {
+    s_k1[i] = gamma[i] * rstd[i];
+    s_c1[i] = sum_dy[i] * inv_cnt;
+    s_c2[i] = sum_dyxh[i] * inv_cnt;
+    s_mu[i] = mean[i];
+    s_rs[i] = rstd[i];
+  }
+  __syncthreads();
+  const int t8 = threadIdx.x * 8;
+  const int ch0 = t8 % C;
+  long base = (long)blockIdx.x * PASS + t8;
+  long stride = (long)gridDim.x * PASS;
+  for (long e = base; e + 7 < E; e += stride) {
+    short8 dv = *reinterpret_cast<const short8*>(dy + e);
+    short8 xv = *reinterpret_cast<const short8*>(x + e);
+    short8 yv;
+    if (RELU) yv = *reinterpret_cast<const short8*>(y + e);
+    short8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = ch0 + j;
+      float d = bf16_to_f32(dv[j]);
+      if (RELU && bf16_to_f32(yv[j]) <= 0.f) d = 0.f;
+      float xh = (bf16_to_f32(xv[j]) - s_mu[c]) * s_rs[c];
+      out[j] = f32_to_bf16(s_k1[c] * (d - s_c1[c] - xh * s_c2[c]));
+    }
+    *reinterpret_cast<short8*>(dx + e) = out;
+  }
+}
+
+int grid_for(long E) {
+  long b = (E + PASS - 1) / PASS;
+  return (int)std::min<long>(b > 0 ? b : 1, 2048);
+}
+
+void check_bn(const torch::Tensor& x, int C) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(C % 8 == 0 && PASS % C == 0,
+              "fused BN needs C%8==0 and 2048%C==0, got C=", C);
+}
+
+}  // namespace
+
+// x viewed as [E/C rows][C]; returns (y, mean, rstd) in training mode
+// or (y,) with given stats.  gamma/beta/mean/rstd are fp32 [C].
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta,
+                                  c10::optional<torch::Tensor> use_mean,
+                                  c10::optional<torch::Tensor> use_rstd,
+                                  double eps, bool relu) {
+  const int C = x.size(-1);
+  check_bn(x, C);
+  const long E = x.numel();
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  torch::Tensor mean, rstd;
+  if (use_mean.has_value()) {
+    mean = *use_mean;
+    rstd = *use_rstd;
+  } else {
+    auto sum = torch::zeros({C}, fopt);
+    auto sumsq = torch::zeros({C}, fopt);
+    hipLaunchKernelGGL((bn_stats_kernel<false, false>),
+                       dim3(grid_for(E)), dim3(NT), 2 * C * 4,
+                       stream.stream(), (const short*)x.data_ptr(),
+                       nullptr, nullptr, nullptr, nullptr,
+                       sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                       E, C);
+    double inv = 1.0 / (double)(E / C);
+    mean = sum * inv;
+    rstd = (sumsq * inv - mean * mean + eps).rsqrt_();
+  }
+  auto scale = gamma * rstd;
+  auto shift = beta - mean * scale;
+  auto y = torch::empty_like(x);
+  if (relu) {
+    hipLaunchKernelGGL((bn_apply_kernel<true>), dim3(grid_for(E)),
+                       dim3(NT), 2 * C * 4, stream.stream(),
+                       (const short*)x.data_ptr(), (short*)y.data_ptr(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       E, C);
+  } else {
+    hipLaunchKernelGGL((bn_apply_kernel<false>), dim3(grid_for(E)),
+                       dim3(NT), 2 * C * 4, stream.stream(),
+                       (const short*)x.data_ptr(), (short*)y.data_ptr(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       E, C);
+  }
+  HIP_CHECK_LAST();
+  return {y, mean, rstd};
+}
+
+// returns (dx, sum_dy (=dbeta), sum_dyxh (=dgamma)); y needed iff relu
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  c10::optional<torch::Tensor> y,
+                                  torch::Tensor mean, torch::Tensor rstd,
+                                  torch::Tensor gamma, bool relu) {
+  const int C = x.size(-1);
+  check_bn(x, C);
+  const long E = x.numel();
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  auto sum_dy = torch::zeros({C}, fopt);
+  auto sum_dyxh = torch::zeros({C}, fopt);
+  const short* y_p = relu ? (const short*)y->data_ptr() : nullptr;
+  if (relu) {
+    hipLaunchKernelGGL((bn_stats_kernel<true, true>), dim3(grid_for(E)),
+                       dim3(NT), 2 * C * 4, stream.stream(),
+                       (const short*)x.data_ptr(),
+                       (const short*)dy.data_ptr(), y_p,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       sum_dy.data_ptr<float>(),
+                       sum_dyxh.data_ptr<float>(), E, C);
+  } else {
+    hipLaunchKernelGGL((bn_stats_kernel<true, false>), dim3(grid_for(E)),
+                       dim3(NT), 2 * C * 4, stream.stream(),
+                       (const short*)x.data_ptr(),
+                       (const short*)dy.data_ptr(), nullptr,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       sum_dy.data_ptr<float>(),
+                       sum_dyxh.data_ptr<float>(), E, C);
+  }
+  auto dx = torch::empty_like(x);
+  float inv_cnt = 1.0f / (float)(E / C);
+  if (relu) {
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<true>), dim3(grid_for(E)),
+                       dim3(NT), 5 * C * 4, stream.stream(),
+                       (const short*)dy.data_ptr(),
+                       (const short*)x.data_ptr(), y_p,
+                       (short*)dx.data_ptr(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       sum_dy.data_ptr<float>(),
+                       sum_dyxh.data_ptr<float>(), E, C, inv_cnt);
+  } else {
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<false>), dim3(grid_for(E)),
+                       dim3(NT), 5 * C * 4, stream.stream(),
+                       (const short*)dy.data_ptr(),
+                       (const short*)x.data_ptr(), nullptr,
+                       (short*)dx.data_ptr(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       sum_dy.data_ptr<float>(),
+                       sum_dyxh.data_ptr<float>(), E, C, inv_cnt);
+  }
+  HIP_CHECK_LAST();
+  return {dx, sum_dy, sum_dyxh};
+}

@@ -182,3 +182,56 @@ def test_resnet18_job_step_on_gpu():
         r.step(1)
         losses.append(r.last_loss)
     assert all(torch.isfinite(torch.tensor(losses)))
+
+
+# ---- fused BatchNorm ----------------------------------------------
+
+@pytest.mark.parametrize("relu", [False, True])
+@pytest.mark.parametrize("shape", [(16, 8, 8, 64), (8, 14, 14, 256),
+                                   (4, 7, 7, 2048)])
+def test_fused_bn_fwd_bwd(shape, relu):
+    torch.manual_seed(8)
+    N, H, W, C = shape
+    rows = N * H * W
+    x = torch.randn(rows, C, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(C, device="cuda").abs() + 0.5
+    b = torch.randn(C, device="cuda")
+    dy = torch.randn(rows, C, device="cuda", dtype=torch.bfloat16)
+
+    y, mean, rstd = ext().bn_fwd(x, g, b, None, None, 1e-5, relu)
+    xf = x.float().requires_grad_(True)
+    gf = g.clone().requires_grad_(True)
+    bf = b.clone().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(
+        xf.t().unsqueeze(0), None, None, gf, bf, True, 0.0, 1e-5)
+    ref = ref.squeeze(0).t()
+    if relu:
+        ref = torch.relu(ref)
+    assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 1e-2
+    ref.backward(dy.float())
+    dx, dbeta, dgamma = ext().bn_bwd(dy, x, y if relu else None, mean,
+                                     rstd, g, relu)
+    s = xf.grad.abs().max().item() + 1e-4
+    assert mx(dx, xf.grad) / s < 2e-2
+    s = gf.grad.abs().max().item() + 1e-3
+    assert mx(dgamma, gf.grad) / s < 2e-2
+    s = bf.grad.abs().max().item() + 1e-3
+    assert mx(dbeta, bf.grad) / s < 2e-2
+
+
+def test_fused_bn_module_matches_torch_bn():
+    from gpuschedule_amd.ops.bn import FusedBN2d
+    torch.manual_seed(9)
+    C = 64
+    fb = FusedBN2d(C, relu=False).cuda()
+    tb = torch.nn.BatchNorm2d(C).cuda()
+    x = torch.randn(8, C, 16, 16, device="cuda", dtype=torch.bfloat16)
+    y1 = fb(x)
+    y2 = tb(x.float())
+    assert (y1.float() - y2).abs().max().item() < 3e-2
+    assert torch.allclose(fb.running_mean, tb.running_mean, atol=1e-3)
+    assert torch.allclose(fb.running_var, tb.running_var, atol=1e-2)
+    fb.eval(); tb.eval()
+    y1 = fb(x)
+    y2 = tb(x.float())
+    assert (y1.float() - y2).abs().max().item() < 3e-2
