@@ -24,11 +24,12 @@ def run_jobs(specs, policy, tmp_path, policy_kw=None, ipt=2):
 
 
 def test_gandiva_timeslice_two_jobs_gpu(tmp_path):
-    # Two jobs share one GPU via in-HBM suspend/resume.
-    specs = [J("a", iters=6), J("b", iters=6, model="resnet18",
+    # Two jobs share one GPU via in-HBM suspend/resume.  Quantum much
+    # smaller than total job runtime so rotation must happen.
+    specs = [J("a", iters=30), J("b", iters=30, model="resnet18",
                batch=32)]
     log = run_jobs(specs, "gandiva", tmp_path,
-                   {"quantum": 0.5})
+                   {"quantum": 0.05}, ipt=1)
     s = log.summary()
     assert int(s["n_complete"]) == 2
     assert s["preemptions"] >= 1  # actually time-sliced
