@@ -55,6 +55,15 @@ class JobRunner:
             self.load_state(state)
         self.reducer = GradReducer(self.model, gang_group)
         self.last_loss = float("nan")
+        # hipGraph capture of the whole training step (fwd+bwd+fused
+        # optimizer): per-job static shapes make the step replayable;
+        # launch-bound models (small CNNs) gain the most.  Single-GPU
+        # jobs only (gang collectives stay out of capture).
+        self.use_graph = (str(device) != "cpu" and gang_world == 1)
+        self._graph = None
+        self._graph_warmup = 3  # eager iters before capture (lazy opt
+        #                         state init + allocator warmup)
+        self._static = None     # (x, y, loss)
 
     # ---- state movement -----------------------------------------
 
@@ -85,25 +94,68 @@ class JobRunner:
         y = y.to(dev, non_blocking=nb)
         return x, y
 
+    def _eager_iter(self, x, y):
+        self.optim.zero_grad(set_to_none=True)
+        loss = compute_loss(self.model_name, self.model, x, y)
+        loss.backward()
+        self.reducer.finalize()
+        self.optim.step()
+        return loss
+
+    def _try_capture(self, x, y) -> bool:
+        try:
+            sx = x.clone()
+            sy = y.clone()
+            torch.cuda.synchronize(self.device)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                sloss = self._eager_iter(sx, sy)
+            self._graph = g
+            self._static = (sx, sy, sloss)
+            return True
+        except Exception:
+            # capture-unsafe model (CPU syncs, dynamic shapes): stay
+            # eager for this job
+            self._graph = None
+            self.use_graph = False
+            return False
+
     def step(self, n_iters: int = 1) -> tuple:
         """Run up to n_iters training iterations; returns
         (iters_done, wall_seconds)."""
         t0 = time.perf_counter()
+        loss = None
         for _ in range(n_iters):
             x, y = self._batch(self.done_iters)
-            self.optim.zero_grad(set_to_none=False)
-            loss = compute_loss(self.model_name, self.model, x, y)
-            loss.backward()
-            self.reducer.finalize()
-            self.optim.step()
-            self.last_loss = float(loss.detach())
+            if self._graph is not None:
+                sx, sy, sloss = self._static
+                sx.copy_(x, non_blocking=True)
+                sy.copy_(y, non_blocking=True)
+                self._graph.replay()
+                loss = sloss
+            else:
+                if self.use_graph and self._graph_warmup <= 0:
+                    self._try_capture(x, y)
+                if self._graph is not None:
+                    sx, sy, sloss = self._static
+                    sx.copy_(x, non_blocking=True)
+                    sy.copy_(y, non_blocking=True)
+                    self._graph.replay()
+                    loss = sloss
+                else:
+                    loss = self._eager_iter(x, y)
+                    self._graph_warmup -= 1
             self.done_iters += 1
         if str(self.device) != "cpu":
             torch.cuda.synchronize(self.device)
+        if loss is not None:
+            self.last_loss = float(loss.detach())
         return n_iters, time.perf_counter() - t0
 
     def release(self) -> None:
         """Drop GPU state (eviction or completion)."""
         self.reducer.detach()
+        self._graph = None
+        self._static = None
         self.model = None
         self.optim = None

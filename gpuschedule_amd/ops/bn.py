@@ -16,14 +16,15 @@ from ._ext import get_ext
 
 class _BNFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2d, weight, bias, mean, rstd, eps, relu, training):
+    def forward(ctx, x2d, weight, bias, mean, rstd, rm, rv, momentum,
+                eps, relu, training):
         ext = get_ext(required=True)
         if training:
             y, mean, rstd = ext.bn_fwd(x2d, weight, bias, None, None,
-                                       eps, relu)
+                                       rm, rv, momentum, eps, relu)
         else:
-            y, _, _ = ext.bn_fwd(x2d, weight, bias, mean, rstd, eps,
-                                 relu)
+            y, _, _ = ext.bn_fwd(x2d, weight, bias, mean, rstd, None,
+                                 None, momentum, eps, relu)
         ctx.save_for_backward(x2d, y if relu else None, mean, rstd,
                               weight)
         ctx.relu = relu
@@ -35,7 +36,8 @@ class _BNFn(torch.autograd.Function):
         x2d, y, mean, rstd, weight = ctx.saved_tensors
         dx, dbeta, dgamma = ext.bn_bwd(dy.contiguous(), x2d, y, mean,
                                        rstd, weight, ctx.relu)
-        return dx, dgamma, dbeta, None, None, None, None, None
+        return (dx, dgamma, dbeta, None, None, None, None,
+                None, None, None, None)
 
 
 class FusedBN2d(nn.Module):
@@ -68,17 +70,10 @@ class FusedBN2d(nn.Module):
         else:
             mean = self.running_mean
             rstd = (self.running_var + self.eps).rsqrt()
-        y2d, mean, rstd = _BNFn.apply(x2d, self.weight, self.bias, mean,
-                                      rstd, self.eps, self.relu,
-                                      self.training)
-        if self.training:
-            with torch.no_grad():
-                n = x2d.shape[0]
-                var = (1.0 / (rstd * rstd) - self.eps) * (n / max(1, n - 1))
-                self.running_mean.mul_(1 - self.momentum) \
-                    .add_(mean, alpha=self.momentum)
-                self.running_var.mul_(1 - self.momentum) \
-                    .add_(var, alpha=self.momentum)
+        y2d, mean, rstd = _BNFn.apply(
+            x2d, self.weight, self.bias, mean, rstd, self.running_mean,
+            self.running_var, self.momentum, self.eps, self.relu,
+            self.training)
         return y2d.view(N, H, W, C).permute(0, 3, 1, 2)
 
     def extra_repr(self):

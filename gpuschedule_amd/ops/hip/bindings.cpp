@@ -22,7 +22,11 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   c10::optional<torch::Tensor> use_mean,
                                   c10::optional<torch::Tensor> use_rstd,
-                                  double eps, bool relu);
+                                  c10::optional<torch::Tensor> run_mean,
+                                  c10::optional<torch::Tensor> run_var,
+                                  double momentum, double eps, bool relu);
+void sgd_step_fp32(torch::Tensor p, torch::Tensor grad, torch::Tensor buf,
+                   double lr, double momentum, double wd);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
@@ -36,6 +40,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
   m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax)");
+  m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");
   m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");
   m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+ReLU) backward, channels_last");
 }

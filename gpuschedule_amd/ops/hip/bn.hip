@@ -167,6 +167,43 @@ __global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
   }
 }
 
+// finalize: sums -> mean/rstd/scale/shift (+running-stat update), one
+// kernel instead of ~10 C-length torch ops per BN layer per step.
+__global__ void bn_finalize_kernel(
+    const float* __restrict__ sums, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, float* __restrict__ scale,
+    float* __restrict__ shift, float* __restrict__ running_mean,
+    float* __restrict__ running_var, int C, float inv_cnt,
+    float unbias, float momentum, float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mu = sums[c] * inv_cnt;
+  float var = sums[C + c] * inv_cnt - mu * mu;
+  float rs = rsqrtf(var + eps);
+  mean_out[c] = mu;
+  rstd_out[c] = rs;
+  float sc = gamma[c] * rs;
+  scale[c] = sc;
+  shift[c] = beta[c] - mu * sc;
+  if (running_mean) {
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
+    running_var[c] = (1.f - momentum) * running_var[c]
+        + momentum * var * unbias;
+  }
+}
+
+__global__ void bn_scale_shift_kernel(
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ scale, float* __restrict__ shift, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float sc = gamma[c] * rstd[c];
+  scale[c] = sc;
+  shift[c] = beta[c] - mean[c] * sc;
+}
+
 int grid_for(long E) {
   long b = (E + PASS - 1) / PASS;
   return (int)std::min<long>(b > 0 ? b : 1, 2048);
@@ -186,31 +223,50 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   c10::optional<torch::Tensor> use_mean,
                                   c10::optional<torch::Tensor> use_rstd,
-                                  double eps, bool relu) {
+                                  c10::optional<torch::Tensor> run_mean,
+                                  c10::optional<torch::Tensor> run_var,
+                                  double momentum, double eps, bool relu) {
   const int C = x.size(-1);
   check_bn(x, C);
   const long E = x.numel();
   auto fopt = x.options().dtype(torch::kFloat32);
   auto stream = c10::cuda::getCurrentCUDAStream();
+  const int cgrid = (C + NT - 1) / NT;
   torch::Tensor mean, rstd;
+  auto scale = torch::empty({C}, fopt);
+  auto shift = torch::empty({C}, fopt);
   if (use_mean.has_value()) {
     mean = *use_mean;
     rstd = *use_rstd;
+    hipLaunchKernelGGL(bn_scale_shift_kernel, dim3(cgrid), dim3(NT), 0,
+                       stream.stream(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), scale.data_ptr<float>(),
+                       shift.data_ptr<float>(), C);
   } else {
-    auto sum = torch::zeros({C}, fopt);
-    auto sumsq = torch::zeros({C}, fopt);
+    auto sums = torch::zeros({2 * C}, fopt);
     hipLaunchKernelGGL((bn_stats_kernel<false, false>),
                        dim3(grid_for(E)), dim3(NT), 2 * C * 4,
                        stream.stream(), (const short*)x.data_ptr(),
                        nullptr, nullptr, nullptr, nullptr,
-                       sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-                       E, C);
-    double inv = 1.0 / (double)(E / C);
-    mean = sum * inv;
-    rstd = (sumsq * inv - mean * mean + eps).rsqrt_();
+                       sums.data_ptr<float>(),
+                       sums.data_ptr<float>() + C, E, C);
+    long cnt = E / C;
+    mean = torch::empty({C}, fopt);
+    rstd = torch::empty({C}, fopt);
+    float* rm = run_mean.has_value() ? run_mean->data_ptr<float>()
+                                     : nullptr;
+    float* rv = run_var.has_value() ? run_var->data_ptr<float>()
+                                    : nullptr;
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(cgrid), dim3(NT), 0,
+                       stream.stream(), sums.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       rm, rv, C, 1.0f / (float)cnt,
+                       (float)cnt / (float)std::max<long>(cnt - 1, 1),
+                       (float)momentum, (float)eps);
   }
-  auto scale = gamma * rstd;
-  auto shift = beta - mean * scale;
   auto y = torch::empty_like(x);
   if (relu) {
     hipLaunchKernelGGL((bn_apply_kernel<true>), dim3(grid_for(E)),
@@ -239,8 +295,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   const long E = x.numel();
   auto fopt = x.options().dtype(torch::kFloat32);
   auto stream = c10::cuda::getCurrentCUDAStream();
-  auto sum_dy = torch::zeros({C}, fopt);
-  auto sum_dyxh = torch::zeros({C}, fopt);
+  auto sums = torch::zeros({2 * C}, fopt);
+  auto sum_dy = sums.narrow(0, 0, C);
+  auto sum_dyxh = sums.narrow(0, C, C);
   const short* y_p = relu ? (const short*)y->data_ptr() : nullptr;
   if (relu) {
     hipLaunchKernelGGL((bn_stats_kernel<true, true>), dim3(grid_for(E)),

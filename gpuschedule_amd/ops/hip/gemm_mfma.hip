@@ -165,6 +165,147 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
   }
 }
 
+// ==== 256x256-tile phase-interleaved GEMM =========================
+// 8 waves (512 threads) in a 2(M) x 4(N) wave grid; per-wave output
+// 128x64 = 8x4 fragments; BK=64; both operands' K-tiles double-
+// buffered in 128 KiB LDS.  Each K-tile is processed as 4 phases
+// (64-row x 32-col C-quadrants, 16 MFMAs each); every phase also
+// issues the LDS-DMA for one half-tile of the NEXT K-tile, so 8 glds
+// spread across the tile's compute and the end-of-tile barrier waits
+// on loads that had ~4 MFMA clusters to land.  s_setprio(1) around
+// each MFMA cluster lets the CU scheduler favor compute waves
+// (phase-split creates the wave role diversity it needs).
+constexpr int BM2 = 256, BN2 = 256, BK2 = 64;
+constexpr int NT2 = 512;
+
+__device__ __forceinline__ void glds_stage_half(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int k0, char* lds_img, int half, int wid, int lane) {
+  const int rl = lane >> 3;
+  const int gcolb = ((lane & 7) * 16) ^ (rl << 4);
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int rp = half * 128 + p * 64 + wid * 8;
+    int gr = row0 + rp + rl;
+    if (gr >= rows_max) gr = rows_max - 1;
+    const void* g = src + (long)gr * ld + k0 + gcolb / 2;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (lds_img + rp * ROWB),
+        16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+  char* Ac = lds;
+  char* Bc = lds + BM2 * ROWB;
+  char* An = lds + 2 * BM2 * ROWB;
+  char* Bn = lds + 3 * BM2 * ROWB;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    glds_stage_half(A, K, bm0, M, 0, Ac, h, wid, lane);
+    glds_stage_half(B, K, bn0, N, 0, Bc, h, wid, lane);
+  }
+  __syncthreads();
+
+  const int nk = K / BK2;
+  const int arow = (lane & 15);
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK2;
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int jr = ph >> 1, jc = ph & 1;
+      bf16x8 af[4][2], bf[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int fi = 0; fi < 4; ++fi)
+          af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+              Ac + swz(wr + jr * 64 + fi * 16 + arow, kk * 64 + kb));
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          bf[fj][kk] = *reinterpret_cast<const bf16x8*>(
+              Bc + swz(wc + jc * 32 + fj * 16 + arow, kk * 64 + kb));
+      }
+      if (pre) {
+        // one half-tile of the next K-tile per phase: A0 A1 B0 B1
+        if (ph < 2) glds_stage_half(A, K, bm0, M, k1, An, ph, wid, lane);
+        else glds_stage_half(B, K, bn0, N, k1, Bn, ph - 2, wid, lane);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+          for (int fj = 0; fj < 2; ++fj)
+            acc[jr * 4 + fi][jc * 2 + fj] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[fi][kk], bf[fj][kk],
+                    acc[jr * 4 + fi][jc * 2 + fj], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (pre) {
+      char* tmp = Ac; Ac = An; An = tmp;
+      tmp = Bc; Bc = Bn; Bn = tmp;
+    }
+    __syncthreads();  // drains this tile's prefetch DMA + barrier
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  const int crow0 = bm0 + wr + (lane >> 4) * 4;
+  const int ccol0 = bn0 + wc + (lane & 15);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = ccol0 + j * 16;
+      if (ntail && col >= N) continue;
+      float bv = bias ? bf16_to_f32(bias[col]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow0 + i * 16 + r;
+        if (mtail && row >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (relu) v = fmaxf(v, 0.f);
+        C[(long)row * N + col] = f32_to_bf16(v);
+      }
+    }
+  }
+}
+
 // ---- bf16 2-D transpose (LDS-tiled, vectorized loads) --------------
 // grid.z = batch index (stride R*C per matrix) for batched use.
 constexpr int TT = 64;  // tile
@@ -234,12 +375,20 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                 bias->numel() == N);
     bias_p = (const short*)bias->data_ptr();
   }
-  int grid = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
   auto stream = c10::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(gemm_nt_kernel, dim3(grid), dim3(NTHREADS), 0,
-                     stream.stream(), (const short*)a.data_ptr(),
-                     (const short*)b.data_ptr(), (short*)c.data_ptr(),
-                     bias_p, M, N, K, relu ? 1 : 0);
+  if (M >= 256 && N >= 256) {
+    int grid = ((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+    hipLaunchKernelGGL(gemm_nt256_kernel, dim3(grid), dim3(NT2), 0,
+                       stream.stream(), (const short*)a.data_ptr(),
+                       (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                       bias_p, M, N, K, relu ? 1 : 0);
+  } else {
+    int grid = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+    hipLaunchKernelGGL(gemm_nt_kernel, dim3(grid), dim3(NTHREADS), 0,
+                       stream.stream(), (const short*)a.data_ptr(),
+                       (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                       bias_p, M, N, K, relu ? 1 : 0);
+  }
   HIP_CHECK_LAST();
   return c;
 }

@@ -75,12 +75,40 @@ __global__ __launch_bounds__(NT) void sgd_kernel(
   }
 }
 
+// fp32-native SGD (BatchNorm affine params train in fp32)
+__global__ __launch_bounds__(NT) void sgd_fp32_kernel(
+    float* __restrict__ p, const float* __restrict__ grad,
+    float* __restrict__ buf, long n, float lr, float momentum,
+    float wd) {
+  long i = blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float g = grad[i] + wd * p[i];
+    float b = momentum * buf[i] + g;
+    buf[i] = b;
+    p[i] -= lr * b;
+  }
+}
+
 int grid_for(long n) {
   long blocks = (n / 4 + NT - 1) / NT;
   return (int)std::min<long>(blocks > 0 ? blocks : 1, 2048);
 }
 
 }  // namespace
+
+void sgd_step_fp32(torch::Tensor p, torch::Tensor grad, torch::Tensor buf,
+                   double lr, double momentum, double wd) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
+  long n = p.numel();
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(sgd_fp32_kernel,
+                     dim3((int)std::min<long>((n + NT - 1) / NT, 2048)),
+                     dim3(NT), 0, stream.stream(), p.data_ptr<float>(),
+                     grad.data_ptr<float>(), buf.data_ptr<float>(), n,
+                     (float)lr, (float)momentum, (float)wd);
+  HIP_CHECK_LAST();
+}
 
 void adamw_step(torch::Tensor p_bf16, torch::Tensor p32, torch::Tensor grad,
                 torch::Tensor m, torch::Tensor v, double lr, double beta1,

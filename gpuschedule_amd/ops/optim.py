@@ -100,6 +100,10 @@ class FusedSGD(_MasterStateMixin, torch.optim.Optimizer):
                     ext = get_ext(required=True)
                     ext.sgd_step(state["p32"], p.data, p.grad.contiguous(),
                                  state["buf"], lr, mom, wd)
+                elif p.is_cuda and p.dtype == torch.float32:
+                    ext = get_ext(required=True)
+                    ext.sgd_step_fp32(p.data, p.grad.contiguous(),
+                                      state["buf"], lr, mom, wd)
                 else:
                     g = p.grad.float() + wd * state["p32"]
                     state["buf"].mul_(mom).add_(g)

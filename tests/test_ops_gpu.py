@@ -198,7 +198,8 @@ def test_fused_bn_fwd_bwd(shape, relu):
     b = torch.randn(C, device="cuda")
     dy = torch.randn(rows, C, device="cuda", dtype=torch.bfloat16)
 
-    y, mean, rstd = ext().bn_fwd(x, g, b, None, None, 1e-5, relu)
+    y, mean, rstd = ext().bn_fwd(x, g, b, None, None, None, None,
+                                 0.1, 1e-5, relu)
     xf = x.float().requires_grad_(True)
     gf = g.clone().requires_grad_(True)
     bf = b.clone().requires_grad_(True)
