@@ -9,6 +9,7 @@ differ and a migrated/restored job replays a deterministic stream.
 """
 from __future__ import annotations
 
+import os
 import time
 import zlib
 from typing import Optional
@@ -59,11 +60,19 @@ class JobRunner:
         # optimizer): per-job static shapes make the step replayable;
         # launch-bound models (small CNNs) gain the most.  Single-GPU
         # jobs only (gang collectives stay out of capture).
-        self.use_graph = (str(device) != "cpu" and gang_world == 1)
+        self.use_graph = (str(device) != "cpu" and gang_world == 1
+                          and not os.environ.get("GSAMD_NO_GRAPHS"))
         self._graph = None
         self._graph_warmup = 3  # eager iters before capture (lazy opt
         #                         state init + allocator warmup)
         self._static = None     # (x, y, loss)
+        # Per-job HIP stream: all of this job's work (including eager
+        # warmup) runs on it, so autograd's AccumulateGrad nodes bind
+        # to the SAME stream the hipGraph captures on (the default-
+        # stream mismatch breaks capture), and concurrently-resident
+        # jobs' kernels can overlap.
+        self.stream = (torch.cuda.Stream(device)
+                       if str(device) != "cpu" else None)
 
     # ---- state movement -----------------------------------------
 
@@ -106,9 +115,9 @@ class JobRunner:
         try:
             sx = x.clone()
             sy = y.clone()
-            torch.cuda.synchronize(self.device)
+            self.stream.synchronize()
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            with torch.cuda.graph(g, stream=self.stream):
                 sloss = self._eager_iter(sx, sy)
             self._graph = g
             self._static = (sx, sy, sloss)
@@ -120,34 +129,38 @@ class JobRunner:
             self.use_graph = False
             return False
 
+    def _one_iter(self, x, y):
+        if self._graph is None and self.use_graph \
+                and self._graph_warmup <= 0:
+            self._try_capture(x, y)
+        if self._graph is not None:
+            sx, sy, sloss = self._static
+            sx.copy_(x, non_blocking=True)
+            sy.copy_(y, non_blocking=True)
+            self._graph.replay()
+            return sloss
+        self._graph_warmup -= 1
+        return self._eager_iter(x, y)
+
     def step(self, n_iters: int = 1) -> tuple:
         """Run up to n_iters training iterations; returns
         (iters_done, wall_seconds)."""
         t0 = time.perf_counter()
         loss = None
-        for _ in range(n_iters):
-            x, y = self._batch(self.done_iters)
-            if self._graph is not None:
-                sx, sy, sloss = self._static
-                sx.copy_(x, non_blocking=True)
-                sy.copy_(y, non_blocking=True)
-                self._graph.replay()
-                loss = sloss
-            else:
-                if self.use_graph and self._graph_warmup <= 0:
-                    self._try_capture(x, y)
-                if self._graph is not None:
-                    sx, sy, sloss = self._static
-                    sx.copy_(x, non_blocking=True)
-                    sy.copy_(y, non_blocking=True)
-                    self._graph.replay()
-                    loss = sloss
-                else:
-                    loss = self._eager_iter(x, y)
-                    self._graph_warmup -= 1
-            self.done_iters += 1
-        if str(self.device) != "cpu":
-            torch.cuda.synchronize(self.device)
+        if self.stream is None:
+            for _ in range(n_iters):
+                x, y = self._batch(self.done_iters)
+                loss = self._one_iter(x, y)
+                self.done_iters += 1
+        else:
+            self.stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self.stream):
+                for _ in range(n_iters):
+                    x, y = self._batch(self.done_iters)
+                    loss = None  # drop autograd refs before capture
+                    loss = self._one_iter(x, y)
+                    self.done_iters += 1
+            self.stream.synchronize()
         if loss is not None:
             self.last_loss = float(loss.detach())
         return n_iters, time.perf_counter() - t0

@@ -376,8 +376,12 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
     bias_p = (const short*)bias->data_ptr();
   }
   auto stream = c10::cuda::getCurrentCUDAStream();
-  if (M >= 256 && N >= 256) {
-    int grid = ((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+  // 256-tile kernel runs 1 block/CU (128 KiB LDS): only profitable
+  // when its grid fills most of the 256 CUs; otherwise the 128-tile
+  // kernel's 4x block count wins.
+  long blocks256 = (long)((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+  if (M >= 256 && N >= 256 && blocks256 >= 200) {
+    int grid = (int)blocks256;
     hipLaunchKernelGGL(gemm_nt256_kernel, dim3(grid), dim3(NT2), 0,
                        stream.stream(), (const short*)a.data_ptr(),
                        (const short*)b.data_ptr(), (short*)c.data_ptr(),
