@@ -14,6 +14,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.attention import flash_attention
+from ..ops.ce import fused_cross_entropy
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedLayerNorm
 
@@ -79,5 +80,4 @@ class BertMLM(nn.Module):
 
     def loss(self, tokens, labels):
         logits = self(tokens)
-        return F.cross_entropy(logits.view(-1, self.cfg.vocab_size).float(),
-                               labels.view(-1))
+        return fused_cross_entropy(logits, labels)

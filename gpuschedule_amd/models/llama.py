@@ -16,6 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.attention import flash_attention
+from ..ops.ce import fused_cross_entropy
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedRMSNorm
 
@@ -125,5 +126,4 @@ class LlamaLM(nn.Module):
 
     def loss(self, tokens, labels):
         logits = self(tokens)
-        return F.cross_entropy(
-            logits.view(-1, self.cfg.vocab_size).float(), labels.view(-1))
+        return fused_cross_entropy(logits, labels)

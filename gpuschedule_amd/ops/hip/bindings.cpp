@@ -27,6 +27,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   double momentum, double eps, bool relu);
 void sgd_step_fp32(torch::Tensor p, torch::Tensor grad, torch::Tensor buf,
                    double lr, double momentum, double wd);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits,
+                                  torch::Tensor labels);
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
+                     torch::Tensor lse, torch::Tensor gscale);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
@@ -40,6 +44,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
   m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax)");
+  m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");
+  m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward");
   m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");
   m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");
   m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+ReLU) backward, channels_last");

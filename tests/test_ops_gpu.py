@@ -236,3 +236,27 @@ def test_fused_bn_module_matches_torch_bn():
     y1 = fb(x)
     y2 = tb(x.float())
     assert (y1.float() - y2).abs().max().item() < 3e-2
+
+
+# ---- fused cross-entropy ------------------------------------------
+
+@pytest.mark.parametrize("shape", [(512, 1024), (4096, 30528),
+                                   (128, 1000)])
+def test_fused_ce(shape):
+    torch.manual_seed(10)
+    T, V = shape
+    logits = torch.randn(T, V, device="cuda", dtype=torch.bfloat16)
+    labels = torch.randint(0, V, (T,), device="cuda")
+    loss, lse = ext().ce_fwd(logits, labels)
+    ref = torch.nn.functional.cross_entropy(
+        logits.float().requires_grad_(False), labels)
+    assert abs(loss.item() - ref.item()) / ref.item() < 2e-3
+
+    from gpuschedule_amd.ops.ce import fused_cross_entropy
+    lg = logits.clone().requires_grad_(True)
+    out = fused_cross_entropy(lg, labels)
+    out.backward()
+    lf = logits.float().requires_grad_(True)
+    torch.nn.functional.cross_entropy(lf, labels).backward()
+    scale = lf.grad.abs().max().item() + 1e-9
+    assert (lg.grad.float() - lf.grad).abs().max().item() / scale < 2e-2
