@@ -31,30 +31,19 @@ class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal):
         ext = get_ext(required=True)
-        o = ext.attn_fwd(q, k, v, causal)
-        ctx.save_for_backward(q, k, v)
+        o, lse = ext.attn_fwd(q, k, v, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
         return o
 
     @staticmethod
     def backward(ctx, do):
-        # Recompute-backward through the bf16 composite (hipBLASLt
-        # matmuls); exact softmax, no saved S x S activation.
-        q, k, v = ctx.saved_tensors
-        with torch.enable_grad():
-            q2 = q.detach().requires_grad_(True)
-            k2 = k.detach().requires_grad_(True)
-            v2 = v.detach().requires_grad_(True)
-            scale = 1.0 / math.sqrt(q.shape[-1])
-            s = torch.matmul(q2, k2.transpose(-1, -2)) * scale
-            if ctx.causal:
-                S = q.shape[-2]
-                mask = torch.ones(S, S, dtype=torch.bool,
-                                  device=q.device).triu(1)
-                s = s.masked_fill(mask, float("-inf"))
-            p = torch.softmax(s.float(), dim=-1).to(q.dtype)
-            o = torch.matmul(p, v2)
-            dq, dk, dv = torch.autograd.grad(o, (q2, k2, v2), do)
+        # hand-written FA2-style backward (ops/hip/attention.hip):
+        # no S x S activation ever materialized.
+        ext = get_ext(required=True)
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = ext.attn_bwd(q, k, v, o, do.contiguous(), lse,
+                                  ctx.causal)
         return dq, dk, dv, None
 
 

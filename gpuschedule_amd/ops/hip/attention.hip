@@ -52,7 +52,7 @@ template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ Vt, short* __restrict__ O,
-    int B, int H, int S, float scale) {
+    float* __restrict__ LSE, int B, int H, int S, float scale) {
   constexpr int DF = D / 16;       // PV col fragments
   constexpr int KC = D / 32;       // QK^T k-steps over D
   constexpr int ROWB = D * 2;      // K tile LDS row bytes
@@ -241,7 +241,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: O / l ------------------------------------------
+  // ---- epilogue: O / l, plus per-row logsumexp for backward -----
   const int orow0 = q0 + (lane >> 4) * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -252,6 +252,378 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
 #pragma unroll
     for (int f = 0; f < DF; ++f)
       orow[f * 16 + (lane & 15)] = f32_to_bf16(o_acc[f][r] * inv);
+    if (LSE && (lane & 15) == 0)
+      LSE[(long)(b * H + h) * S + row] =
+          (l[r] > 0.f) ? m[r] + __logf(l[r]) : -1e30f;
+  }
+}
+
+// ==================== backward =====================================
+// FA2-style two-kernel backward (no atomics):
+//   bwd_kv: grid over KV tiles, accumulates dK/dV looping q-tiles;
+//   bwd_q:  grid over Q tiles, accumulates dQ looping kv-tiles.
+// All matmuls are the SAME validated NT MFMA pattern; operands whose
+// reduction dim is the row dim (Q/K/dO in the dK/dV/dQ products) come
+// from batched pre-transposed copies (Qt/Kt/dOt, one fast pass each),
+// so every B-fragment reads 8 bf16 contiguous along the reduction.
+// P / dS convert C-layout -> A-fragment layout through wave-private
+// swizzled LDS, as in forward.  Definitions (scale folded into the
+// staged K/Q fragments so exp uses the saved scaled-score lse):
+//   P  = exp(S - lse[q]);  dP = dO . V^T;  Drow = rowsum(dO*O)
+//   dS = P * (dP - Drow[q]) * scale
+//   dV = P^T dO;  dK = dS^T Q;  dQ = dS K
+
+// generic glds tile stage: ROWS x ROWBYTES image, 4 waves, source-
+// swizzled so the linear LDS image is the swizzled image.  Row AND
+// column clamped (clamped lanes load valid-but-wrong data; consumers
+// mask those rows/cols to zero before they reach any stored output).
+template <int ROWS, int ROWBYTES>
+__device__ __forceinline__ void glds_tile(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int col0, int cols_max, char* img, int wid, int lane) {
+  constexpr int RPI = 1024 / ROWBYTES;     // rows per instruction
+  constexpr int CPR = ROWBYTES / 16;       // 16B chunks per row
+  constexpr int NI = (ROWS / 4) / RPI;     // instructions per wave
+  const int rl = lane / CPR;
+  const int cb = (lane % CPR) * 16;
+#pragma unroll
+  for (int p = 0; p < NI; ++p) {
+    int rp = wid * (ROWS / 4) + p * RPI + rl;
+    int gr = row0 + rp;
+    if (gr >= rows_max) gr = rows_max - 1;
+    int scol = (cb ^ ((rp & 7) << 4)) / 2 + col0;
+    if (scol > cols_max - 8) scol = cols_max - 8;
+    const void* g = src + (long)gr * ld + scol;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (img + (wid * (ROWS / 4) + p * RPI) * ROWBYTES),
+        16, 0, 0);
+  }
+}
+
+// Drow = rowsum(dO * O): one 64-lane wave per row.
+__global__ __launch_bounds__(256) void attn_drow_kernel(
+    const short* __restrict__ dO, const short* __restrict__ O,
+    float* __restrict__ Drow, long n_rows, int D) {
+  long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= n_rows) return;
+  const int lane = threadIdx.x & 63;
+  float s = 0.f;
+  for (int i = lane * 2; i + 1 < D; i += 128) {
+    s += bf16_to_f32(dO[row * D + i]) * bf16_to_f32(O[row * D + i]);
+    s += bf16_to_f32(dO[row * D + i + 1]) *
+         bf16_to_f32(O[row * D + i + 1]);
+  }
+  s = wave_reduce_sum(s);
+  if (lane == 0) Drow[row] = s;
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const short* __restrict__ Qt, const short* __restrict__ dOt,
+    const float* __restrict__ LSE, const float* __restrict__ DR,
+    short* __restrict__ dK, short* __restrict__ dV,
+    int B, int H, int S, float scale) {
+  constexpr int DF = D / 16;
+  constexpr int KC = D / 32;
+  constexpr int RB = D * 2;
+  __shared__ __attribute__((aligned(16))) char lds[
+      2 * KVB * D * 2 +      // Q, dO  [64][D]
+      2 * D * KVB * 2 +      // Qt, dOt [D][64]
+      4 * 2 * 16 * KVB * 2 + // per-wave P^T and dS^T  [16][64]
+      2 * KVB * 4];          // lse, Drow (f32[64])
+  char* q_lds = lds;
+  char* do_lds = q_lds + KVB * D * 2;
+  char* qt_lds = do_lds + KVB * D * 2;
+  char* dot_lds = qt_lds + D * KVB * 2;
+  char* p_base = dot_lds + D * KVB * 2;
+  float* lse_lds = (float*)(p_base + 4 * 2 * 16 * KVB * 2);
+  float* dr_lds = lse_lds + KVB;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  char* my_p = p_base + wid * 16 * KVB * 2;
+  char* my_ds = p_base + (4 + wid) * 16 * KVB * 2;
+
+  const int n_kvt = (S + KVB - 1) / KVB;
+  const int bh = blockIdx.x / n_kvt;
+  const int kvt = blockIdx.x % n_kvt;
+  const int b = bh / H, h = bh % H;
+  const int kv0 = kvt * KVB;
+  const int krow = kv0 + wid * 16 + (lane & 15);
+
+  const long off = ((long)(b * H + h) * S) * D;
+  const long offT = ((long)(b * H + h) * D) * S;
+  const long roff = (long)(b * H + h) * S;
+
+  // K (scaled) and V fragments in registers
+  bf16x8 kreg[KC], vreg[KC];
+  {
+    int row = krow < S ? krow : S - 1;
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      int d0 = c * 32 + (lane >> 4) * 8;
+      bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
+          K + off + (long)row * D + d0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        kv8[j] = (__bf16)((float)kv8[j] * scale);
+      kreg[c] = kv8;
+      vreg[c] = *reinterpret_cast<const bf16x8*>(
+          V + off + (long)row * D + d0);
+    }
+  }
+
+  float4v dv_acc[DF], dk_acc[DF];
+#pragma unroll
+  for (int f = 0; f < DF; ++f) {
+    dv_acc[f] = {0.f, 0.f, 0.f, 0.f};
+    dk_acc[f] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int qt_start = CAUSAL ? kv0 / 64 : 0;
+  const int n_qt = (S + 63) / 64;
+  for (int qt = qt_start; qt < n_qt; ++qt) {
+    const int q0 = qt * 64;
+    glds_tile<KVB, RB>(Q, D, q0, S, 0, D, q_lds, wid, lane);
+    glds_tile<KVB, RB>(dO, D, q0, S, 0, D, do_lds, wid, lane);
+    glds_tile<D, KVB * 2>(Qt, S, 0, D, q0, S, qt_lds, wid, lane);
+    glds_tile<D, KVB * 2>(dOt, S, 0, D, q0, S, dot_lds, wid, lane);
+    for (int i = tid; i < KVB; i += NTHREADS) {
+      int qr = q0 + i;
+      lse_lds[i] = qr < S ? LSE[roff + qr] : 3e30f;
+      dr_lds[i] = qr < S ? DR[roff + qr] : 0.f;
+    }
+    __syncthreads();
+
+    // S^T[16k][64q] and dP^T[16k][64q]
+    float4v st[4], dpt[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      st[f] = {0.f, 0.f, 0.f, 0.f};
+      dpt[f] = {0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int kk = 0; kk < KC; ++kk) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+            q_lds + swz(f * 16 + (lane & 15),
+                        kk * 64 + (lane >> 4) * 16, RB));
+        st[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            kreg[kk], qf, st[f], 0, 0, 0);
+        bf16x8 dof = *reinterpret_cast<const bf16x8*>(
+            do_lds + swz(f * 16 + (lane & 15),
+                         kk * 64 + (lane >> 4) * 16, RB));
+        dpt[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            vreg[kk], dof, dpt[f], 0, 0, 0);
+      }
+    }
+    // P^T, dS^T -> wave-private LDS
+    const int krow0 = kv0 + wid * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int qcol = q0 + f * 16 + (lane & 15);
+      float lse_q = lse_lds[f * 16 + (lane & 15)];
+      float dr_q = dr_lds[f * 16 + (lane & 15)];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        bool dead = qcol >= S || lse_q > 1e30f;
+        if (CAUSAL) dead |= qcol < krow0 + r;
+        float p = dead ? 0.f : __expf(st[f][r] - lse_q);
+        float ds = p * (dpt[f][r] - dr_q) * scale;
+        int row = (lane >> 4) * 4 + r;
+        int cb = (f * 16 + (lane & 15)) * 2;
+        *reinterpret_cast<short*>(
+            my_p + swz(row, cb, KVB * 2)) = f32_to_bf16(p);
+        *reinterpret_cast<short*>(
+            my_ds + swz(row, cb, KVB * 2)) = f32_to_bf16(ds);
+      }
+    }
+    // dV += P^T dO (B from dOt image), dK += dS^T Q (B from Qt image)
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 pf = *reinterpret_cast<const bf16x8*>(
+          my_p + swz(lane & 15, kk * 64 + (lane >> 4) * 16, KVB * 2));
+      bf16x8 dsf = *reinterpret_cast<const bf16x8*>(
+          my_ds + swz(lane & 15, kk * 64 + (lane >> 4) * 16, KVB * 2));
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        bf16x8 dotf = *reinterpret_cast<const bf16x8*>(
+            dot_lds + swz(f * 16 + (lane & 15),
+                          kk * 64 + (lane >> 4) * 16, KVB * 2));
+        dv_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pf, dotf, dv_acc[f], 0, 0, 0);
+        bf16x8 qtf = *reinterpret_cast<const bf16x8*>(
+            qt_lds + swz(f * 16 + (lane & 15),
+                         kk * 64 + (lane >> 4) * 16, KVB * 2));
+        dk_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf, qtf, dk_acc[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int orow0 = kv0 + wid * 16 + (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = orow0 + r;
+    if (row >= S) continue;
+    short* dkr = dK + off + (long)row * D;
+    short* dvr = dV + off + (long)row * D;
+#pragma unroll
+    for (int f = 0; f < DF; ++f) {
+      dkr[f * 16 + (lane & 15)] = f32_to_bf16(dk_acc[f][r]);
+      dvr[f * 16 + (lane & 15)] = f32_to_bf16(dv_acc[f][r]);
+    }
+  }
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const short* __restrict__ Kt, const float* __restrict__ LSE,
+    const float* __restrict__ DR, short* __restrict__ dQ,
+    int B, int H, int S, float scale) {
+  constexpr int DF = D / 16;
+  constexpr int KC = D / 32;
+  constexpr int RB = D * 2;
+  __shared__ __attribute__((aligned(16))) char lds[
+      2 * KVB * D * 2 +      // K, V [64][D]
+      D * KVB * 2 +          // Kt [D][64]
+      4 * 16 * KVB * 2];     // per-wave dS [16][64]
+  char* k_lds = lds;
+  char* v_lds = k_lds + KVB * D * 2;
+  char* kt_lds = v_lds + KVB * D * 2;
+  char* p_base = kt_lds + D * KVB * 2;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  char* my_ds = p_base + wid * 16 * KVB * 2;
+
+  const int n_qt = (S + 63) / 64;
+  const int bh = blockIdx.x / n_qt;
+  const int qt = blockIdx.x % n_qt;
+  const int b = bh / H, h = bh % H;
+  const int q0 = qt * 64 + wid * 16;
+
+  const long off = ((long)(b * H + h) * S) * D;
+  const long offT = ((long)(b * H + h) * D) * S;
+  const long roff = (long)(b * H + h) * S;
+
+  // Q (scaled) and dO fragments
+  bf16x8 qreg[KC], doreg[KC];
+  {
+    int row = q0 + (lane & 15);
+    if (row >= S) row = S - 1;
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      int d0 = c * 32 + (lane >> 4) * 8;
+      bf16x8 v8 = *reinterpret_cast<const bf16x8*>(
+          Q + off + (long)row * D + d0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v8[j] = (__bf16)((float)v8[j] * scale);
+      qreg[c] = v8;
+      doreg[c] = *reinterpret_cast<const bf16x8*>(
+          dO + off + (long)row * D + d0);
+    }
+  }
+  // per-lane lse/Drow for this lane's 4 C-rows
+  float lse_r[4], dr_r[4];
+  {
+    int r0 = qt * 64 + wid * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = r0 + r;
+      lse_r[r] = row < S ? LSE[roff + row] : 3e30f;
+      dr_r[r] = row < S ? DR[roff + row] : 0.f;
+    }
+  }
+
+  float4v dq_acc[DF];
+#pragma unroll
+  for (int f = 0; f < DF; ++f) dq_acc[f] = {0.f, 0.f, 0.f, 0.f};
+
+  const int q_tile_end = qt * 64 + 63;
+  const int n_kv = CAUSAL
+      ? min((q_tile_end / KVB) + 1, (S + KVB - 1) / KVB)
+      : (S + KVB - 1) / KVB;
+  for (int t = 0; t < n_kv; ++t) {
+    const int kv0 = t * KVB;
+    glds_tile<KVB, RB>(K, D, kv0, S, 0, D, k_lds, wid, lane);
+    glds_tile<KVB, RB>(V, D, kv0, S, 0, D, v_lds, wid, lane);
+    glds_tile<D, KVB * 2>(Kt, S, 0, D, kv0, S, kt_lds, wid, lane);
+    __syncthreads();
+
+    float4v st[4], dpt[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      st[f] = {0.f, 0.f, 0.f, 0.f};
+      dpt[f] = {0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int kk = 0; kk < KC; ++kk) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+            k_lds + swz(f * 16 + (lane & 15),
+                        kk * 64 + (lane >> 4) * 16, RB));
+        st[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qreg[kk], kf, st[f], 0, 0, 0);
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+            v_lds + swz(f * 16 + (lane & 15),
+                        kk * 64 + (lane >> 4) * 16, RB));
+        dpt[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            doreg[kk], vf, dpt[f], 0, 0, 0);
+      }
+    }
+    const int lrow0 = qt * 64 + wid * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int kcol = kv0 + f * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        bool dead = kcol >= S || lse_r[r] > 1e30f;
+        if (CAUSAL) dead |= kcol > lrow0 + r;
+        float p = dead ? 0.f : __expf(st[f][r] - lse_r[r]);
+        float ds = p * (dpt[f][r] - dr_r[r]) * scale;
+        int row = (lane >> 4) * 4 + r;
+        int cb = (f * 16 + (lane & 15)) * 2;
+        *reinterpret_cast<short*>(
+            my_ds + swz(row, cb, KVB * 2)) = f32_to_bf16(ds);
+      }
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 dsf = *reinterpret_cast<const bf16x8*>(
+          my_ds + swz(lane & 15, kk * 64 + (lane >> 4) * 16, KVB * 2));
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        bf16x8 ktf = *reinterpret_cast<const bf16x8*>(
+            kt_lds + swz(f * 16 + (lane & 15),
+                         kk * 64 + (lane >> 4) * 16, KVB * 2));
+        dq_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf, ktf, dq_acc[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int orow0 = qt * 64 + wid * 16 + (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = orow0 + r;
+    if (row >= S) continue;
+    short* dqr = dQ + off + (long)row * D;
+#pragma unroll
+    for (int f = 0; f < DF; ++f)
+      dqr[f * 16 + (lane & 15)] = f32_to_bf16(dq_acc[f][r]);
   }
 }
 
@@ -259,8 +631,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
 
 torch::Tensor transpose2d(torch::Tensor x);  // gemm_mfma.hip (batched)
 
-torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                       bool causal) {
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.dim() == 4, "attn_fwd: [B,H,S,D]");
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
@@ -275,6 +647,8 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   auto vt = transpose2d(v.reshape({(long)B * H, S, D}))
                 .reshape({B, H, D, S});
   auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, S},
+                          q.options().dtype(torch::kFloat32));
   float scale = 1.0f / sqrtf((float)D);
   int grid = B * H * ((S + 63) / 64);
   auto stream = c10::cuda::getCurrentCUDAStream();
@@ -284,7 +658,8 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                      (const short*)q.data_ptr(),                         \
                      (const short*)k.data_ptr(),                         \
                      (const short*)vt.data_ptr(),                        \
-                     (short*)o.data_ptr(), B, H, S, scale)
+                     (short*)o.data_ptr(), lse.data_ptr<float>(),        \
+                     B, H, S, scale)
   if (D == 64) {
     if (causal) { LAUNCH(64, true); } else { LAUNCH(64, false); }
   } else {
@@ -292,5 +667,62 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   }
 #undef LAUNCH
   HIP_CHECK_LAST();
-  return o;
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dout,
+                                    torch::Tensor lse, bool causal) {
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  TORCH_CHECK(dout.is_contiguous() && o.is_contiguous());
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  // pre-transposes (batched LDS-tiled kernel) + Drow
+  auto qt = transpose2d(q.reshape({(long)B * H, S, D}));
+  auto kt = transpose2d(k.reshape({(long)B * H, S, D}));
+  auto dot = transpose2d(dout.reshape({(long)B * H, S, D}));
+  long n_rows = (long)B * H * S;
+  auto drow = torch::empty({n_rows}, q.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(attn_drow_kernel,
+                     dim3((int)((n_rows + 3) / 4)), dim3(256), 0,
+                     stream.stream(), (const short*)dout.data_ptr(),
+                     (const short*)o.data_ptr(), drow.data_ptr<float>(),
+                     n_rows, D);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  float scale = 1.0f / sqrtf((float)D);
+  int grid_kv = B * H * ((S + KVB - 1) / KVB);
+  int grid_q = B * H * ((S + 63) / 64);
+#define LAUNCH_BWD(DD, CC)                                              \
+  do {                                                                  \
+    hipLaunchKernelGGL((attn_bwd_kv_kernel<DD, CC>), dim3(grid_kv),     \
+                       dim3(NTHREADS), 0, stream.stream(),              \
+                       (const short*)q.data_ptr(),                      \
+                       (const short*)k.data_ptr(),                      \
+                       (const short*)v.data_ptr(),                      \
+                       (const short*)dout.data_ptr(),                   \
+                       (const short*)qt.data_ptr(),                     \
+                       (const short*)dot.data_ptr(),                    \
+                       lse.data_ptr<float>(), drow.data_ptr<float>(),   \
+                       (short*)dk.data_ptr(), (short*)dv.data_ptr(),    \
+                       B, H, S, scale);                                 \
+    hipLaunchKernelGGL((attn_bwd_q_kernel<DD, CC>), dim3(grid_q),       \
+                       dim3(NTHREADS), 0, stream.stream(),              \
+                       (const short*)q.data_ptr(),                      \
+                       (const short*)k.data_ptr(),                      \
+                       (const short*)v.data_ptr(),                      \
+                       (const short*)dout.data_ptr(),                   \
+                       (const short*)kt.data_ptr(),                     \
+                       lse.data_ptr<float>(), drow.data_ptr<float>(),   \
+                       (short*)dq.data_ptr(), B, H, S, scale);          \
+  } while (0)
+  if (D == 64) {
+    if (causal) LAUNCH_BWD(64, true); else LAUNCH_BWD(64, false);
+  } else {
+    if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false);
+  }
+#undef LAUNCH_BWD
+  HIP_CHECK_LAST();
+  return {dq, dk, dv};
 }

@@ -16,8 +16,12 @@ void adamw_step(torch::Tensor p_bf16, torch::Tensor p32, torch::Tensor grad,
                 double beta2, double eps, double wd, long step);
 void sgd_step(torch::Tensor p32, torch::Tensor p_bf16, torch::Tensor grad,
               torch::Tensor buf, double lr, double momentum, double wd);
-torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                       bool causal);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dout,
+                                    torch::Tensor lse, bool causal);
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   c10::optional<torch::Tensor> use_mean,
@@ -43,7 +47,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
-  m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax)");
+  m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax); returns (o, lse)");
+  m.def("attn_bwd", &attn_bwd, "flash-attention backward (FA2 two-kernel)");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward");
   m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");

@@ -24,6 +24,7 @@ def ref_attn(q, k, v, causal):
 
 
 def check(B, H, S, D, causal, seed=0, spike=False):
+    # (shape args named for the lse reference below)
     torch.manual_seed(seed)
     q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
     k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
@@ -33,8 +34,16 @@ def check(B, H, S, D, causal, seed=0, spike=False):
         # the LAST kv tile for the FIRST q rows (max jumps at the end)
         k[:, :, -3, :] = 8.0
         q[:, :, :5, :] = q[:, :, :5, :].abs()
-    o = ext().attn_fwd(q, k, v, causal)
+    o, lse = ext().attn_fwd(q, k, v, causal)
     ref = ref_attn(q, k, v, causal)
+    # lse vs reference logsumexp of scaled scores
+    scale = 1.0 / math.sqrt(D)
+    s_ref = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.ones(S, S, dtype=torch.bool, device=q.device).triu(1)
+        s_ref = s_ref.masked_fill(mask, float("-inf"))
+    lse_ref = torch.logsumexp(s_ref, dim=-1)
+    assert (lse - lse_ref).abs().max().item() < 5e-2
     err = (o.float() - ref).abs().max().item()
     scale = ref.abs().max().item() + 1e-3
     assert err / scale < 1.5e-2, f"rel err {err/scale:.4f} (abs {err:.4f})"
@@ -102,3 +111,28 @@ def test_llama_tiny_job_step_on_gpu():
     assert all(torch.isfinite(torch.tensor(losses)))
     # training sanity: loss should move (weights updating)
     assert abs(losses[-1] - losses[0]) > 1e-4
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("shape", [(2, 3, 128, 64), (1, 2, 256, 128),
+                                   (1, 1, 200, 64), (2, 2, 512, 128)])
+def test_attn_bwd_kernels(shape, causal):
+    B, H, S, D = shape
+    torch.manual_seed(11)
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    do = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    o, lse = ext().attn_fwd(q, k, v, causal)
+    dq, dk, dv = ext().attn_bwd(q, k, v, o, do, lse, causal)
+
+    qf = q.float().requires_grad_(True)
+    kf = k.float().requires_grad_(True)
+    vf = v.float().requires_grad_(True)
+    ref = ref_attn(qf, kf, vf, causal)
+    ref.backward(do.float())
+    for got, want, name in ((dq, qf.grad, "dq"), (dk, kf.grad, "dk"),
+                            (dv, vf.grad, "dv")):
+        scale = want.abs().max().item() + 1e-4
+        err = (got.float() - want).abs().max().item() / scale
+        assert err < 2.5e-2, f"{name} rel err {err:.4f}"
