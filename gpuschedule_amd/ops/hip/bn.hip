@@ -83,10 +83,10 @@ __global__ __launch_bounds__(NT) void bn_stats_kernel(
     atomicAdd(lds_b + ch0 + j, acc_b[j]);
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < C; i += NT) {
-    atomicAdd(out_a + i, lds_a[i]);
-    atomicAdd(out_b + i, lds_b[i]);
-  }
+  // per-block partial slot [blockIdx][2C]: no pre-zeroed global
+  // accumulator, no global atomics; finalize sums the slots.
+  float* slot = out_a + (long)blockIdx.x * 2 * C;
+  for (int i = threadIdx.x; i < 2 * C; i += NT) slot[i] = lds_a[i];
 }
 
 // ---- apply: y = x * scale[c] + shift[c] (+relu) -------------------
@@ -169,17 +169,34 @@ __global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
 
 // finalize: sums -> mean/rstd/scale/shift (+running-stat update), one
 // kernel instead of ~10 C-length torch ops per BN layer per step.
+// 64 channels x 4 slot-lanes per block: the slot loop is 4-way
+// parallel per channel (a serial P-loop with few threads was the
+// LN-backward mistake; rule: reductions get chip-filling parallelism).
 __global__ void bn_finalize_kernel(
-    const float* __restrict__ sums, const float* __restrict__ gamma,
+    const float* __restrict__ part, int P,
+    const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ mean_out,
     float* __restrict__ rstd_out, float* __restrict__ scale,
     float* __restrict__ shift, float* __restrict__ running_mean,
     float* __restrict__ running_var, int C, float inv_cnt,
     float unbias, float momentum, float eps) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float mu = sums[c] * inv_cnt;
-  float var = sums[C + c] * inv_cnt - mu * mu;
+  __shared__ float sh[2][64][4];
+  const int cl = threadIdx.x & 63, pl = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + cl;
+  float s1 = 0.f, s2 = 0.f;
+  if (c < C)
+    for (int p = pl; p < P; p += 4) {
+      s1 += part[(long)p * 2 * C + c];
+      s2 += part[(long)p * 2 * C + C + c];
+    }
+  sh[0][cl][pl] = s1;
+  sh[1][cl][pl] = s2;
+  __syncthreads();
+  if (pl != 0 || c >= C) return;
+  s1 = sh[0][cl][0] + sh[0][cl][1] + sh[0][cl][2] + sh[0][cl][3];
+  s2 = sh[1][cl][0] + sh[1][cl][1] + sh[1][cl][2] + sh[1][cl][3];
+  float mu = s1 * inv_cnt;
+  float var = s2 * inv_cnt - mu * mu;
   float rs = rsqrtf(var + eps);
   mean_out[c] = mu;
   rstd_out[c] = rs;
@@ -191,6 +208,21 @@ __global__ void bn_finalize_kernel(
     running_var[c] = (1.f - momentum) * running_var[c]
         + momentum * var * unbias;
   }
+}
+
+__global__ void bn_reduce_slots_kernel(
+    const float* __restrict__ part, int P, float* __restrict__ out,
+    int n) {
+  __shared__ float sh[64][4];
+  const int cl = threadIdx.x & 63, pl = threadIdx.x >> 6;
+  const int i = blockIdx.x * 64 + cl;
+  float s = 0.f;
+  if (i < n)
+    for (int p = pl; p < P; p += 4) s += part[(long)p * n + i];
+  sh[cl][pl] = s;
+  __syncthreads();
+  if (pl == 0 && i < n)
+    out[i] = sh[cl][0] + sh[cl][1] + sh[cl][2] + sh[cl][3];
 }
 
 __global__ void bn_scale_shift_kernel(
@@ -205,8 +237,10 @@ __global__ void bn_scale_shift_kernel(
 }
 
 int grid_for(long E) {
+  // enough blocks to saturate HBM, few enough that the finalize
+  // slot-reduce stays cheap
   long b = (E + PASS - 1) / PASS;
-  return (int)std::min<long>(b > 0 ? b : 1, 2048);
+  return (int)std::min<long>(b > 0 ? b : 1, 304);
 }
 
 void check_bn(const torch::Tensor& x, int C) {
@@ -244,13 +278,13 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                        beta.data_ptr<float>(), scale.data_ptr<float>(),
                        shift.data_ptr<float>(), C);
   } else {
-    auto sums = torch::zeros({2 * C}, fopt);
+    int sgrid = grid_for(E);
+    auto part = torch::empty({sgrid, 2 * C}, fopt);
     hipLaunchKernelGGL((bn_stats_kernel<false, false>),
-                       dim3(grid_for(E)), dim3(NT), 2 * C * 4,
+                       dim3(sgrid), dim3(NT), 2 * C * 4,
                        stream.stream(), (const short*)x.data_ptr(),
                        nullptr, nullptr, nullptr, nullptr,
-                       sums.data_ptr<float>(),
-                       sums.data_ptr<float>() + C, E, C);
+                       part.data_ptr<float>(), nullptr, E, C);
     long cnt = E / C;
     mean = torch::empty({C}, fopt);
     rstd = torch::empty({C}, fopt);
@@ -258,8 +292,9 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                      : nullptr;
     float* rv = run_var.has_value() ? run_var->data_ptr<float>()
                                     : nullptr;
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3(cgrid), dim3(NT), 0,
-                       stream.stream(), sums.data_ptr<float>(),
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 63) / 64),
+                       dim3(NT), 0,
+                       stream.stream(), part.data_ptr<float>(), sgrid,
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        scale.data_ptr<float>(), shift.data_ptr<float>(),
@@ -295,27 +330,31 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   const long E = x.numel();
   auto fopt = x.options().dtype(torch::kFloat32);
   auto stream = c10::cuda::getCurrentCUDAStream();
-  auto sums = torch::zeros({2 * C}, fopt);
+  int sgrid = grid_for(E);
+  auto part = torch::empty({sgrid, 2 * C}, fopt);
+  auto sums = torch::empty({2 * C}, fopt);
   auto sum_dy = sums.narrow(0, 0, C);
   auto sum_dyxh = sums.narrow(0, C, C);
   const short* y_p = relu ? (const short*)y->data_ptr() : nullptr;
   if (relu) {
-    hipLaunchKernelGGL((bn_stats_kernel<true, true>), dim3(grid_for(E)),
+    hipLaunchKernelGGL((bn_stats_kernel<true, true>), dim3(sgrid),
                        dim3(NT), 2 * C * 4, stream.stream(),
                        (const short*)x.data_ptr(),
                        (const short*)dy.data_ptr(), y_p,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       sum_dy.data_ptr<float>(),
-                       sum_dyxh.data_ptr<float>(), E, C);
+                       part.data_ptr<float>(), nullptr, E, C);
   } else {
-    hipLaunchKernelGGL((bn_stats_kernel<true, false>), dim3(grid_for(E)),
+    hipLaunchKernelGGL((bn_stats_kernel<true, false>), dim3(sgrid),
                        dim3(NT), 2 * C * 4, stream.stream(),
                        (const short*)x.data_ptr(),
                        (const short*)dy.data_ptr(), nullptr,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       sum_dy.data_ptr<float>(),
-                       sum_dyxh.data_ptr<float>(), E, C);
+                       part.data_ptr<float>(), nullptr, E, C);
   }
+  hipLaunchKernelGGL(bn_reduce_slots_kernel,
+                     dim3((2 * C + 63) / 64), dim3(NT), 0,
+                     stream.stream(), part.data_ptr<float>(), sgrid,
+                     sums.data_ptr<float>(), 2 * C);
   auto dx = torch::empty_like(x);
   float inv_cnt = 1.0f / (float)(E / C);
   if (relu) {
