@@ -28,7 +28,7 @@ class BasicBlock(nn.Module):
         self.conv1 = conv3x3(cin, cout, stride)
         self.bn1 = FusedBN2d(cout, relu=True)
         self.conv2 = conv3x3(cout, cout)
-        self.bn2 = FusedBN2d(cout)
+        self.bn2 = FusedBN2d(cout, relu=True)  # fused +residual +relu
         self.down = None
         if stride != 1 or cin != cout:
             self.down = nn.Sequential(
@@ -38,8 +38,7 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         idt = x if self.down is None else self.down(x)
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        return F.relu(out + idt)
+        return self.bn2(self.conv2(out), residual=idt)
 
 
 class Bottleneck(nn.Module):
@@ -53,7 +52,7 @@ class Bottleneck(nn.Module):
         self.conv2 = conv3x3(cmid, cmid, stride)
         self.bn2 = FusedBN2d(cmid, relu=True)
         self.conv3 = nn.Conv2d(cmid, cout, 1, bias=False)
-        self.bn3 = FusedBN2d(cout)
+        self.bn3 = FusedBN2d(cout, relu=True)  # fused +residual +relu
         self.down = None
         if stride != 1 or cin != cout:
             self.down = nn.Sequential(
@@ -64,8 +63,7 @@ class Bottleneck(nn.Module):
         idt = x if self.down is None else self.down(x)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
-        return F.relu(out + idt)
+        return self.bn3(self.conv3(out), residual=idt)
 
 
 class ResNet(nn.Module):

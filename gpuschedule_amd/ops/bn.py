@@ -16,27 +16,31 @@ from ._ext import get_ext
 
 class _BNFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2d, weight, bias, mean, rstd, rm, rv, momentum,
-                eps, relu, training):
+    def forward(ctx, x2d, res2d, weight, bias, mean, rstd, rm, rv,
+                momentum, eps, relu, training):
         ext = get_ext(required=True)
         if training:
             y, mean, rstd = ext.bn_fwd(x2d, weight, bias, None, None,
-                                       rm, rv, momentum, eps, relu)
+                                       rm, rv, res2d, momentum, eps,
+                                       relu)
         else:
             y, _, _ = ext.bn_fwd(x2d, weight, bias, mean, rstd, None,
-                                 None, momentum, eps, relu)
+                                 None, res2d, momentum, eps, relu)
         ctx.save_for_backward(x2d, y if relu else None, mean, rstd,
                               weight)
         ctx.relu = relu
+        ctx.has_res = res2d is not None
         return y, mean, rstd
 
     @staticmethod
     def backward(ctx, dy, _dm, _dr):
         ext = get_ext(required=True)
         x2d, y, mean, rstd, weight = ctx.saved_tensors
-        dx, dbeta, dgamma = ext.bn_bwd(dy.contiguous(), x2d, y, mean,
-                                       rstd, weight, ctx.relu)
-        return (dx, dgamma, dbeta, None, None, None, None,
+        out = ext.bn_bwd(dy.contiguous(), x2d, y, mean,
+                         rstd, weight, ctx.relu, ctx.has_res)
+        dx, dbeta, dgamma = out[0], out[1], out[2]
+        dres = out[3] if ctx.has_res else None
+        return (dx, dres, dgamma, dbeta, None, None, None, None,
                 None, None, None, None)
 
 
@@ -53,27 +57,34 @@ class FusedBN2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(channels))
         self.register_buffer("running_var", torch.ones(channels))
 
-    def _cpu(self, x):
+    def _cpu(self, x, residual=None):
         y = F.batch_norm(x.float(), self.running_mean, self.running_var,
                          self.weight, self.bias, self.training,
                          self.momentum, self.eps).to(x.dtype)
+        if residual is not None:
+            y = y + residual
         return F.relu(y) if self.relu else y
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         if not x.is_cuda:
-            return self._cpu(x)
+            return self._cpu(x, residual)
         N, C, H, W = x.shape
         x2d = x.contiguous(memory_format=torch.channels_last) \
                .permute(0, 2, 3, 1).reshape(-1, C)
+        res2d = None
+        if residual is not None:
+            res2d = residual.contiguous(
+                memory_format=torch.channels_last) \
+                .permute(0, 2, 3, 1).reshape(-1, C)
         if self.training:
             mean = rstd = self.running_mean  # placeholders (unused)
         else:
             mean = self.running_mean
             rstd = (self.running_var + self.eps).rsqrt()
         y2d, mean, rstd = _BNFn.apply(
-            x2d, self.weight, self.bias, mean, rstd, self.running_mean,
-            self.running_var, self.momentum, self.eps, self.relu,
-            self.training)
+            x2d, res2d, self.weight, self.bias, mean, rstd,
+            self.running_mean, self.running_var, self.momentum,
+            self.eps, self.relu, self.training)
         return y2d.view(N, H, W, C).permute(0, 3, 1, 2)
 
     def extra_repr(self):

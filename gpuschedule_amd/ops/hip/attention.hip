@@ -389,10 +389,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   const int n_qt = (S + 63) / 64;
   for (int qt = qt_start; qt < n_qt; ++qt) {
     const int q0 = qt * 64;
-    glds_tile<KVB, RB>(Q, D, q0, S, 0, D, q_lds, wid, lane);
-    glds_tile<KVB, RB>(dO, D, q0, S, 0, D, do_lds, wid, lane);
-    glds_tile<D, KVB * 2>(Qt, S, 0, D, q0, S, qt_lds, wid, lane);
-    glds_tile<D, KVB * 2>(dOt, S, 0, D, q0, S, dot_lds, wid, lane);
+    glds_tile<KVB, RB>(Q + off, D, q0, S, 0, D, q_lds, wid, lane);
+    glds_tile<KVB, RB>(dO + off, D, q0, S, 0, D, do_lds, wid, lane);
+    glds_tile<D, KVB * 2>(Qt + offT, S, 0, D, q0, S, qt_lds, wid, lane);
+    glds_tile<D, KVB * 2>(dOt + offT, S, 0, D, q0, S, dot_lds, wid, lane);
     for (int i = tid; i < KVB; i += NTHREADS) {
       int qr = q0 + i;
       lse_lds[i] = qr < S ? LSE[roff + qr] : 3e30f;
@@ -556,9 +556,9 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
       : (S + KVB - 1) / KVB;
   for (int t = 0; t < n_kv; ++t) {
     const int kv0 = t * KVB;
-    glds_tile<KVB, RB>(K, D, kv0, S, 0, D, k_lds, wid, lane);
-    glds_tile<KVB, RB>(V, D, kv0, S, 0, D, v_lds, wid, lane);
-    glds_tile<D, KVB * 2>(Kt, S, 0, D, kv0, S, kt_lds, wid, lane);
+    glds_tile<KVB, RB>(K + off, D, kv0, S, 0, D, k_lds, wid, lane);
+    glds_tile<KVB, RB>(V + off, D, kv0, S, 0, D, v_lds, wid, lane);
+    glds_tile<D, KVB * 2>(Kt + offT, S, 0, D, kv0, S, kt_lds, wid, lane);
     __syncthreads();
 
     float4v st[4], dpt[4];

@@ -28,6 +28,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   c10::optional<torch::Tensor> use_rstd,
                                   c10::optional<torch::Tensor> run_mean,
                                   c10::optional<torch::Tensor> run_var,
+                                  c10::optional<torch::Tensor> residual,
                                   double momentum, double eps, bool relu);
 void sgd_step_fp32(torch::Tensor p, torch::Tensor grad, torch::Tensor buf,
                    double lr, double momentum, double wd);
@@ -38,7 +39,8 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
-                                  torch::Tensor gamma, bool relu);
+                                  torch::Tensor gamma, bool relu,
+                                  bool want_dres);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "fused LayerNorm/RMSNorm forward (bf16)");

@@ -90,9 +90,10 @@ __global__ __launch_bounds__(NT) void bn_stats_kernel(
 }
 
 // ---- apply: y = x * scale[c] + shift[c] (+relu) -------------------
-template <bool RELU>
+template <bool RELU, bool RES>
 __global__ __launch_bounds__(NT) void bn_apply_kernel(
-    const short* __restrict__ x, short* __restrict__ y,
+    const short* __restrict__ x, const short* __restrict__ res,
+    short* __restrict__ y,
     const float* __restrict__ scale, const float* __restrict__ shift,
     long E, int C) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -109,10 +110,13 @@ __global__ __launch_bounds__(NT) void bn_apply_kernel(
   long stride = (long)gridDim.x * PASS;
   for (long e = base; e + 7 < E; e += stride) {
     short8 xv = *reinterpret_cast<const short8*>(x + e);
+    short8 rv;
+    if (RES) rv = *reinterpret_cast<const short8*>(res + e);
     short8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = bf16_to_f32(xv[j]) * s_sc[ch0 + j] + s_sh[ch0 + j];
+      if (RES) f += bf16_to_f32(rv[j]);
       if (RELU) f = fmaxf(f, 0.f);
       out[j] = f32_to_bf16(f);
     }
@@ -122,10 +126,11 @@ __global__ __launch_bounds__(NT) void bn_apply_kernel(
 
 // ---- backward apply:
 // dx = g*rstd * (dy' - sum_dy/Nc - xhat * sum_dyxhat/Nc)
-template <bool RELU>
+template <bool RELU, bool RES>
 __global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const short* __restrict__ y, short* __restrict__ dx,
+    short* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const float* __restrict__ gamma, const float* __restrict__ sum_dy,
     const float* __restrict__ sum_dyxh, long E, int C, float inv_cnt) {
@@ -154,16 +159,18 @@ __global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
     short8 xv = *reinterpret_cast<const short8*>(x + e);
     short8 yv;
     if (RELU) yv = *reinterpret_cast<const short8*>(y + e);
-    short8 out;
+    short8 out, rout;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = ch0 + j;
       float d = bf16_to_f32(dv[j]);
       if (RELU && bf16_to_f32(yv[j]) <= 0.f) d = 0.f;
+      if (RES) rout[j] = f32_to_bf16(d);
       float xh = (bf16_to_f32(xv[j]) - s_mu[c]) * s_rs[c];
       out[j] = f32_to_bf16(s_k1[c] * (d - s_c1[c] - xh * s_c2[c]));
     }
     *reinterpret_cast<short8*>(dx + e) = out;
+    if (RES) *reinterpret_cast<short8*>(dres + e) = rout;
   }
 }
 
@@ -259,6 +266,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   c10::optional<torch::Tensor> use_rstd,
                                   c10::optional<torch::Tensor> run_mean,
                                   c10::optional<torch::Tensor> run_var,
+                                  c10::optional<torch::Tensor> residual,
                                   double momentum, double eps, bool relu) {
   const int C = x.size(-1);
   check_bn(x, C);
@@ -303,19 +311,17 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                        (float)momentum, (float)eps);
   }
   auto y = torch::empty_like(x);
-  if (relu) {
-    hipLaunchKernelGGL((bn_apply_kernel<true>), dim3(grid_for(E)),
-                       dim3(NT), 2 * C * 4, stream.stream(),
-                       (const short*)x.data_ptr(), (short*)y.data_ptr(),
-                       scale.data_ptr<float>(), shift.data_ptr<float>(),
-                       E, C);
-  } else {
-    hipLaunchKernelGGL((bn_apply_kernel<false>), dim3(grid_for(E)),
-                       dim3(NT), 2 * C * 4, stream.stream(),
-                       (const short*)x.data_ptr(), (short*)y.data_ptr(),
-                       scale.data_ptr<float>(), shift.data_ptr<float>(),
-                       E, C);
-  }
+  const short* res_p = residual.has_value()
+      ? (const short*)residual->data_ptr() : nullptr;
+#define BN_APPLY(RL, RS)                                                 \
+  hipLaunchKernelGGL((bn_apply_kernel<RL, RS>), dim3(grid_for(E)),       \
+                     dim3(NT), 2 * C * 4, stream.stream(),               \
+                     (const short*)x.data_ptr(), res_p,                  \
+                     (short*)y.data_ptr(), scale.data_ptr<float>(),      \
+                     shift.data_ptr<float>(), E, C)
+  if (relu) { if (res_p) BN_APPLY(true, true); else BN_APPLY(true, false); }
+  else { if (res_p) BN_APPLY(false, true); else BN_APPLY(false, false); }
+#undef BN_APPLY
   HIP_CHECK_LAST();
   return {y, mean, rstd};
 }
@@ -324,7 +330,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
-                                  torch::Tensor gamma, bool relu) {
+                                  torch::Tensor gamma, bool relu,
+                                  bool want_dres) {
   const int C = x.size(-1);
   check_bn(x, C);
   const long E = x.numel();
@@ -356,26 +363,32 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                      stream.stream(), part.data_ptr<float>(), sgrid,
                      sums.data_ptr<float>(), 2 * C);
   auto dx = torch::empty_like(x);
-  float inv_cnt = 1.0f / (float)(E / C);
-  if (relu) {
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<true>), dim3(grid_for(E)),
-                       dim3(NT), 5 * C * 4, stream.stream(),
-                       (const short*)dy.data_ptr(),
-                       (const short*)x.data_ptr(), y_p,
-                       (short*)dx.data_ptr(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                       sum_dy.data_ptr<float>(),
-                       sum_dyxh.data_ptr<float>(), E, C, inv_cnt);
-  } else {
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<false>), dim3(grid_for(E)),
-                       dim3(NT), 5 * C * 4, stream.stream(),
-                       (const short*)dy.data_ptr(),
-                       (const short*)x.data_ptr(), nullptr,
-                       (short*)dx.data_ptr(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                       sum_dy.data_ptr<float>(),
-                       sum_dyxh.data_ptr<float>(), E, C, inv_cnt);
+  torch::Tensor dres;
+  short* dres_p = nullptr;
+  if (want_dres) {
+    dres = torch::empty_like(x);
+    dres_p = (short*)dres.data_ptr();
   }
+  float inv_cnt = 1.0f / (float)(E / C);
+#define BN_BWD_APPLY(RL, RS, YP)                                         \
+  hipLaunchKernelGGL((bn_bwd_apply_kernel<RL, RS>), dim3(grid_for(E)),   \
+                     dim3(NT), 5 * C * 4, stream.stream(),               \
+                     (const short*)dy.data_ptr(),                        \
+                     (const short*)x.data_ptr(), YP,                     \
+                     (short*)dx.data_ptr(), dres_p,                      \
+                     mean.data_ptr<float>(),                             \
+                     rstd.data_ptr<float>(), gamma.data_ptr<float>(),    \
+                     sum_dy.data_ptr<float>(),                           \
+                     sum_dyxh.data_ptr<float>(), E, C, inv_cnt)
+  if (relu) {
+    if (want_dres) BN_BWD_APPLY(true, true, y_p);
+    else BN_BWD_APPLY(true, false, y_p);
+  } else {
+    if (want_dres) BN_BWD_APPLY(false, true, nullptr);
+    else BN_BWD_APPLY(false, false, nullptr);
+  }
+#undef BN_BWD_APPLY
   HIP_CHECK_LAST();
+  if (want_dres) return {dx, sum_dy, sum_dyxh, dres};
   return {dx, sum_dy, sum_dyxh};
 }

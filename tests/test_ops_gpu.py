@@ -199,7 +199,7 @@ def test_fused_bn_fwd_bwd(shape, relu):
     dy = torch.randn(rows, C, device="cuda", dtype=torch.bfloat16)
 
     y, mean, rstd = ext().bn_fwd(x, g, b, None, None, None, None,
-                                 0.1, 1e-5, relu)
+                                 None, 0.1, 1e-5, relu)
     xf = x.float().requires_grad_(True)
     gf = g.clone().requires_grad_(True)
     bf = b.clone().requires_grad_(True)
@@ -211,7 +211,7 @@ def test_fused_bn_fwd_bwd(shape, relu):
     assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 1e-2
     ref.backward(dy.float())
     dx, dbeta, dgamma = ext().bn_bwd(dy, x, y if relu else None, mean,
-                                     rstd, g, relu)
+                                     rstd, g, relu, False)
     s = xf.grad.abs().max().item() + 1e-4
     assert mx(dx, xf.grad) / s < 2e-2
     s = gf.grad.abs().max().item() + 1e-3
@@ -260,3 +260,27 @@ def test_fused_ce(shape):
     torch.nn.functional.cross_entropy(lf, labels).backward()
     scale = lf.grad.abs().max().item() + 1e-9
     assert (lg.grad.float() - lf.grad).abs().max().item() / scale < 2e-2
+
+
+def test_fused_bn_residual_relu():
+    from gpuschedule_amd.ops.bn import FusedBN2d
+    torch.manual_seed(12)
+    C = 128
+    fb = FusedBN2d(C, relu=True).cuda()
+    x = torch.randn(4, C, 8, 8, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    res = torch.randn_like(x).requires_grad_(True)
+    y = fb(x, residual=res)
+    y.float().square().mean().backward()
+
+    xf = x.detach().float().requires_grad_(True)
+    rf = res.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(
+        xf, None, None, fb.weight.detach().clone().requires_grad_(False),
+        fb.bias.detach().clone(), True, 0.0, fb.eps)
+    ref = torch.relu(ref + rf)
+    ref.square().mean().backward()
+    assert (y.float() - ref).abs().max().item() < 5e-2
+    for got, want in ((x.grad, xf.grad), (res.grad, rf.grad)):
+        s = want.abs().max().item() + 1e-6
+        assert (got.float() - want).abs().max().item() / s < 3e-2
