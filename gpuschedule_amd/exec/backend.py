@@ -188,7 +188,11 @@ class TorchBackend:
             try:
                 done, secs = r.step(n)
                 if r.gang_rank == 0:
-                    report[job_id] = ("ok", done, secs, r.last_loss)
+                    mem = 0
+                    if r.measured_mem and not r._mem_reported:
+                        mem = r.measured_mem
+                        r._mem_reported = True
+                    report[job_id] = ("ok", done, secs, r.last_loss, mem)
             except Exception as e:  # job-level fault isolation
                 report[job_id] = ("fail", 0, 0.0, f"{type(e).__name__}: {e}")
 
@@ -201,7 +205,9 @@ class TorchBackend:
                         self.sched.on_fail(job_id, r[3], now)
                         self._pending_cleanup.append(job_id)
                     else:
-                        _, done, secs, _ = r
+                        _, done, secs, _, mem = r
+                        if mem:
+                            self.sched.cluster.update_mem(job_id, mem)
                         width = len(self.sched.jobs[job_id].placement)
                         fin = self.sched.on_progress(
                             job_id, done, secs * width, now,

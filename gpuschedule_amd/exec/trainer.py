@@ -73,6 +73,12 @@ class JobRunner:
         # jobs' kernels can overlap.
         self.stream = (torch.cuda.Stream(device)
                        if str(device) != "cpu" else None)
+        # measured per-GPU footprint (reported once to the scheduler
+        # for bin-packing refinement)
+        self._mem_base = (torch.cuda.memory_allocated(device)
+                          if self.stream is not None else 0)
+        self.measured_mem = 0
+        self._mem_reported = False
 
     # ---- state movement -----------------------------------------
 
@@ -161,6 +167,10 @@ class JobRunner:
                     loss = self._one_iter(x, y)
                     self.done_iters += 1
             self.stream.synchronize()
+            if self.measured_mem == 0 and self.done_iters >= 4:
+                used = (torch.cuda.memory_allocated(self.device)
+                        - self._mem_base)
+                self.measured_mem = max(used, 1)
         if loss is not None:
             self.last_loss = float(loss.detach())
         return n_iters, time.perf_counter() - t0

@@ -164,3 +164,17 @@ class Cluster:
 
     def job_mem(self, job_id: str) -> int:
         return self._mem.get(job_id, 0)
+
+    def update_mem(self, job_id: str, measured: int) -> None:
+        """Refine a job's footprint from the backend's measured HBM use
+        (catalog numbers are conservative priors; measured values give
+        tighter bin-packing).  Floor at 256 MB."""
+        old = self._mem.get(job_id)
+        new = max(measured, 256 << 20)
+        if old is None or new == old:
+            return
+        delta = new - old
+        for g in self.gpus:
+            if job_id in g.resident:
+                g.resident_bytes += delta
+        self._mem[job_id] = new
