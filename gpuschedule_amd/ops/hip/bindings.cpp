@@ -11,6 +11,7 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor transpose2d(torch::Tensor x);
+std::vector<torch::Tensor> transpose2d_colsum(torch::Tensor x);
 void adamw_step(torch::Tensor p_bf16, torch::Tensor p32, torch::Tensor grad,
                 torch::Tensor m, torch::Tensor v, double lr, double beta1,
                 double beta2, double eps, double wd, long step);
@@ -47,6 +48,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_bwd", &norm_bwd, "fused LayerNorm/RMSNorm backward (bf16)");
   m.def("gemm_nt", &gemm_nt, "MFMA bf16 GEMM C = A @ B^T (+bias, +relu)");
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
+  m.def("transpose2d_colsum", &transpose2d_colsum,
+        "bf16 transpose + fp32 column sums (bias grad)");
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
   m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax); returns (o, lse)");

@@ -187,12 +187,12 @@ __global__ void bn_finalize_kernel(
     float* __restrict__ shift, float* __restrict__ running_mean,
     float* __restrict__ running_var, int C, float inv_cnt,
     float unbias, float momentum, float eps) {
-  __shared__ float sh[2][64][4];
+  __shared__ float sh[2][64][16];
   const int cl = threadIdx.x & 63, pl = threadIdx.x >> 6;
   const int c = blockIdx.x * 64 + cl;
   float s1 = 0.f, s2 = 0.f;
   if (c < C)
-    for (int p = pl; p < P; p += 4) {
+    for (int p = pl; p < P; p += 16) {
       s1 += part[(long)p * 2 * C + c];
       s2 += part[(long)p * 2 * C + C + c];
     }
@@ -200,8 +200,9 @@ __global__ void bn_finalize_kernel(
   sh[1][cl][pl] = s2;
   __syncthreads();
   if (pl != 0 || c >= C) return;
-  s1 = sh[0][cl][0] + sh[0][cl][1] + sh[0][cl][2] + sh[0][cl][3];
-  s2 = sh[1][cl][0] + sh[1][cl][1] + sh[1][cl][2] + sh[1][cl][3];
+  s1 = 0.f; s2 = 0.f;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) { s1 += sh[0][cl][j]; s2 += sh[1][cl][j]; }
   float mu = s1 * inv_cnt;
   float var = s2 * inv_cnt - mu * mu;
   float rs = rsqrtf(var + eps);
@@ -220,16 +221,19 @@ __global__ void bn_finalize_kernel(
 __global__ void bn_reduce_slots_kernel(
     const float* __restrict__ part, int P, float* __restrict__ out,
     int n) {
-  __shared__ float sh[64][4];
+  __shared__ float sh[64][16];
   const int cl = threadIdx.x & 63, pl = threadIdx.x >> 6;
   const int i = blockIdx.x * 64 + cl;
   float s = 0.f;
   if (i < n)
-    for (int p = pl; p < P; p += 4) s += part[(long)p * n + i];
+    for (int p = pl; p < P; p += 16) s += part[(long)p * n + i];
   sh[cl][pl] = s;
   __syncthreads();
-  if (pl == 0 && i < n)
-    out[i] = sh[cl][0] + sh[cl][1] + sh[cl][2] + sh[cl][3];
+  if (pl != 0 || i >= n) return;
+  s = 0.f;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) s += sh[cl][j];
+  out[i] = s;
 }
 
 __global__ void bn_scale_shift_kernel(
@@ -301,7 +305,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     float* rv = run_var.has_value() ? run_var->data_ptr<float>()
                                     : nullptr;
     hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 63) / 64),
-                       dim3(NT), 0,
+                       dim3(1024), 0,
                        stream.stream(), part.data_ptr<float>(), sgrid,
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
@@ -359,7 +363,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        part.data_ptr<float>(), nullptr, E, C);
   }
   hipLaunchKernelGGL(bn_reduce_slots_kernel,
-                     dim3((2 * C + 63) / 64), dim3(NT), 0,
+                     dim3((2 * C + 63) / 64), dim3(1024), 0,
                      stream.stream(), part.data_ptr<float>(), sgrid,
                      sums.data_ptr<float>(), 2 * C);
   auto dx = torch::empty_like(x);

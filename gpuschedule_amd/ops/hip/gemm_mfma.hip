@@ -309,9 +309,13 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_kernel(
 // ---- bf16 2-D transpose (LDS-tiled, vectorized loads) --------------
 // grid.z = batch index (stride R*C per matrix) for batched use.
 constexpr int TT = 64;  // tile
+// colsum (optional): fp32 per-input-column sums accumulated while the
+// tile is already in LDS — gives nn.Linear's bias grad (dy.sum(0)) for
+// free instead of a separate full-tensor torch reduce pass.
+template <bool COLSUM>
 __global__ __launch_bounds__(256) void transpose_kernel(
     const short* __restrict__ in0, short* __restrict__ out0,
-    int R, int Cc) {
+    float* __restrict__ colsum, int R, int Cc) {
   const short* in = in0 + (long)blockIdx.z * R * Cc;
   short* out = out0 + (long)blockIdx.z * R * Cc;
   __shared__ short tile[TT][TT + 8];
@@ -336,6 +340,15 @@ __global__ __launch_bounds__(256) void transpose_kernel(
     }
   }
   __syncthreads();
+  if (COLSUM) {
+    // threads 0..63 each own one tile column
+    if (tid < TT && tc0 + tid < Cc) {
+      float s = 0.f;
+      int rmax = min(TT, R - tr0);
+      for (int r = 0; r < rmax; ++r) s += bf16_to_f32(tile[r][tid]);
+      atomicAdd(colsum + (long)blockIdx.z * Cc + tc0 + tid, s);
+    }
+  }
   // store: out[c][r]; thread writes 8 contiguous out-cols (= in-rows)
 #pragma unroll
   for (int p = 0; p < 2; ++p) {
@@ -409,9 +422,25 @@ torch::Tensor transpose2d(torch::Tensor x) {
                      : torch::empty({C, R}, x.options());
   dim3 grid((C + TT - 1) / TT, (R + TT - 1) / TT, Bn);
   auto stream = c10::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(transpose_kernel, grid, dim3(256), 0, stream.stream(),
-                     (const short*)x.data_ptr(), (short*)out.data_ptr(),
-                     R, C);
+  hipLaunchKernelGGL(transpose_kernel<false>, grid, dim3(256), 0,
+                     stream.stream(), (const short*)x.data_ptr(),
+                     (short*)out.data_ptr(), nullptr, R, C);
   HIP_CHECK_LAST();
   return out;
+}
+
+// transpose + fp32 column sums (dy -> dy^T and bias grad in one pass)
+std::vector<torch::Tensor> transpose2d_colsum(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  const int R = x.size(0), C = x.size(1);
+  auto out = torch::empty({C, R}, x.options());
+  auto cs = torch::zeros({C}, x.options().dtype(torch::kFloat32));
+  dim3 grid((C + TT - 1) / TT, (R + TT - 1) / TT, 1);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(transpose_kernel<true>, grid, dim3(256), 0,
+                     stream.stream(), (const short*)x.data_ptr(),
+                     (short*)out.data_ptr(), cs.data_ptr<float>(), R, C);
+  HIP_CHECK_LAST();
+  return {out, cs};
 }

@@ -63,10 +63,14 @@ class _MFMALinearFn(torch.autograd.Function):
         dx = _gemm_nt(ext, dy, wt)              # [M, K]
         dyp = _pad_rows64(dy)
         xp = _pad_rows64(x2d)
-        dyt = ext.transpose2d(dyp)              # [N, Mp]
+        if ctx.has_bias:
+            dyt, db32 = ext.transpose2d_colsum(dyp)  # [N, Mp] + colsums
+            db = db32.to(dy.dtype)
+        else:
+            dyt = ext.transpose2d(dyp)
+            db = None
         xt = ext.transpose2d(xp)                # [K, Mp]
         dw = _gemm_nt(ext, dyt, xt)             # [N, K]
-        db = dy.sum(dim=0) if ctx.has_bias else None
         return dx, dw, db, None
 
 
