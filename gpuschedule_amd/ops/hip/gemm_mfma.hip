@@ -143,24 +143,51 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt_kernel(
     __syncthreads();
   }
 
-  // Epilogue: C row = bm0+wr+i*16+(lane>>4)*4+r, col = bn0+wc+j*16+(lane&15)
-  const int crow0 = bm0 + wr + (lane >> 4) * 4;
-  const int ccol0 = bn0 + wc + (lane & 15);
+  // Epilogue via LDS bounce: the MFMA C-layout's natural stores are
+  // 2-B per lane (32-B transactions, 4x wasted store bandwidth), so
+  // each wave writes its fragments into a linear LDS image of the
+  // 128x128 tile and the block streams it out in coalesced 16-B rows.
+  __syncthreads();
+  short* c_lds = (short*)lds;  // [BM][BN] bf16 = 32 KiB
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int col = ccol0 + j * 16;
-      if (ntail && col >= N) continue;
-      float bv = bias ? bf16_to_f32(bias[col]) : 0.f;
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = crow0 + i * 16 + r;
-        if (mtail && row >= M) continue;
-        float v = acc[i][j][r] + bv;
-        if (relu) v = fmaxf(v, 0.f);
-        C[(long)row * N + col] = f32_to_bf16(v);
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN + col] = f32_to_bf16(v);
+        }
       }
+  }
+  __syncthreads();
+  // 128 rows x 256 B: 2048 16-B chunks over 256 threads
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int idx = p * NTHREADS + tid;
+    int row = idx >> 4;
+    int colb = (idx & 15) * 8;       // 8 bf16 per chunk
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if ((mtail && grow >= M) || (ntail && gcol >= N)) {
+      if (!mtail || grow < M) {
+        for (int j = 0; j < 8 && gcol + j < N; ++j)
+          C[(long)grow * N + gcol + j] = c_lds[row * BN + colb + j];
+      }
+      continue;
+    }
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN + colb + j];
+    } else {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN + colb);
     }
   }
 }
@@ -285,23 +312,42 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_kernel(
 
   const bool mtail = bm0 + BM2 > M;
   const bool ntail = bn0 + BN2 > N;
-  const int crow0 = bm0 + wr + (lane >> 4) * 4;
-  const int ccol0 = bn0 + wc + (lane & 15);
+  // LDS-bounce epilogue (same rationale as the 128-tile kernel): the
+  // 256x256 bf16 tile exactly fills the 128 KiB LDS.
+  __syncthreads();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < 8; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int col = ccol0 + j * 16;
-      if (ntail && col >= N) continue;
-      float bv = bias ? bf16_to_f32(bias[col]) : 0.f;
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = crow0 + i * 16 + r;
-        if (mtail && row >= M) continue;
-        float v = acc[i][j][r] + bv;
-        if (relu) v = fmaxf(v, 0.f);
-        C[(long)row * N + col] = f32_to_bf16(v);
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
       }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
     }
   }
 }
