@@ -56,12 +56,14 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
   constexpr int DF = D / 16;       // PV col fragments
   constexpr int KC = D / 32;       // QK^T k-steps over D
   constexpr int ROWB = D * 2;      // K tile LDS row bytes
-  // LDS: K tile [KVB][D] + Vt tile [D][KVB] + 4 wave-private P tiles
+  // LDS: double-buffered K [KVB][D] + Vt [D][KVB] + 4 wave-private P
   __shared__ __attribute__((aligned(16))) char lds[
-      KVB * D * 2 + D * KVB * 2 + 4 * 16 * KVB * 2];
-  char* k_lds = lds;
-  char* v_lds = lds + KVB * D * 2;
-  char* p_lds = lds + 2 * KVB * D * 2;
+      2 * (KVB * D * 2 + D * KVB * 2) + 4 * 16 * KVB * 2];
+  char* k_cur = lds;
+  char* v_cur = lds + KVB * D * 2;
+  char* k_nxt = lds + 2 * KVB * D * 2;
+  char* v_nxt = lds + 3 * KVB * D * 2;
+  char* p_lds = lds + 4 * KVB * D * 2;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -110,51 +112,70 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
       ? min((q_tile_end / KVB) + 1, (S + KVB - 1) / KVB)
       : (S + KVB - 1) / KVB;
 
-  for (int t = 0; t < n_kv; ++t) {
-    const int kv0 = t * KVB;
-    // ---- stage K tile [KVB][D] and Vt tile [D][KVB] -------------
-    {
-      // K: KVB*D*2 bytes, 16B pieces: piece p -> row r, chunk c
-      constexpr int PIECES_K = KVB * D * 2 / (NTHREADS * 16);
-      constexpr int CPR = D * 2 / 16;   // 16B chunks per K row
+  // ---- staging helpers (register path: T14 async-split) ---------
+  constexpr int PIECES_K = KVB * D * 2 / (NTHREADS * 16);
+  constexpr int CPR = D * 2 / 16;     // 16B chunks per K row
+  constexpr int PIECES_V = D * KVB * 2 / (NTHREADS * 16);
+  constexpr int CPRV = KVB * 2 / 16;
+  short8 kst[PIECES_K], vst[PIECES_V];
+
+  auto stage_load = [&](int kv0) {
 #pragma unroll
-      for (int p = 0; p < PIECES_K; ++p) {
-        int idx = p * NTHREADS + tid;
-        int r = idx / CPR, cb = (idx % CPR) * 16;
-        int kvr = kv0 + r;
-        short8 val;
-        if (kvr < S) {
-          val = *reinterpret_cast<const short8*>(
-              K + qoff + (long)kvr * D + cb / 2);
-        } else {
+    for (int p = 0; p < PIECES_K; ++p) {
+      int idx = p * NTHREADS + tid;
+      int r = idx / CPR, cb = (idx % CPR) * 16;
+      int kvr = kv0 + r;
+      if (kvr < S) {
+        kst[p] = *reinterpret_cast<const short8*>(
+            K + qoff + (long)kvr * D + cb / 2);
+      } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) val[j] = 0;
-        }
-        *reinterpret_cast<short8*>(k_lds + swz(r, cb, ROWB)) = val;
-      }
-      // Vt: D rows x KVB cols
-      constexpr int PIECES_V = D * KVB * 2 / (NTHREADS * 16);
-      constexpr int CPRV = KVB * 2 / 16;
-#pragma unroll
-      for (int p = 0; p < PIECES_V; ++p) {
-        int idx = p * NTHREADS + tid;
-        int r = idx / CPRV, cb = (idx % CPRV) * 16;
-        short8 val;
-        int kvc = kv0 + cb / 2;
-        if (kvc + 7 < S) {
-          val = *reinterpret_cast<const short8*>(
-              Vt + voff + (long)r * S + kvc);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            val[j] = (kvc + j < S)
-                ? (Vt + voff + (long)r * S)[kvc + j] : (short)0;
-        }
-        *reinterpret_cast<short8*>(
-            v_lds + swz(r, cb, KVB * 2)) = val;
+        for (int j = 0; j < 8; ++j) kst[p][j] = 0;
       }
     }
-    __syncthreads();
+#pragma unroll
+    for (int p = 0; p < PIECES_V; ++p) {
+      int idx = p * NTHREADS + tid;
+      int r = idx / CPRV, cb = (idx % CPRV) * 16;
+      int kvc = kv0 + cb / 2;
+      if (kvc + 7 < S) {
+        vst[p] = *reinterpret_cast<const short8*>(
+            Vt + voff + (long)r * S + kvc);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          vst[p][j] = (kvc + j < S)
+              ? (Vt + voff + (long)r * S)[kvc + j] : (short)0;
+      }
+    }
+  };
+  auto stage_write = [&](char* kbuf, char* vbuf) {
+#pragma unroll
+    for (int p = 0; p < PIECES_K; ++p) {
+      int idx = p * NTHREADS + tid;
+      int r = idx / CPR, cb = (idx % CPR) * 16;
+      *reinterpret_cast<short8*>(kbuf + swz(r, cb, ROWB)) = kst[p];
+    }
+#pragma unroll
+    for (int p = 0; p < PIECES_V; ++p) {
+      int idx = p * NTHREADS + tid;
+      int r = idx / CPRV, cb = (idx % CPRV) * 16;
+      *reinterpret_cast<short8*>(vbuf + swz(r, cb, KVB * 2)) = vst[p];
+    }
+  };
+
+  stage_load(0);
+  stage_write(k_cur, v_cur);
+  __syncthreads();
+
+  for (int t = 0; t < n_kv; ++t) {
+    const int kv0 = t * KVB;
+    char* k_lds = k_cur;
+    char* v_lds = v_cur;
+    // T14 async-split: issue next tile's global loads NOW; their
+    // HBM latency hides under QK^T + softmax + PV; the ds_write's
+    // implicit wait lands just before the end-of-tile barrier.
+    if (t + 1 < n_kv) stage_load((t + 1) * KVB);
 
     // ---- S = QK^T: 4 col-frags x KC k-steps ---------------------
     float4v s_acc[4];
@@ -237,6 +258,11 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
         o_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             pf, vf, o_acc[f], 0, 0, 0);
       }
+    }
+    if (t + 1 < n_kv) {
+      stage_write(k_nxt, v_nxt);
+      char* tmp = k_cur; k_cur = k_nxt; k_nxt = tmp;
+      tmp = v_cur; v_cur = v_nxt; v_nxt = tmp;
     }
     __syncthreads();
   }

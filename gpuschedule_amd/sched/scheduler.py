@@ -56,6 +56,7 @@ class Scheduler:
         self.log = log if log is not None else EventLog()
         self.jobs: Dict[str, Job] = {}
         self._pending: List[JobSpec] = []   # sorted by submit_time
+        self._ticks = 0
 
     # ---- submission ---------------------------------------------
 
@@ -120,6 +121,13 @@ class Scheduler:
         alloc, evictions = self._allocate(requests)
         plan = self._diff(alloc, evictions, now)
         self._apply(plan, now)
+        self._ticks += 1
+        if self._ticks % 16 == 1:  # periodic cluster utilization sample
+            self.log.emit(now, "tick",
+                          util=round(self.cluster.utilization(), 3),
+                          running=len(self.cluster.running_jobs()),
+                          queued=sum(1 for j in self.jobs.values()
+                                     if j.state == JobState.QUEUED))
         return plan
 
     # ---- allocation ---------------------------------------------
