@@ -110,7 +110,8 @@ class JobRunner:
         return x, y
 
     def _eager_iter(self, x, y):
-        self.optim.zero_grad(set_to_none=True)
+        if not getattr(self.optim, "fused_zero_grad", False):
+            self.optim.zero_grad(set_to_none=True)
         loss = compute_loss(self.model_name, self.model, x, y)
         loss.backward()
         self.reducer.finalize()

@@ -33,6 +33,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   double momentum, double eps, bool relu);
 void sgd_step_fp32(torch::Tensor p, torch::Tensor grad, torch::Tensor buf,
                    double lr, double momentum, double wd);
+void adamw_multi(torch::Tensor ptrs, torch::Tensor sizes,
+                 torch::Tensor chunk_t, torch::Tensor chunk_off,
+                 torch::Tensor stepbuf, double lr, double beta1,
+                 double beta2, double eps, double wd);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits,
                                   torch::Tensor labels);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
@@ -56,6 +60,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, "flash-attention backward (FA2 two-kernel)");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward");
+  m.def("adamw_multi", &adamw_multi, "multi-tensor fused AdamW (one launch, zeroes grads)");
   m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");
   m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");
   m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+ReLU) backward, channels_last");

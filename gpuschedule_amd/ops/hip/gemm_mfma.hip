@@ -284,14 +284,12 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_kernel(
           bf[fj][kk] = *reinterpret_cast<const bf16x8*>(
               Bc + swz(wc + jc * 32 + fj * 16 + arow, kk * 64 + kb));
       }
-      if (pre && ph == 0) {
-        // issue the WHOLE next K-tile's LDS-DMA up front: the loads
-        // get ~4 MFMA phases (~1k cycles) of cover before the
-        // end-of-tile drain instead of 1 phase for the last half
-        glds_stage_half(A, K, bm0, M, k1, An, 0, wid, lane);
-        glds_stage_half(A, K, bm0, M, k1, An, 1, wid, lane);
-        glds_stage_half(B, K, bn0, N, k1, Bn, 0, wid, lane);
-        glds_stage_half(B, K, bn0, N, k1, Bn, 1, wid, lane);
+      if (pre) {
+        // one half-tile of the next K-tile per phase: A0 A1 B0 B1
+        // (issuing all four at phase 0 measured WORSE at 8192^3:
+        // the burst serializes against phase 0's ds_reads)
+        if (ph < 2) glds_stage_half(A, K, bm0, M, k1, An, ph, wid, lane);
+        else glds_stage_half(B, K, bn0, N, k1, Bn, ph - 2, wid, lane);
       }
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);

@@ -75,6 +75,56 @@ __global__ __launch_bounds__(NT) void sgd_kernel(
   }
 }
 
+// Multi-tensor AdamW: ONE launch updates every bf16 parameter of the
+// optimizer (vs one launch per tensor).  A chunk table (built once —
+// pointers are stable across steps and hipGraph replays) maps blocks
+// to (tensor, offset) pieces; the kernel also ZEROES the consumed
+// grad, so no zero_grad pass exists at all and grad tensors keep
+// their identity for graph capture.
+constexpr int MT_CHUNK = 8192;  // elements per table entry
+
+// device-side step counter + bias corrections: host-passed values
+// would be frozen by hipGraph capture; sb = [step, bc1, bc2] f32.
+__global__ void adamw_mt_prestep_kernel(float* sb, float beta1,
+                                        float beta2) {
+  float st = sb[0] + 1.f;
+  sb[0] = st;
+  sb[1] = 1.f - powf(beta1, st);
+  sb[2] = 1.f - powf(beta2, st);
+}
+
+__global__ __launch_bounds__(NT) void adamw_mt_kernel(
+    const long* __restrict__ ptrs,   // [T][5]: p,p32,g,m,v
+    const long* __restrict__ sizes,  // [T]
+    const int* __restrict__ chunk_t, const long* __restrict__ chunk_off,
+    int n_chunks, float lr, float beta1, float beta2, float eps,
+    float wd, const float* __restrict__ sb) {
+  const float bc1 = sb[1], bc2 = sb[2];
+  for (int c = blockIdx.x; c < n_chunks; c += gridDim.x) {
+    const int t = chunk_t[c];
+    const long off = chunk_off[c];
+    short* p = (short*)ptrs[t * 5 + 0] + off;
+    float* p32 = (float*)ptrs[t * 5 + 1] + off;
+    short* g = (short*)ptrs[t * 5 + 2] + off;
+    float* m = (float*)ptrs[t * 5 + 3] + off;
+    float* v = (float*)ptrs[t * 5 + 4] + off;
+    const long n = min((long)MT_CHUNK, sizes[t] - off);
+    for (long i = threadIdx.x * 4; i < n; i += NT * 4) {
+      long lim = min(i + 4, n);
+      for (long k = i; k < lim; ++k) {
+        float gg = bf16_to_f32(g[k]);
+        float mm = beta1 * m[k] + (1.f - beta1) * gg;
+        float vv = beta2 * v[k] + (1.f - beta2) * gg * gg;
+        float pw = p32[k];
+        pw -= lr * ((mm / bc1) / (sqrtf(vv / bc2) + eps) + wd * pw);
+        m[k] = mm; v[k] = vv; p32[k] = pw;
+        p[k] = f32_to_bf16(pw);
+        g[k] = 0;  // fused zero_grad (backward accumulates from 0)
+      }
+    }
+  }
+}
+
 // fp32-native SGD (BatchNorm affine params train in fp32)
 __global__ __launch_bounds__(NT) void sgd_fp32_kernel(
     float* __restrict__ p, const float* __restrict__ grad,
@@ -96,6 +146,26 @@ int grid_for(long n) {
 }
 
 }  // namespace
+
+void adamw_multi(torch::Tensor ptrs, torch::Tensor sizes,
+                 torch::Tensor chunk_t, torch::Tensor chunk_off,
+                 torch::Tensor stepbuf, double lr, double beta1,
+                 double beta2, double eps, double wd) {
+  TORCH_CHECK(ptrs.is_cuda() && ptrs.scalar_type() == torch::kLong);
+  int n_chunks = chunk_t.numel();
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(adamw_mt_prestep_kernel, dim3(1), dim3(1), 0,
+                     stream.stream(), stepbuf.data_ptr<float>(),
+                     (float)beta1, (float)beta2);
+  int grid = std::min(n_chunks, 2048);
+  hipLaunchKernelGGL(adamw_mt_kernel, dim3(grid), dim3(NT), 0,
+                     stream.stream(), ptrs.data_ptr<long>(),
+                     sizes.data_ptr<long>(), chunk_t.data_ptr<int>(),
+                     chunk_off.data_ptr<long>(), n_chunks, (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, (float)wd,
+                     stepbuf.data_ptr<float>());
+  HIP_CHECK_LAST();
+}
 
 void sgd_step_fp32(torch::Tensor p, torch::Tensor grad, torch::Tensor buf,
                    double lr, double momentum, double wd) {

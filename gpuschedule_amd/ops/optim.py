@@ -35,11 +35,70 @@ class _MasterStateMixin:
 
 
 class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
+    """Fused AdamW.  When every param of a group is a bf16 CUDA tensor
+    with a grad, the MULTI-TENSOR path runs: ONE kernel updates all
+    params AND zeroes the grads in place (``fused_zero_grad`` tells
+    the trainer to skip zero_grad, which keeps grad tensors stable —
+    required for the chunk table and for hipGraph capture).  The step
+    counter/bias corrections live in a device buffer so graph replays
+    keep advancing them."""
+
+    MT_CHUNK = 8192
+
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=0.01):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self.fused_zero_grad = False
+        self._mt = None  # (ptrs, sizes, chunk_t, chunk_off, stepbuf, params)
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        self._mt = None  # state tensors replaced: table is stale
+        self.fused_zero_grad = False
+
+    def state_dict(self):
+        # sync host step mirrors from the device counter (python-side
+        # step() does not run under hipGraph replay)
+        if self._mt is not None:
+            step = int(self._mt[4][0].item())
+            for p in self._mt[5]:
+                self.state[p]["step"] = step
+        return super().state_dict()
+
+    def _try_build_mt(self, group):
+        params = [p for p in group["params"] if p.requires_grad]
+        if not params or not all(
+                p.is_cuda and p.dtype == torch.bfloat16
+                and p.grad is not None and p.grad.is_contiguous()
+                and p.is_contiguous() for p in params):
+            return None
+        dev = params[0].device
+        ptrs, sizes, ct, co = [], [], [], []
+        step0 = None
+        for ti, p in enumerate(params):
+            st = self.state[p]
+            ptrs.append([p.data.data_ptr(), st["p32"].data_ptr(),
+                         p.grad.data_ptr(), st["m"].data_ptr(),
+                         st["v"].data_ptr()])
+            n = p.numel()
+            sizes.append(n)
+            step0 = st["step"] if step0 is None else step0
+            if st["step"] != step0:
+                return None  # inconsistent restore; stay per-tensor
+            off = 0
+            while off < n:
+                ct.append(ti)
+                co.append(off)
+                off += self.MT_CHUNK
+        sb = torch.zeros(3, dtype=torch.float32, device=dev)
+        sb[0] = float(step0)
+        return (torch.tensor(ptrs, dtype=torch.int64, device=dev),
+                torch.tensor(sizes, dtype=torch.int64, device=dev),
+                torch.tensor(ct, dtype=torch.int32, device=dev),
+                torch.tensor(co, dtype=torch.int64, device=dev),
+                sb, params)
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -48,6 +107,7 @@ class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
             lr = group["lr"]
             beta1, beta2 = group["betas"]
             eps, wd = group["eps"], group["weight_decay"]
+            # lazy state init for all params with grads
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -57,6 +117,21 @@ class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
                     state["p32"] = p.detach().float().clone()
                     state["m"] = torch.zeros_like(state["p32"])
                     state["v"] = torch.zeros_like(state["p32"])
+            # multi-tensor fast path
+            if self._mt is None and len(self.param_groups) == 1:
+                self._mt = self._try_build_mt(group)
+                if self._mt is not None:
+                    self.fused_zero_grad = True
+            if self._mt is not None:
+                ext = get_ext(required=True)
+                ptrs, sizes, ct, co, sb, params = self._mt
+                ext.adamw_multi(ptrs, sizes, ct, co, sb, lr, beta1,
+                                beta2, eps, wd)
+                return loss
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
                 state["step"] += 1
                 t = state["step"]
                 if p.is_cuda and p.dtype == torch.bfloat16:
