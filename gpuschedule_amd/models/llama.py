@@ -17,6 +17,7 @@ import torch.nn.functional as F
 
 from ..ops.attention import flash_attention
 from ..ops.ce import fused_cross_entropy
+from ..ops.elementwise import fused_rope, silu_mul
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedRMSNorm
 
@@ -43,23 +44,14 @@ LLAMA3_TINY = LlamaConfig(vocab_size=1024, hidden=512, layers=4, heads=8,
                           rope_theta=10000.0)
 
 
-def rope_table(cfg: LlamaConfig, device, dtype):
+def rope_table(cfg: LlamaConfig, device, dtype=None):
+    # fp32 tables (the fused gfx950 RoPE kernel reads them directly)
     d = cfg.head_dim
     inv = 1.0 / (cfg.rope_theta **
                  (torch.arange(0, d, 2, dtype=torch.float32) / d))
     t = torch.arange(cfg.max_seq, dtype=torch.float32)
     freqs = torch.outer(t, inv)
-    return (freqs.cos().to(device=device, dtype=dtype),
-            freqs.sin().to(device=device, dtype=dtype))
-
-
-def apply_rope(x, cos, sin):
-    # x: [B, H, S, D]; cos/sin: [S, D/2]
-    S = x.shape[-2]
-    c = cos[:S][None, None]
-    s = sin[:S][None, None]
-    x1, x2 = x.chunk(2, dim=-1)
-    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+    return freqs.cos().to(device=device), freqs.sin().to(device=device)
 
 
 class LlamaLayer(nn.Module):
@@ -86,15 +78,15 @@ class LlamaLayer(nn.Module):
         q = q.view(B, S, cfg.heads, hd).transpose(1, 2)
         k = k.view(B, S, cfg.kv_heads, hd).transpose(1, 2)
         v = v.view(B, S, cfg.kv_heads, hd).transpose(1, 2)
-        q = apply_rope(q, cos, sin)
-        k = apply_rope(k, cos, sin)
+        q = fused_rope(q, cos, sin)
+        k = fused_rope(k, cos, sin)
         att = flash_attention(q, k, v, causal=True)
         att = att.transpose(1, 2).reshape(B, S, cfg.hidden)
         x = x + self.o(att)
         n = self.norm2(x)
         gu = self.gate_up(n)
         g, u = gu.chunk(2, dim=-1)
-        x = x + self.down(F.silu(g) * u)
+        x = x + self.down(silu_mul(g, u))
         return x
 
 
@@ -111,9 +103,9 @@ class LlamaLM(nn.Module):
         self.register_buffer("rope_cos", None, persistent=False)
         self.register_buffer("rope_sin", None, persistent=False)
 
-    def _rope(self, device, dtype):
+    def _rope(self, device, dtype=None):
         if self.rope_cos is None or self.rope_cos.device != device:
-            cos, sin = rope_table(self.cfg, device, dtype)
+            cos, sin = rope_table(self.cfg, device)
             self.rope_cos, self.rope_sin = cos, sin
         return self.rope_cos, self.rope_sin
 

@@ -16,7 +16,8 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 BUILD_DIR = os.path.join(HERE, "_build")
 SOURCES = [os.path.join(HERE, "hip", f)
            for f in ("bindings.cpp", "norms.hip", "gemm_mfma.hip",
-                     "optim.hip", "attention.hip", "bn.hip", "ce.hip")]
+                     "optim.hip", "attention.hip", "bn.hip", "ce.hip",
+                     "elementwise.hip")]
 EXT_NAME = "_hip_ops"
 
 

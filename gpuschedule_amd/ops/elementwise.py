@@ -1,0 +1,54 @@
+"""Fused RoPE and SwiGLU-gate wrappers (ops/hip/elementwise.hip)."""
+from __future__ import annotations
+
+import torch
+
+from ._ext import get_ext
+
+
+class _RopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cosb, sinb):
+        ext = get_ext(required=True)
+        ctx.save_for_backward(cosb, sinb)
+        return ext.rope_apply(x, cosb, sinb, False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        cosb, sinb = ctx.saved_tensors
+        return ext.rope_apply(dy.contiguous(), cosb, sinb, True), \
+            None, None
+
+
+def fused_rope(x: torch.Tensor, cosb: torch.Tensor,
+               sinb: torch.Tensor) -> torch.Tensor:
+    """Rotate-half RoPE on [B, H, S, D] bf16 with fp32 [S, D/2] tables."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _RopeFn.apply(x.contiguous(), cosb, sinb)
+    S = x.shape[-2]
+    c = cosb[:S].to(x.dtype)[None, None]
+    s = sinb[:S].to(x.dtype)[None, None]
+    x1, x2 = x.chunk(2, dim=-1)
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+
+
+class _SiluMulFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g, u):
+        ext = get_ext(required=True)
+        ctx.save_for_backward(g, u)
+        return ext.silu_mul_fwd(g, u)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        g, u = ctx.saved_tensors
+        dg, du = ext.silu_mul_bwd(dy.contiguous(), g, u)
+        return dg, du
+
+
+def silu_mul(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
+    if g.is_cuda and g.dtype == torch.bfloat16:
+        return _SiluMulFn.apply(g.contiguous(), u.contiguous())
+    return torch.nn.functional.silu(g) * u

@@ -37,6 +37,11 @@ void adamw_multi(torch::Tensor ptrs, torch::Tensor sizes,
                  torch::Tensor chunk_t, torch::Tensor chunk_off,
                  torch::Tensor stepbuf, double lr, double beta1,
                  double beta2, double eps, double wd);
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosb,
+                         torch::Tensor sinb, bool backward);
+torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor g,
+                                        torch::Tensor u);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits,
                                   torch::Tensor labels);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
@@ -58,6 +63,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
   m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax); returns (o, lse)");
   m.def("attn_bwd", &attn_bwd, "flash-attention backward (FA2 two-kernel)");
+  m.def("rope_apply", &rope_apply, "fused rotate-half RoPE fwd/bwd");
+  m.def("silu_mul_fwd", &silu_mul_fwd, "fused SwiGLU gate fwd");
+  m.def("silu_mul_bwd", &silu_mul_bwd, "fused SwiGLU gate bwd");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward");
   m.def("adamw_multi", &adamw_multi, "multi-tensor fused AdamW (one launch, zeroes grads)");

@@ -1,0 +1,167 @@
+// Fused elementwise kernels for the Llama path on gfx950:
+//  * RoPE (rotate-half) forward/backward from a host-precomputed
+//    fp32 cos/sin table (guide: no on-device trig) — replaces ~8
+//    torch ops (chunk/mul/cat) per projection per direction.
+//  * SwiGLU gate: y = silu(g) * u, with fused backward — replaces the
+//    silu/mul kernel pair and their backward chain.
+// All memory-bound: bf16x8 vectorized, grid-stride.
+#include <torch/extension.h>
+#include <c10/cuda/CUDAStream.h>
+#include "common.h"
+
+namespace {
+
+constexpr int NT = 256;
+
+// x [rows, D] where row r maps to sequence position pos = r % S
+// (rows are [B,H,S] flattened).  cos/sin: [S, D/2] fp32.
+template <bool BWD>
+__global__ __launch_bounds__(NT) void rope_kernel(
+    const short* __restrict__ x, const float* __restrict__ cosb,
+    const float* __restrict__ sinb, short* __restrict__ out,
+    long rows, int S, int D) {
+  const int HD = D / 2;
+  const int chunks = HD / 8;  // 8-wide pieces per half-row
+  long total = rows * chunks;
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  long stride = (long)gridDim.x * NT;
+  for (; i < total; i += stride) {
+    long row = i / chunks;
+    int c8 = (int)(i % chunks) * 8;
+    int pos = (int)(row % S);
+    const short* xr = x + row * D;
+    short* orow = out + row * D;
+    short8 x1 = *reinterpret_cast<const short8*>(xr + c8);
+    short8 x2 = *reinterpret_cast<const short8*>(xr + HD + c8);
+    const float* cr = cosb + (long)pos * HD + c8;
+    const float* sr = sinb + (long)pos * HD + c8;
+    short8 o1, o2;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float a = bf16_to_f32(x1[j]), b = bf16_to_f32(x2[j]);
+      float cc = cr[j], ss = sr[j];
+      if (!BWD) {
+        o1[j] = f32_to_bf16(a * cc - b * ss);
+        o2[j] = f32_to_bf16(b * cc + a * ss);
+      } else {  // transpose of the rotation
+        o1[j] = f32_to_bf16(a * cc + b * ss);
+        o2[j] = f32_to_bf16(b * cc - a * ss);
+      }
+    }
+    *reinterpret_cast<short8*>(orow + c8) = o1;
+    *reinterpret_cast<short8*>(orow + HD + c8) = o2;
+  }
+}
+
+__global__ __launch_bounds__(NT) void silu_mul_fwd_kernel(
+    const short* __restrict__ g, const short* __restrict__ u,
+    short* __restrict__ y, long n) {
+  long i = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * NT * 8;
+  for (; i + 7 < n; i += stride) {
+    short8 gv = *reinterpret_cast<const short8*>(g + i);
+    short8 uv = *reinterpret_cast<const short8*>(u + i);
+    short8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gg = bf16_to_f32(gv[j]);
+      float sig = 1.f / (1.f + __expf(-gg));
+      out[j] = f32_to_bf16(gg * sig * bf16_to_f32(uv[j]));
+    }
+    *reinterpret_cast<short8*>(y + i) = out;
+  }
+}
+
+__global__ __launch_bounds__(NT) void silu_mul_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ g,
+    const short* __restrict__ u, short* __restrict__ dg,
+    short* __restrict__ du, long n) {
+  long i = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * NT * 8;
+  for (; i + 7 < n; i += stride) {
+    short8 dv = *reinterpret_cast<const short8*>(dy + i);
+    short8 gv = *reinterpret_cast<const short8*>(g + i);
+    short8 uv = *reinterpret_cast<const short8*>(u + i);
+    short8 og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float d = bf16_to_f32(dv[j]);
+      float gg = bf16_to_f32(gv[j]);
+      float sig = 1.f / (1.f + __expf(-gg));
+      float si = gg * sig;
+      ou[j] = f32_to_bf16(d * si);
+      og[j] = f32_to_bf16(d * bf16_to_f32(uv[j]) *
+                          (sig * (1.f + gg * (1.f - sig))));
+    }
+    *reinterpret_cast<short8*>(dg + i) = og;
+    *reinterpret_cast<short8*>(du + i) = ou;
+  }
+}
+
+int grid_for_n(long items) {
+  return (int)std::min<long>((items + NT - 1) / NT, 2048);
+}
+
+}  // namespace
+
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosb,
+                         torch::Tensor sinb, bool backward) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(cosb.scalar_type() == torch::kFloat32);
+  const int D = x.size(-1);
+  const int S = x.size(-2);
+  TORCH_CHECK(D % 16 == 0, "rope: D must be a multiple of 16");
+  TORCH_CHECK(cosb.size(-1) == D / 2 && cosb.size(0) >= S);
+  long rows = x.numel() / D;
+  auto out = torch::empty_like(x);
+  long total = rows * (D / 16);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  if (backward) {
+    hipLaunchKernelGGL(rope_kernel<true>, dim3(grid_for_n(total)),
+                       dim3(NT), 0, stream.stream(),
+                       (const short*)x.data_ptr(),
+                       cosb.data_ptr<float>(), sinb.data_ptr<float>(),
+                       (short*)out.data_ptr(), rows, S, D);
+  } else {
+    hipLaunchKernelGGL(rope_kernel<false>, dim3(grid_for_n(total)),
+                       dim3(NT), 0, stream.stream(),
+                       (const short*)x.data_ptr(),
+                       cosb.data_ptr<float>(), sinb.data_ptr<float>(),
+                       (short*)out.data_ptr(), rows, S, D);
+  }
+  HIP_CHECK_LAST();
+  return out;
+}
+
+torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u) {
+  TORCH_CHECK(g.is_cuda() && g.is_contiguous() && u.is_contiguous());
+  TORCH_CHECK(g.scalar_type() == torch::kBFloat16);
+  long n = g.numel();
+  TORCH_CHECK(n % 8 == 0);
+  auto y = torch::empty_like(g);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(silu_mul_fwd_kernel, dim3(grid_for_n(n / 8)),
+                     dim3(NT), 0, stream.stream(),
+                     (const short*)g.data_ptr(),
+                     (const short*)u.data_ptr(), (short*)y.data_ptr(),
+                     n);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor g,
+                                        torch::Tensor u) {
+  long n = g.numel();
+  auto dg = torch::empty_like(g);
+  auto du = torch::empty_like(u);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(silu_mul_bwd_kernel, dim3(grid_for_n(n / 8)),
+                     dim3(NT), 0, stream.stream(),
+                     (const short*)dy.data_ptr(),
+                     (const short*)g.data_ptr(),
+                     (const short*)u.data_ptr(), (short*)dg.data_ptr(),
+                     (short*)du.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return {dg, du};
+}

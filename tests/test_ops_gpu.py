@@ -314,3 +314,49 @@ def test_multi_tensor_adamw_matches_torch():
     # grads were zeroed by the kernel
     for p in a.parameters():
         assert p.grad is not None and p.grad.abs().max().item() == 0
+
+
+# ---- fused RoPE / SwiGLU gate -------------------------------------
+
+def test_fused_rope_fwd_bwd():
+    from gpuschedule_amd.ops.elementwise import fused_rope
+    torch.manual_seed(14)
+    B, H, S, D = 2, 4, 128, 64
+    inv = 1.0 / (10000 ** (torch.arange(0, D, 2).float() / D))
+    t = torch.arange(S).float()
+    fr = torch.outer(t, inv)
+    cosb = fr.cos().cuda()
+    sinb = fr.sin().cuda()
+    x = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = fused_rope(x, cosb, sinb)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.detach().float().requires_grad_(True)
+    c = cosb[None, None]
+    s = sinb[None, None]
+    x1, x2 = xf.chunk(2, dim=-1)
+    ref = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+    ref.backward(dy.float())
+    assert (y.float() - ref).abs().max().item() < 2e-2
+    assert (x.grad.float() - xf.grad).abs().max().item() < 2e-2
+
+
+def test_fused_silu_mul():
+    from gpuschedule_amd.ops.elementwise import silu_mul
+    torch.manual_seed(15)
+    g = torch.randn(1024, 512, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    u = torch.randn_like(g).requires_grad_(True)
+    y = silu_mul(g, u)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    gf = g.detach().float().requires_grad_(True)
+    uf = u.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.silu(gf) * uf
+    ref.backward(dy.float())
+    assert (y.float() - ref).abs().max().item() < 2e-2
+    sc = gf.grad.abs().max().item()
+    assert (g.grad.float() - gf.grad).abs().max().item() / sc < 2e-2
+    assert (u.grad.float() - uf.grad).abs().max().item() < 2e-2
