@@ -47,9 +47,9 @@ class BertLayer(nn.Module):
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
         att = flash_attention(q, k, v, causal=False)
         att = att.transpose(1, 2).reshape(B, S, Hd)
-        x = self.ln1(x + self.proj(att))
+        x = self.ln1(self.proj(att), residual=x)  # fused add+LN
         h = F.gelu(self.fc1(x), approximate="tanh")
-        x = self.ln2(x + self.fc2(h))
+        x = self.ln2(self.fc2(h), residual=x)
         return x
 
 

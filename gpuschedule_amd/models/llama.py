@@ -82,8 +82,8 @@ class LlamaLayer(nn.Module):
         k = fused_rope(k, cos, sin)
         att = flash_attention(q, k, v, causal=True)
         att = att.transpose(1, 2).reshape(B, S, cfg.hidden)
-        x = x + self.o(att)
-        n = self.norm2(x)
+        # fused residual-add + RMSNorm; summed stream comes back out
+        n, x = self.norm2(self.o(att), residual=x, return_sum=True)
         gu = self.gate_up(n)
         g, u = gu.chunk(2, dim=-1)
         x = x + self.down(silu_mul(g, u))

@@ -1,7 +1,9 @@
 // Python bindings for the gfx950 kernel library.
 #include <torch/extension.h>
 
-std::vector<torch::Tensor> norm_fwd(torch::Tensor x, torch::Tensor gamma,
+std::vector<torch::Tensor> norm_fwd(torch::Tensor x,
+                                    c10::optional<torch::Tensor> residual,
+                                    torch::Tensor gamma,
                                     c10::optional<torch::Tensor> beta,
                                     double eps, bool rms);
 std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,

@@ -20,9 +20,10 @@ constexpr int CHUNK = NT * VEC;  // elements covered per chunk pass
 // ---------------- forward ------------------------------------------
 // one block per row; NCHUNK = ceil(D / 2048) template-dispatched so the
 // row stays in registers (runtime-indexed arrays go to scratch).
-template <int NCHUNK, bool RMS>
+template <int NCHUNK, bool RMS, bool RES>
 __global__ __launch_bounds__(NT) void norm_fwd_kernel(
-    const short* __restrict__ x, const short* __restrict__ gamma,
+    const short* __restrict__ x, const short* __restrict__ res,
+    short* __restrict__ sum_out, const short* __restrict__ gamma,
     const short* __restrict__ beta, short* __restrict__ y,
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
     int n_rows, int D, float eps) {
@@ -36,6 +37,18 @@ __global__ __launch_bounds__(NT) void norm_fwd_kernel(
       int i = c * CHUNK + threadIdx.x * VEC;
       if (i < D) {
         xv[c] = *reinterpret_cast<const short8*>(xr + i);
+        if (RES) {
+          // fused residual add; the summed activation IS the tensor
+          // backward normalizes against, so write it out
+          short8 rv = *reinterpret_cast<const short8*>(
+              res + (long)row * D + i);
+#pragma unroll
+          for (int j = 0; j < VEC; ++j)
+            xv[c][j] = f32_to_bf16(bf16_to_f32(xv[c][j]) +
+                                   bf16_to_f32(rv[j]));
+          *reinterpret_cast<short8*>(sum_out + (long)row * D + i) =
+              xv[c];
+        }
 #pragma unroll
         for (int j = 0; j < VEC; ++j) {
           float f = bf16_to_f32(xv[c][j]);
@@ -218,8 +231,13 @@ void check_norm_inputs(const torch::Tensor& x, const torch::Tensor& gamma) {
 
 }  // namespace
 
-// y, mean, rstd = layernorm_fwd(x, gamma, beta, eps); RMS: mean unused.
-std::vector<torch::Tensor> norm_fwd(torch::Tensor x, torch::Tensor gamma,
+// (y, sum_or_x, mean, rstd) = norm_fwd(x, residual?, gamma, beta?, ...)
+// With a residual the normalized tensor is x+residual and is written
+// out (backward normalizes against it; callers may reuse it as the
+// residual stream).  RMS: mean slot returns rstd.
+std::vector<torch::Tensor> norm_fwd(torch::Tensor x,
+                                    c10::optional<torch::Tensor> residual,
+                                    torch::Tensor gamma,
                                     c10::optional<torch::Tensor> beta,
                                     double eps, bool rms) {
   check_norm_inputs(x, gamma);
@@ -237,23 +255,37 @@ std::vector<torch::Tensor> norm_fwd(torch::Tensor x, torch::Tensor gamma,
     TORCH_CHECK(beta.has_value(), "layernorm needs beta");
     beta_p = (const short*)beta->data_ptr();
   }
+  const bool has_res = residual.has_value() && residual->defined();
+  torch::Tensor sum_out;
+  const short* res_p = nullptr;
+  short* sum_p = nullptr;
+  if (has_res) {
+    TORCH_CHECK(residual->is_contiguous() &&
+                residual->scalar_type() == torch::kBFloat16);
+    sum_out = torch::empty_like(x);
+    res_p = (const short*)residual->data_ptr();
+    sum_p = (short*)sum_out.data_ptr();
+  }
+#define NORM_FWD_LAUNCH(kNC, RMSB, RESB)                                  \
+  hipLaunchKernelGGL((norm_fwd_kernel<kNC, RMSB, RESB>), dim3(grid),      \
+                     dim3(NT), 0, stream.stream(),                        \
+                     (const short*)x.data_ptr(), res_p, sum_p,            \
+                     (const short*)gamma.data_ptr(), beta_p,              \
+                     (short*)y.data_ptr(),                                \
+                     RMSB ? nullptr : mean.data_ptr<float>(),             \
+                     rstd.data_ptr<float>(), (int)n_rows, D, (float)eps)
   DISPATCH_NCHUNK(nc, rms, {
     if (rms) {
-      hipLaunchKernelGGL((norm_fwd_kernel<kNC, true>), dim3(grid), dim3(NT),
-                         0, stream.stream(), (const short*)x.data_ptr(),
-                         (const short*)gamma.data_ptr(), nullptr,
-                         (short*)y.data_ptr(), nullptr,
-                         rstd.data_ptr<float>(), (int)n_rows, D, (float)eps);
+      if (has_res) NORM_FWD_LAUNCH(kNC, true, true);
+      else NORM_FWD_LAUNCH(kNC, true, false);
     } else {
-      hipLaunchKernelGGL((norm_fwd_kernel<kNC, false>), dim3(grid), dim3(NT),
-                         0, stream.stream(), (const short*)x.data_ptr(),
-                         (const short*)gamma.data_ptr(), beta_p,
-                         (short*)y.data_ptr(), mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), (int)n_rows, D, (float)eps);
+      if (has_res) NORM_FWD_LAUNCH(kNC, false, true);
+      else NORM_FWD_LAUNCH(kNC, false, false);
     }
   });
+#undef NORM_FWD_LAUNCH
   HIP_CHECK_LAST();
-  return {y, mean.defined() ? mean : rstd, rstd};
+  return {y, has_res ? sum_out : x, mean.defined() ? mean : rstd, rstd};
 }
 
 std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,

@@ -13,23 +13,37 @@ from ._ext import get_ext
 
 
 class _NormFn(torch.autograd.Function):
-    @staticmethod
-    def forward(ctx, x, weight, bias, eps, rms):
-        ext = get_ext(required=True)
-        x = x.contiguous()
-        y, mean, rstd = ext.norm_fwd(x, weight, bias, eps, rms)
-        ctx.save_for_backward(x, weight, mean, rstd)
-        ctx.rms = rms
-        return y
+    """Fused norm with optional residual-add: y = norm(x + res).
+    Returns (y, summed): ``summed`` is x+res (the residual stream) when
+    a residual is given, else x."""
 
     @staticmethod
-    def backward(ctx, dy):
+    def forward(ctx, x, res, weight, bias, eps, rms):
         ext = get_ext(required=True)
-        x, weight, mean, rstd = ctx.saved_tensors
-        dx, dw, db = ext.norm_bwd(dy.contiguous(), x,
+        ctx.set_materialize_grads(False)
+        x = x.contiguous()
+        y, summed, mean, rstd = ext.norm_fwd(x, res, weight, bias, eps,
+                                             rms)
+        has_res = res is not None
+        ctx.save_for_backward(summed, weight, mean, rstd)
+        ctx.rms = rms
+        ctx.has_res = has_res
+        # second output: the residual stream (x+res); dummy when no res
+        return y, (summed if has_res else x.new_empty(0))
+
+    @staticmethod
+    def backward(ctx, dy, dsum):
+        ext = get_ext(required=True)
+        summed, weight, mean, rstd = ctx.saved_tensors
+        if dy is None:
+            dy = torch.zeros_like(summed)
+        dx, dw, db = ext.norm_bwd(dy.contiguous(), summed,
                                   None if ctx.rms else mean, rstd,
                                   weight, ctx.rms)
-        return dx, dw, (None if ctx.rms else db), None, None
+        if dsum is not None:  # residual stream used downstream
+            dx = dx + dsum
+        dres = dx if ctx.has_res else None
+        return dx, dres, dw, (None if ctx.rms else db), None, None
 
 
 def _cpu_layernorm(x, w, b, eps):
@@ -52,10 +66,17 @@ class FusedLayerNorm(nn.Module):
         self.weight = nn.Parameter(torch.ones(dim))
         self.bias = nn.Parameter(torch.zeros(dim))
 
-    def forward(self, x):
+    def forward(self, x, residual=None, return_sum=False):
         if x.is_cuda:
-            return _NormFn.apply(x, self.weight, self.bias, self.eps, False)
-        return _cpu_layernorm(x, self.weight, self.bias, self.eps)
+            y, summed = _NormFn.apply(
+                x, residual.contiguous() if residual is not None
+                else None, self.weight, self.bias, self.eps, False)
+            return (y, summed) if return_sum else y
+        y = _cpu_layernorm(x + residual if residual is not None else x,
+                           self.weight, self.bias, self.eps)
+        if return_sum:
+            return y, (x + residual if residual is not None else x)
+        return y
 
     def extra_repr(self):
         return f"{self.dim}, eps={self.eps}"
@@ -67,10 +88,15 @@ class FusedRMSNorm(nn.Module):
         self.dim, self.eps = dim, eps
         self.weight = nn.Parameter(torch.ones(dim))
 
-    def forward(self, x):
+    def forward(self, x, residual=None, return_sum=False):
         if x.is_cuda:
-            return _NormFn.apply(x, self.weight, None, self.eps, True)
-        return _cpu_rmsnorm(x, self.weight, self.eps)
+            y, summed = _NormFn.apply(
+                x, residual.contiguous() if residual is not None
+                else None, self.weight, None, self.eps, True)
+            return (y, summed) if return_sum else y
+        xs = x + residual if residual is not None else x
+        y = _cpu_rmsnorm(xs, self.weight, self.eps)
+        return (y, xs) if return_sum else y
 
     def extra_repr(self):
         return f"{self.dim}, eps={self.eps}"

@@ -359,4 +359,29 @@ def test_fused_silu_mul():
     assert (y.float() - ref).abs().max().item() < 2e-2
     sc = gf.grad.abs().max().item()
     assert (g.grad.float() - gf.grad).abs().max().item() / sc < 2e-2
-    assert (u.grad.float() - uf.grad).abs().max().item() < 2e-2
+    su = uf.grad.abs().max().item()
+    assert (u.grad.float() - uf.grad).abs().max().item() / su < 2e-2
+
+
+def test_norm_residual_fusion():
+    from gpuschedule_amd.ops import FusedLayerNorm, FusedRMSNorm
+    torch.manual_seed(16)
+    for cls, kw in ((FusedLayerNorm, {}), (FusedRMSNorm, {})):
+        m = cls(256, **kw).cuda().bfloat16()
+        x = torch.randn(64, 256, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        r = torch.randn_like(x).requires_grad_(True)
+        y, summed = m(x, residual=r, return_sum=True)
+        (y.float().square().mean() + summed.float().mean()).backward()
+
+        m2 = cls(256, **kw).cuda().bfloat16()
+        xf = x.detach().clone().requires_grad_(True)
+        rf = r.detach().clone().requires_grad_(True)
+        s2 = xf + rf
+        y2 = m2(s2)
+        (y2.float().square().mean() + s2.float().mean()).backward()
+        assert (y.float() - y2.float()).abs().max().item() < 3e-2
+        for got, want in ((x.grad, xf.grad), (r.grad, rf.grad)):
+            sc = want.float().abs().max().item() + 1e-5
+            assert (got.float() - want.float()).abs().max().item() / sc \
+                < 5e-2
