@@ -61,8 +61,8 @@ def bench_norms(ext, out):
         g = torch.ones(D, device="cuda", dtype=torch.bfloat16)
         b = torch.zeros(D, device="cuda", dtype=torch.bfloat16)
         dy = torch.randn_like(x)
-        t_f = timeit(lambda: ext.norm_fwd(x, g, b, 1e-5, False))
-        y, mean, rstd = ext.norm_fwd(x, g, b, 1e-5, False)
+        t_f = timeit(lambda: ext.norm_fwd(x, None, g, b, 1e-5, False))
+        y, _, mean, rstd = ext.norm_fwd(x, None, g, b, 1e-5, False)
         t_b = timeit(lambda: ext.norm_bwd(dy, x, mean, rstd, g, False))
         t_t = timeit(lambda: torch.nn.functional.layer_norm(
             x, (D,), g, b, 1e-5))

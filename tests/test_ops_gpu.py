@@ -30,7 +30,7 @@ def test_layernorm_fwd_bwd(shape):
     b = torch.randn(D, device="cuda", dtype=torch.bfloat16)
     dy = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
 
-    y, mean, rstd = ext().norm_fwd(x, g, b, 1e-5, False)
+    y, _, mean, rstd = ext().norm_fwd(x, None, g, b, 1e-5, False)
     xf = x.float().requires_grad_(True)
     gf = g.float().requires_grad_(True)
     bf = b.float().requires_grad_(True)
@@ -53,7 +53,7 @@ def test_rmsnorm_fwd_bwd(shape):
     x = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
     g = torch.randn(D, device="cuda", dtype=torch.bfloat16)
     dy = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
-    y, _, rstd = ext().norm_fwd(x, g, None, 1e-6, True)
+    y, _, _, rstd = ext().norm_fwd(x, None, g, None, 1e-6, True)
     xf = x.float().requires_grad_(True)
     ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) \
         * g.float()
