@@ -339,8 +339,10 @@ def test_fused_rope_fwd_bwd():
     x1, x2 = xf.chunk(2, dim=-1)
     ref = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
     ref.backward(dy.float())
-    assert (y.float() - ref).abs().max().item() < 2e-2
-    assert (x.grad.float() - xf.grad).abs().max().item() < 2e-2
+    sy = ref.abs().max().item()
+    assert (y.float() - ref).abs().max().item() / sy < 1e-2
+    sg = xf.grad.abs().max().item()
+    assert (x.grad.float() - xf.grad).abs().max().item() / sg < 1e-2
 
 
 def test_fused_silu_mul():
@@ -356,7 +358,8 @@ def test_fused_silu_mul():
     uf = u.detach().float().requires_grad_(True)
     ref = torch.nn.functional.silu(gf) * uf
     ref.backward(dy.float())
-    assert (y.float() - ref).abs().max().item() < 2e-2
+    sy = ref.abs().max().item()
+    assert (y.float() - ref).abs().max().item() / sy < 2e-2
     sc = gf.grad.abs().max().item()
     assert (g.grad.float() - gf.grad).abs().max().item() / sc < 2e-2
     su = uf.grad.abs().max().item()
