@@ -566,9 +566,12 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                        (const short*)b.data_ptr(), (short*)c.data_ptr(),
                        bias_p, M, N, K, relu ? 1 : 0);
   } else {
+    // A/B result: the 32x32 MFMA variant ties at 4k/8k and loses on
+    // small grids (the structure is stage/barrier-bound, not
+    // MFMA-issue-bound), so 16x16 stays the default.
     static const bool use32 = [] {
-      const char* e = getenv("GSAMD_GEMM16");  // opt-out switch
-      return !(e && e[0] == '1');
+      const char* e = getenv("GSAMD_GEMM32");
+      return e && e[0] == '1';
     }();
     int grid = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
     if (use32) {
