@@ -100,6 +100,168 @@ __global__ __launch_bounds__(NT) void norm_fwd_kernel(
   }
 }
 
+// ---- small-D variants: one WAVE per row (D <= 1024) ---------------
+// The block-per-row kernel leaves (1 - D/2048) of each block idle at
+// small D (62% at BERT's 768).  Here: 4 rows per 256-thread block,
+// pure wave-shuffle reductions, no LDS in the row path.
+template <bool RMS, bool RES>
+__global__ __launch_bounds__(NT) void norm_fwd_wave_kernel(
+    const short* __restrict__ x, const short* __restrict__ res,
+    short* __restrict__ sum_out, const short* __restrict__ gamma,
+    const short* __restrict__ beta, short* __restrict__ y,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    long n_rows, int D, float eps) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  for (long row = (long)blockIdx.x * 4 + wid; row < n_rows;
+       row += (long)gridDim.x * 4) {
+    const short* xr = x + row * D;
+    short8 xv[2];
+    float sum = 0.f, sq = 0.f;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int i = c * 512 + lane * VEC;
+      if (i < D) {
+        xv[c] = *reinterpret_cast<const short8*>(xr + i);
+        if (RES) {
+          short8 rv = *reinterpret_cast<const short8*>(
+              res + row * D + i);
+#pragma unroll
+          for (int j = 0; j < VEC; ++j)
+            xv[c][j] = f32_to_bf16(bf16_to_f32(xv[c][j]) +
+                                   bf16_to_f32(rv[j]));
+          *reinterpret_cast<short8*>(sum_out + row * D + i) = xv[c];
+        }
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float f = bf16_to_f32(xv[c][j]);
+          sum += f;
+          sq += f * f;
+        }
+      }
+    }
+    float mean = 0.f;
+    if (!RMS) mean = wave_reduce_sum(sum) / D;
+    float ms = wave_reduce_sum(sq) / D;
+    float rstd = rsqrtf((RMS ? ms : ms - mean * mean) + eps);
+    if (lane == 0) {
+      if (!RMS && mean_out) mean_out[row] = mean;
+      if (rstd_out) rstd_out[row] = rstd;
+    }
+    short* yr = y + row * D;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int i = c * 512 + lane * VEC;
+      if (i < D) {
+        short8 gv = *reinterpret_cast<const short8*>(gamma + i);
+        short8 out;
+        if (!RMS) {
+          short8 bv = *reinterpret_cast<const short8*>(beta + i);
+#pragma unroll
+          for (int j = 0; j < VEC; ++j)
+            out[j] = f32_to_bf16(
+                (bf16_to_f32(xv[c][j]) - mean) * rstd *
+                    bf16_to_f32(gv[j]) + bf16_to_f32(bv[j]));
+        } else {
+#pragma unroll
+          for (int j = 0; j < VEC; ++j)
+            out[j] = f32_to_bf16(bf16_to_f32(xv[c][j]) * rstd *
+                                 bf16_to_f32(gv[j]));
+        }
+        *reinterpret_cast<short8*>(yr + i) = out;
+      }
+    }
+  }
+}
+
+template <bool RMS>
+__global__ __launch_bounds__(NT) void norm_bwd_wave_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const short* __restrict__ gamma, short* __restrict__ dx,
+    float* __restrict__ part_dg, float* __restrict__ part_db,
+    long n_rows, int D) {
+  // per-thread dgamma/dbeta accumulators for its fixed columns; the
+  // block's 4 waves cover the SAME columns -> merge through LDS once.
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* merge = (float*)smem;  // [2][D]
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  float acc_dg[2][VEC], acc_db[2][VEC];
+#pragma unroll
+  for (int c = 0; c < 2; ++c)
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      acc_dg[c][j] = 0.f;
+      acc_db[c][j] = 0.f;
+    }
+  for (long row = (long)blockIdx.x * 4 + wid; row < n_rows;
+       row += (long)gridDim.x * 4) {
+    const short* dyr = dy + row * D;
+    const short* xr = x + row * D;
+    const float mu = RMS ? 0.f : mean[row];
+    const float rs = rstd[row];
+    short8 dyv[2], xv[2], gv[2];
+    float c1 = 0.f, c2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int i = c * 512 + lane * VEC;
+      if (i < D) {
+        dyv[c] = *reinterpret_cast<const short8*>(dyr + i);
+        xv[c] = *reinterpret_cast<const short8*>(xr + i);
+        gv[c] = *reinterpret_cast<const short8*>(gamma + i);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float h = bf16_to_f32(dyv[c][j]) * bf16_to_f32(gv[c][j]);
+          float xn = (bf16_to_f32(xv[c][j]) - mu) * rs;
+          c1 += h;
+          c2 += h * xn;
+        }
+      }
+    }
+    if (!RMS) c1 = wave_reduce_sum(c1) / D;
+    c2 = wave_reduce_sum(c2) / D;
+    short* dxr = dx + row * D;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int i = c * 512 + lane * VEC;
+      if (i < D) {
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float d = bf16_to_f32(dyv[c][j]);
+          float h = d * bf16_to_f32(gv[c][j]);
+          float xn = (bf16_to_f32(xv[c][j]) - mu) * rs;
+          out[j] = f32_to_bf16(RMS ? rs * (h - xn * c2)
+                                   : rs * (h - c1 - xn * c2));
+          acc_dg[c][j] += d * xn;
+          acc_db[c][j] += d;
+        }
+        *reinterpret_cast<short8*>(dxr + i) = out;
+      }
+    }
+  }
+  // merge the 4 waves' column accumulators -> block partial slot
+  for (int i = threadIdx.x; i < 2 * D; i += NT) merge[i] = 0.f;
+  __syncthreads();
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int i = c * 512 + lane * VEC;
+    if (i < D) {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        atomicAdd(merge + i + j, acc_dg[c][j]);
+        atomicAdd(merge + D + i + j, acc_db[c][j]);
+      }
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < D; i += NT) {
+    part_dg[(long)blockIdx.x * D + i] = merge[i];
+    if (part_db) part_db[(long)blockIdx.x * D + i] = merge[D + i];
+  }
+}
+
 // ---------------- backward (dx + per-block dgamma/dbeta partials) ---
 template <int NCHUNK, bool RMS>
 __global__ __launch_bounds__(NT) void norm_bwd_kernel(
