@@ -428,6 +428,28 @@ std::vector<torch::Tensor> norm_fwd(torch::Tensor x,
     res_p = (const short*)residual->data_ptr();
     sum_p = (short*)sum_out.data_ptr();
   }
+  if (D <= 1024) {  // wave-per-row fast path
+    long wgrid = std::min<long>((n_rows + 3) / 4, 8192);
+#define NORM_FWDW_LAUNCH(RMSB, RESB)                                      \
+    hipLaunchKernelGGL((norm_fwd_wave_kernel<RMSB, RESB>),               \
+                       dim3((int)wgrid), dim3(NT), 0, stream.stream(),   \
+                       (const short*)x.data_ptr(), res_p, sum_p,         \
+                       (const short*)gamma.data_ptr(), beta_p,           \
+                       (short*)y.data_ptr(),                             \
+                       RMSB ? nullptr : mean.data_ptr<float>(),          \
+                       rstd.data_ptr<float>(), n_rows, D, (float)eps)
+    if (rms) {
+      if (has_res) NORM_FWDW_LAUNCH(true, true);
+      else NORM_FWDW_LAUNCH(true, false);
+    } else {
+      if (has_res) NORM_FWDW_LAUNCH(false, true);
+      else NORM_FWDW_LAUNCH(false, false);
+    }
+#undef NORM_FWDW_LAUNCH
+    HIP_CHECK_LAST();
+    return {y, has_res ? sum_out : x, mean.defined() ? mean : rstd,
+            rstd};
+  }
 #define NORM_FWD_LAUNCH(kNC, RMSB, RESB)                                  \
   hipLaunchKernelGGL((norm_fwd_kernel<kNC, RMSB, RESB>), dim3(grid),      \
                      dim3(NT), 0, stream.stream(),                        \
@@ -461,10 +483,35 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dx = torch::empty_like(x);
   int nc = pick_nchunk(D);
   int grid = (int)std::min<long>(n_rows, 512);
+  if (D <= 1024)
+    grid = (int)std::min<long>((n_rows + 3) / 4, 512);
   auto fopt = x.options().dtype(torch::kFloat32);
   auto part_dg = torch::empty({grid, D}, fopt);
   auto part_db = rms ? torch::Tensor() : torch::empty({grid, D}, fopt);
   auto stream = c10::cuda::getCurrentCUDAStream();
+  if (D <= 1024) {
+    if (rms) {
+      hipLaunchKernelGGL((norm_bwd_wave_kernel<true>), dim3(grid),
+                         dim3(NT), 2 * D * 4, stream.stream(),
+                         (const short*)dy.data_ptr(),
+                         (const short*)x.data_ptr(), nullptr,
+                         rstd.data_ptr<float>(),
+                         (const short*)gamma.data_ptr(),
+                         (short*)dx.data_ptr(),
+                         part_dg.data_ptr<float>(), nullptr, n_rows, D);
+    } else {
+      hipLaunchKernelGGL((norm_bwd_wave_kernel<false>), dim3(grid),
+                         dim3(NT), 2 * D * 4, stream.stream(),
+                         (const short*)dy.data_ptr(),
+                         (const short*)x.data_ptr(),
+                         mean->data_ptr<float>(),
+                         rstd.data_ptr<float>(),
+                         (const short*)gamma.data_ptr(),
+                         (short*)dx.data_ptr(),
+                         part_dg.data_ptr<float>(),
+                         part_db.data_ptr<float>(), n_rows, D);
+    }
+  } else
   DISPATCH_NCHUNK(nc, rms, {
     if (rms) {
       hipLaunchKernelGGL((norm_bwd_kernel<kNC, true>), dim3(grid), dim3(NT),
