@@ -69,10 +69,10 @@ def test_eviction_restore_cycle_gpu(tmp_path):
     # Artificially small per-GPU HBM so two jobs cannot stay resident:
     # suspending one forces checkpoint-eviction, resuming restores it.
     from gpuschedule_amd.trace.job import Job
-    hbm = 10 * 1024**3  # 10 GB "GPU"
+    hbm = 16 * 1024**3  # 16 GB "GPU" (bert est. 12 GB: one fits)
     sched = Scheduler(Cluster(1, 1), make_policy("gandiva",
                                                  quantum=0.2))
-    # catalog says bert-base needs 12GB > 10/2: only one resident
+    # catalog says bert-base needs 12GB: two cannot co-reside
     specs = [J("e1", iters=8, model="bert-base", batch=8),
              J("e2", iters=8, model="bert-base", batch=8)]
     for g in sched.cluster.gpus:
