@@ -98,9 +98,6 @@ class TorchBackend:
 
     # ---- plan application ---------------------------------------
 
-    def _leader(self, gpus: List[int]) -> bool:
-        return self.rank == min(gpus)
-
     def _apply_phase_a(self, plan: Plan, meta: Dict[str, tuple]) -> None:
         for job_id, reason in plan.preempts:
             r = self.runners.get(job_id)
@@ -171,7 +168,8 @@ class TorchBackend:
             r = self.runners.pop(job_id, None)
             if r is not None:
                 r.release()
-            self.ckpt.drop(job_id) if self.rank == 0 else None
+            if self.rank == 0:
+                self.ckpt.drop(job_id)
         if msg.stop:
             return True
 

@@ -6,7 +6,6 @@ is padded to a multiple of 64 for the MFMA LM head.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 
 import torch

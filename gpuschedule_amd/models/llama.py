@@ -13,7 +13,6 @@ from dataclasses import dataclass
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops.attention import flash_attention
 from ..ops.ce import fused_cross_entropy

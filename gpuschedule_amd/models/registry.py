@@ -8,8 +8,6 @@ random-init weights + synthetic data.
 """
 from __future__ import annotations
 
-from typing import Callable, Tuple
-
 import torch
 
 from .. import catalog

@@ -8,7 +8,6 @@ max-pool; config-3 ResNet-50 jobs).
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

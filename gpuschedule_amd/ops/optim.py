@@ -7,7 +7,6 @@ multi-kernel unfused update.  CPU fallback does the same math in torch
 """
 from __future__ import annotations
 
-import math
 
 import torch
 

@@ -16,7 +16,6 @@ also bounds LAS demotion / Gandiva quantum latency).
 """
 from __future__ import annotations
 
-import math
 from typing import Dict, Optional
 
 from .. import catalog
