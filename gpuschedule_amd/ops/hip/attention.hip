@@ -48,17 +48,19 @@ __device__ __forceinline__ float group_sum(float v) {
   return v;
 }
 
-template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
+template <int D, bool CAUSAL, int NW>
+__global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ Vt, short* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int S, float scale) {
+  constexpr int NT = NW * 64;      // threads (NW waves x 16 q rows)
+  constexpr int QB = NW * 16;      // q rows per block
   constexpr int DF = D / 16;       // PV col fragments
   constexpr int KC = D / 32;       // QK^T k-steps over D
   constexpr int ROWB = D * 2;      // K tile LDS row bytes
-  // LDS: double-buffered K [KVB][D] + Vt [D][KVB] + 4 wave-private P
+  // LDS: double-buffered K [KVB][D] + Vt [D][KVB] + NW wave-private P
   __shared__ __attribute__((aligned(16))) char lds[
-      2 * (KVB * D * 2 + D * KVB * 2) + 4 * 16 * KVB * 2];
+      2 * (KVB * D * 2 + D * KVB * 2) + NW * 16 * KVB * 2];
   char* k_cur = lds;
   char* v_cur = lds + KVB * D * 2;
   char* k_nxt = lds + 2 * KVB * D * 2;
@@ -70,10 +72,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
   const int tid = threadIdx.x;
   char* my_p = p_lds + wid * 16 * KVB * 2;
 
-  const int bh = blockIdx.x / ((S + 63) / 64);
-  const int qt = blockIdx.x % ((S + 63) / 64);
+  const int bh = blockIdx.x / ((S + QB - 1) / QB);
+  const int qt = blockIdx.x % ((S + QB - 1) / QB);
   const int b = bh / H, h = bh % H;
-  const int q0 = qt * 64 + wid * 16;      // this wave's first q row
+  const int q0 = qt * QB + wid * 16;      // this wave's first q row
 
   const long qoff = ((long)(b * H + h) * S) * D;
   const long voff = ((long)(b * H + h) * D) * S;  // Vt is [D][S]
@@ -107,22 +109,22 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
 #pragma unroll
   for (int f = 0; f < DF; ++f) o_acc[f] = {0.f, 0.f, 0.f, 0.f};
 
-  const int q_tile_end = qt * 64 + 63;
+  const int q_tile_end = qt * QB + QB - 1;
   const int n_kv = CAUSAL
       ? min((q_tile_end / KVB) + 1, (S + KVB - 1) / KVB)
       : (S + KVB - 1) / KVB;
 
   // ---- staging helpers (register path: T14 async-split) ---------
-  constexpr int PIECES_K = KVB * D * 2 / (NTHREADS * 16);
+  constexpr int PIECES_K = KVB * D * 2 / (NT * 16);
   constexpr int CPR = D * 2 / 16;     // 16B chunks per K row
-  constexpr int PIECES_V = D * KVB * 2 / (NTHREADS * 16);
+  constexpr int PIECES_V = D * KVB * 2 / (NT * 16);
   constexpr int CPRV = KVB * 2 / 16;
   short8 kst[PIECES_K], vst[PIECES_V];
 
   auto stage_load = [&](int kv0) {
 #pragma unroll
     for (int p = 0; p < PIECES_K; ++p) {
-      int idx = p * NTHREADS + tid;
+      int idx = p * NT + tid;
       int r = idx / CPR, cb = (idx % CPR) * 16;
       int kvr = kv0 + r;
       if (kvr < S) {
@@ -135,7 +137,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
     }
 #pragma unroll
     for (int p = 0; p < PIECES_V; ++p) {
-      int idx = p * NTHREADS + tid;
+      int idx = p * NT + tid;
       int r = idx / CPRV, cb = (idx % CPRV) * 16;
       int kvc = kv0 + cb / 2;
       if (kvc + 7 < S) {
@@ -152,13 +154,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_fwd_kernel(
   auto stage_write = [&](char* kbuf, char* vbuf) {
 #pragma unroll
     for (int p = 0; p < PIECES_K; ++p) {
-      int idx = p * NTHREADS + tid;
+      int idx = p * NT + tid;
       int r = idx / CPR, cb = (idx % CPR) * 16;
       *reinterpret_cast<short8*>(kbuf + swz(r, cb, ROWB)) = kst[p];
     }
 #pragma unroll
     for (int p = 0; p < PIECES_V; ++p) {
-      int idx = p * NTHREADS + tid;
+      int idx = p * NT + tid;
       int r = idx / CPRV, cb = (idx % CPRV) * 16;
       *reinterpret_cast<short8*>(vbuf + swz(r, cb, KVB * 2)) = vst[p];
     }
@@ -676,21 +678,29 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   auto lse = torch::empty({B, H, S},
                           q.options().dtype(torch::kFloat32));
   float scale = 1.0f / sqrtf((float)D);
-  int grid = B * H * ((S + 63) / 64);
+  // 8-wave blocks (128 q rows) halve per-FLOP K/V re-reads at long S;
+  // 4-wave blocks keep the grid full at short S.
+  const bool wide = (long)B * H * ((S + 127) / 128) >= 256 && S >= 1024;
+  const int QB = wide ? 128 : 64;
+  int grid = B * H * ((S + QB - 1) / QB);
   auto stream = c10::cuda::getCurrentCUDAStream();
-#define LAUNCH(DD, CC)                                                   \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), dim3(grid),              \
-                     dim3(NTHREADS), 0, stream.stream(),                 \
+#define LAUNCH(DD, CC, NW)                                               \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, NW>), dim3(grid),          \
+                     dim3(NW * 64), 0, stream.stream(),                  \
                      (const short*)q.data_ptr(),                         \
                      (const short*)k.data_ptr(),                         \
                      (const short*)vt.data_ptr(),                        \
                      (short*)o.data_ptr(), lse.data_ptr<float>(),        \
                      B, H, S, scale)
-  if (D == 64) {
-    if (causal) { LAUNCH(64, true); } else { LAUNCH(64, false); }
-  } else {
-    if (causal) { LAUNCH(128, true); } else { LAUNCH(128, false); }
-  }
+#define LAUNCH_D(DD)                                                     \
+  do {                                                                   \
+    if (causal) { if (wide) LAUNCH(DD, true, 8);                         \
+                  else LAUNCH(DD, true, 4); }                            \
+    else { if (wide) LAUNCH(DD, false, 8);                               \
+           else LAUNCH(DD, false, 4); }                                  \
+  } while (0)
+  if (D == 64) LAUNCH_D(64); else LAUNCH_D(128);
+#undef LAUNCH_D
 #undef LAUNCH
   HIP_CHECK_LAST();
   return {o, lse};
