@@ -56,6 +56,8 @@ def check(B, H, S, D, causal, seed=0, spike=False):
     (1, 2, 2048, 128),  # long seq
     (1, 1, 200, 64),    # seq tail (not a multiple of 64)
     (3, 5, 72, 128),    # odd everything
+    (4, 8, 1024, 64),   # triggers the 8-wave (128-row) dispatch
+    (2, 16, 1024, 128), # 8-wave, D=128
 ])
 def test_attn_fwd(shape, causal):
     check(*shape, causal)
