@@ -73,6 +73,7 @@ class JobRunner:
         # jobs' kernels can overlap.
         self.stream = (torch.cuda.Stream(device)
                        if str(device) != "cpu" else None)
+        self._gen = None  # device RNG for synthetic batches
         # measured per-GPU footprint (reported once to the scheduler
         # for bin-packing refinement)
         self._mem_base = (torch.cuda.memory_allocated(device)
@@ -98,16 +99,21 @@ class JobRunner:
         seed = (_job_seed(self.job_id) * 1000003 + it * 8191 +
                 self.gang_rank) & 0x7FFFFFFF
         dev = torch.device(self.device)
-        gen = torch.Generator(device="cpu").manual_seed(seed)
-        # generate on CPU (deterministic across devices), move over
-        x, y = self.make_batch(self.batch_size, "cpu", gen)
-        in_dtype = self.dtype if x.is_floating_point() else None
         nb = str(dev) != "cpu"
-        x = x.to(dev, dtype=in_dtype, non_blocking=nb)
-        if nb and x.dim() == 4:
-            x = x.contiguous(memory_format=torch.channels_last)
-        y = y.to(dev, non_blocking=nb)
-        return x, y
+        if nb:
+            # generate synthetic data ON the GPU: a ResNet-50 batch is
+            # ~38 MB — host-side randn + H2D would dominate the tick
+            if self._gen is None:
+                self._gen = torch.Generator(device=dev)
+            self._gen.manual_seed(seed)
+            x, y = self.make_batch(self.batch_size, dev, self._gen)
+            if x.is_floating_point():
+                x = x.to(self.dtype)
+            if x.dim() == 4:
+                x = x.contiguous(memory_format=torch.channels_last)
+            return x, y
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        return self.make_batch(self.batch_size, "cpu", gen)
 
     def _eager_iter(self, x, y):
         if not getattr(self.optim, "fused_zero_grad", False):

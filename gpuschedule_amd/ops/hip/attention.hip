@@ -680,7 +680,13 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   float scale = 1.0f / sqrtf((float)D);
   // 8-wave blocks (128 q rows) halve per-FLOP K/V re-reads at long S;
   // 4-wave blocks keep the grid full at short S.
-  const bool wide = (long)B * H * ((S + 127) / 128) >= 256 && S >= 1024;
+  static const int force = [] {
+    const char* e = getenv("GSAMD_ATTN_NW");
+    return e ? atoi(e) : 0;
+  }();
+  bool wide = (long)B * H * ((S + 127) / 128) >= 256 && S >= 1024;
+  if (force == 4) wide = false;
+  if (force == 8) wide = true;
   const int QB = wide ? 128 : 64;
   int grid = B * H * ((S + QB - 1) / QB);
   auto stream = c10::cuda::getCurrentCUDAStream();
