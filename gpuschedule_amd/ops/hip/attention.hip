@@ -684,7 +684,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     const char* e = getenv("GSAMD_ATTN_NW");
     return e ? atoi(e) : 0;
   }();
-  bool wide = (long)B * H * ((S + 127) / 128) >= 256 && S >= 1024;
+  // A/B (within-box, repeated): 8-wave wins at S=1024-2048 (+6%),
+  // loses at S=4096 (-8%); gate to the measured window.
+  bool wide = (long)B * H * ((S + 127) / 128) >= 256 && S >= 1024 &&
+              S <= 2048;
   if (force == 4) wide = false;
   if (force == 8) wide = true;
   const int QB = wide ? 128 : 64;
