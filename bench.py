@@ -61,12 +61,21 @@ def build_trace(n_jobs: int, steps: int, n_gpus: int, iters_per_tick: int,
         arrivals.append((tick, JobSpec(
             job_id=f"job{i:03d}", submit_time=0.0, n_gpus=1,
             total_iters=iters, model=models[i % 2])))
-    # per-GPU fill jobs: best-effort, never complete inside the window
-    fill = [JobSpec(job_id=f"bg{g}", submit_time=0.0, n_gpus=1,
-                    total_iters=10**9,
-                    model=("bert-base" if g % 2 == 0 else "resnet50"),
-                    background=True)
-            for g in range(n_gpus)]
+    # per-GPU fill jobs: best-effort, never complete inside the window.
+    # At N>=2 the first fill is a 2-GPU DDP gang so the scale bench
+    # exercises gang scheduling + bucketed RCCL all-reduce over xGMI
+    # (fill jobs are excluded from JCT/makespan, so the strong-scaling
+    # metric over the 1-GPU trace jobs is unaffected).
+    fill = []
+    g = 0
+    while g < n_gpus:
+        width = 2 if (g == 0 and n_gpus >= 2) else 1
+        fill.append(JobSpec(
+            job_id=f"bg{g}", submit_time=0.0, n_gpus=width,
+            total_iters=10**9,
+            model=("bert-base" if g % 2 == 0 else "resnet50"),
+            background=True))
+        g += width
     return arrivals, fill
 
 
