@@ -39,6 +39,9 @@ void adamw_multi(torch::Tensor ptrs, torch::Tensor sizes,
                  torch::Tensor chunk_t, torch::Tensor chunk_off,
                  torch::Tensor stepbuf, double lr, double beta1,
                  double beta2, double eps, double wd);
+void sgd_multi(torch::Tensor ptrs, torch::Tensor sizes,
+               torch::Tensor chunk_t, torch::Tensor chunk_off,
+               double lr, double momentum, double wd, bool bf16);
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosb,
                          torch::Tensor sinb, bool backward);
 torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
@@ -70,6 +73,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul_bwd", &silu_mul_bwd, "fused SwiGLU gate bwd");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward");
+  m.def("sgd_multi", &sgd_multi, "multi-tensor fused SGD (one launch/dtype, zeroes grads)");
   m.def("adamw_multi", &adamw_multi, "multi-tensor fused AdamW (one launch, zeroes grads)");
   m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");
   m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");

@@ -150,6 +150,53 @@ __global__ __launch_bounds__(NT) void adamw_mt_kernel(
   }
 }
 
+// Multi-tensor SGD-momentum: one launch per dtype group (bf16 params
+// with fp32 masters; fp32 params e.g. BatchNorm affine).  Zeroes the
+// consumed grads like the AdamW variant.
+template <bool BF16>
+__global__ __launch_bounds__(NT) void sgd_mt_kernel(
+    const long* __restrict__ ptrs,   // [T][4]: p, p32(or 0), g, buf
+    const long* __restrict__ sizes, const int* __restrict__ chunk_t,
+    const long* __restrict__ chunk_off, int n_chunks, float lr,
+    float momentum, float wd) {
+  for (int c = blockIdx.x; c < n_chunks; c += gridDim.x) {
+    const int t = chunk_t[c];
+    const long off = chunk_off[c];
+    const long n = min((long)MT_CHUNK, sizes[t] - off);
+    float* buf = (float*)ptrs[t * 4 + 3] + off;
+    if (BF16) {
+      short* p = (short*)ptrs[t * 4 + 0] + off;
+      float* p32 = (float*)ptrs[t * 4 + 1] + off;
+      short* g = (short*)ptrs[t * 4 + 2] + off;
+      for (long i = (long)threadIdx.x * 4; i < n; i += NT * 4) {
+        long lim = min(i + 4, n);
+        for (long k = i; k < lim; ++k) {
+          float gg = bf16_to_f32(g[k]) + wd * p32[k];
+          float b = momentum * buf[k] + gg;
+          buf[k] = b;
+          float pw = p32[k] - lr * b;
+          p32[k] = pw;
+          p[k] = f32_to_bf16(pw);
+          g[k] = 0;
+        }
+      }
+    } else {
+      float* p = (float*)ptrs[t * 4 + 0] + off;
+      float* g = (float*)ptrs[t * 4 + 2] + off;
+      for (long i = (long)threadIdx.x * 4; i < n; i += NT * 4) {
+        long lim = min(i + 4, n);
+        for (long k = i; k < lim; ++k) {
+          float gg = g[k] + wd * p[k];
+          float b = momentum * buf[k] + gg;
+          buf[k] = b;
+          p[k] -= lr * b;
+          g[k] = 0.f;
+        }
+      }
+    }
+  }
+}
+
 // fp32-native SGD (BatchNorm affine params train in fp32)
 __global__ __launch_bounds__(NT) void sgd_fp32_kernel(
     float* __restrict__ p, const float* __restrict__ grad,
@@ -189,6 +236,29 @@ void adamw_multi(torch::Tensor ptrs, torch::Tensor sizes,
                      chunk_off.data_ptr<long>(), n_chunks, (float)lr,
                      (float)beta1, (float)beta2, (float)eps, (float)wd,
                      stepbuf.data_ptr<float>());
+  HIP_CHECK_LAST();
+}
+
+void sgd_multi(torch::Tensor ptrs, torch::Tensor sizes,
+               torch::Tensor chunk_t, torch::Tensor chunk_off,
+               double lr, double momentum, double wd, bool bf16) {
+  TORCH_CHECK(ptrs.is_cuda() && ptrs.scalar_type() == torch::kLong);
+  int n_chunks = chunk_t.numel();
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  int grid = std::min(n_chunks, 2048);
+  if (bf16) {
+    hipLaunchKernelGGL((sgd_mt_kernel<true>), dim3(grid), dim3(NT), 0,
+                       stream.stream(), ptrs.data_ptr<long>(),
+                       sizes.data_ptr<long>(), chunk_t.data_ptr<int>(),
+                       chunk_off.data_ptr<long>(), n_chunks, (float)lr,
+                       (float)momentum, (float)wd);
+  } else {
+    hipLaunchKernelGGL((sgd_mt_kernel<false>), dim3(grid), dim3(NT), 0,
+                       stream.stream(), ptrs.data_ptr<long>(),
+                       sizes.data_ptr<long>(), chunk_t.data_ptr<int>(),
+                       chunk_off.data_ptr<long>(), n_chunks, (float)lr,
+                       (float)momentum, (float)wd);
+  }
   HIP_CHECK_LAST();
 }
 

@@ -152,10 +152,59 @@ class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
 
 
 class FusedSGD(_MasterStateMixin, torch.optim.Optimizer):
+    """Fused SGD-momentum with a multi-tensor fast path: one launch
+    per dtype class (bf16 params w/ fp32 masters, fp32 params) that
+    also zeroes the grads in place (see FusedAdamW)."""
+
+    MT_CHUNK = 8192
+
     def __init__(self, params, lr=0.1, momentum=0.9, weight_decay=5e-4):
         defaults = dict(lr=lr, momentum=momentum,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self.fused_zero_grad = False
+        self._mt = None  # list of (ptrs, sizes, ct, co, bf16)
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        self._mt = None
+        self.fused_zero_grad = False
+
+    def _try_build_mt(self, group):
+        params = [p for p in group["params"] if p.requires_grad]
+        if not params or not all(
+                p.is_cuda and p.grad is not None
+                and p.dtype in (torch.bfloat16, torch.float32)
+                and p.is_contiguous() and p.grad.is_contiguous()
+                for p in params):
+            return None
+        dev = params[0].device
+        tables = []
+        for want_bf16 in (True, False):
+            sel = [p for p in params
+                   if (p.dtype == torch.bfloat16) == want_bf16]
+            if not sel:
+                continue
+            ptrs, sizes, ct, co = [], [], [], []
+            for ti, p in enumerate(sel):
+                st = self.state[p]
+                p32 = st["p32"].data_ptr() if want_bf16 else 0
+                ptrs.append([p.data.data_ptr(), p32,
+                             p.grad.data_ptr(), st["buf"].data_ptr()])
+                n = p.numel()
+                sizes.append(n)
+                off = 0
+                while off < n:
+                    ct.append(ti)
+                    co.append(off)
+                    off += self.MT_CHUNK
+            tables.append((
+                torch.tensor(ptrs, dtype=torch.int64, device=dev),
+                torch.tensor(sizes, dtype=torch.int64, device=dev),
+                torch.tensor(ct, dtype=torch.int32, device=dev),
+                torch.tensor(co, dtype=torch.int64, device=dev),
+                want_bf16))
+        return tables
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -170,6 +219,20 @@ class FusedSGD(_MasterStateMixin, torch.optim.Optimizer):
                 if not state:
                     state["p32"] = p.detach().float().clone()
                     state["buf"] = torch.zeros_like(state["p32"])
+            if self._mt is None and len(self.param_groups) == 1:
+                self._mt = self._try_build_mt(group)
+                if self._mt:
+                    self.fused_zero_grad = True
+            if self._mt:
+                ext = get_ext(required=True)
+                for ptrs, sizes, ct, co, bf16 in self._mt:
+                    ext.sgd_multi(ptrs, sizes, ct, co, lr, mom, wd,
+                                  bf16)
+                return loss
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
                 if p.is_cuda and p.dtype == torch.bfloat16:
                     ext = get_ext(required=True)
                     ext.sgd_step(state["p32"], p.data, p.grad.contiguous(),

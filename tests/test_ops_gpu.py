@@ -388,3 +388,35 @@ def test_norm_residual_fusion():
             sc = want.float().abs().max().item() + 1e-5
             assert (got.float() - want.float()).abs().max().item() / sc \
                 < 5e-2
+
+
+def test_multi_tensor_sgd_matches_torch():
+    from gpuschedule_amd.ops import FusedSGD
+    torch.manual_seed(17)
+    # mixed dtypes like ResNet: bf16 "conv" weights + fp32 "BN" params
+    a_bf = torch.nn.Linear(32, 32).cuda().bfloat16()
+    a_fp = torch.nn.Linear(32, 4).cuda()
+    b_bf = torch.nn.Linear(32, 32).cuda()
+    b_fp = torch.nn.Linear(32, 4).cuda()
+    with torch.no_grad():
+        for src, dst in ((a_bf, b_bf), (a_fp, b_fp)):
+            for ps, pd in zip(src.parameters(), dst.parameters()):
+                pd.copy_(ps.float())
+    params_a = list(a_bf.parameters()) + list(a_fp.parameters())
+    oa = FusedSGD(params_a, lr=0.05, momentum=0.9, weight_decay=1e-3)
+    ob = torch.optim.SGD(list(b_bf.parameters()) + list(b_fp.parameters()),
+                         lr=0.05, momentum=0.9, weight_decay=1e-3)
+    x = torch.randn(16, 32, device="cuda")
+    for _ in range(5):
+        if not oa.fused_zero_grad:
+            oa.zero_grad(set_to_none=True)
+        (a_fp(a_bf(x.bfloat16()).float()).square().mean()).backward()
+        oa.step()
+        ob.zero_grad()
+        (b_fp(b_bf(x)).square().mean()).backward()
+        ob.step()
+    assert oa._mt, "multi-tensor SGD did not engage"
+    for pa, pb in zip(params_a,
+                      list(b_bf.parameters()) + list(b_fp.parameters())):
+        err = (pa.float() - pb.float()).abs().max().item()
+        assert err < 3e-2, f"MT sgd drift {err}"
