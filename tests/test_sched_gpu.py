@@ -69,14 +69,14 @@ def test_eviction_restore_cycle_gpu(tmp_path):
     # Artificially small per-GPU HBM so two jobs cannot stay resident:
     # suspending one forces checkpoint-eviction, resuming restores it.
     from gpuschedule_amd.trace.job import Job
-    hbm = 16 * 1024**3  # 16 GB "GPU" (bert est. 12 GB: one fits)
+    hbm = 6 * 1024**3  # 6 GB "GPU" (resnet18 est. 4 GB: one fits)
     # quantum far below per-job runtime so rotation (and therefore
     # eviction) must happen even with warm kernels
     sched = Scheduler(Cluster(1, 1), make_policy("gandiva",
                                                  quantum=0.01))
-    # catalog says bert-base needs 12GB: two cannot co-reside
-    specs = [J("e1", iters=10, model="bert-base", batch=4),
-             J("e2", iters=10, model="bert-base", batch=4)]
+    # catalog says resnet18 needs 4GB: two cannot co-reside on 6GB
+    specs = [J("e1", iters=10, model="resnet18", batch=32),
+             J("e2", iters=10, model="resnet18", batch=32)]
     for g in sched.cluster.gpus:
         g.hbm_bytes = hbm
     sched.submit(specs)
