@@ -20,6 +20,11 @@ from dataclasses import dataclass
 
 GB = 1024**3
 
+# throughput factor when two jobs co-run on one GPU (Gandiva packing):
+# each job's iteration takes ~PACK_SLOWDOWN x longer (they share CUs
+# and HBM bandwidth, but overlap hides launch gaps)
+PACK_SLOWDOWN = 1.7
+
 
 @dataclass(frozen=True)
 class ModelMeta:

@@ -177,21 +177,30 @@ class TorchBackend:
         self._barrier()
         self._apply_phase_b(msg.plan, msg.meta)
 
-        # run this tick's quotas for locally hosted running jobs
+        # run this tick's quotas for locally hosted running jobs.
+        # Launch every job's iterations first (per-job HIP streams:
+        # co-packed jobs overlap), then sync each.
         report = {}
+        launched = []
         for job_id, n in msg.iters.items():
             r = self.runners.get(job_id)
             if r is None or self.rank not in msg.meta[job_id][2]:
                 continue
             try:
-                done, secs = r.step(n)
+                r.step_async(n)
+                launched.append((job_id, r, n))
+            except Exception as e:  # job-level fault isolation
+                report[job_id] = ("fail", 0, 0.0, f"{type(e).__name__}: {e}")
+        for job_id, r, n in launched:
+            try:
+                done, secs = r.step_finish()
                 if r.gang_rank == 0:
                     mem = 0
                     if r.measured_mem and not r._mem_reported:
                         mem = r.measured_mem
                         r._mem_reported = True
                     report[job_id] = ("ok", done, secs, r.last_loss, mem)
-            except Exception as e:  # job-level fault isolation
+            except Exception as e:
                 report[job_id] = ("fail", 0, 0.0, f"{type(e).__name__}: {e}")
 
         reports = self._gather(report)

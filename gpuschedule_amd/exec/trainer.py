@@ -74,6 +74,9 @@ class JobRunner:
         self.stream = (torch.cuda.Stream(device)
                        if str(device) != "cpu" else None)
         self._gen = None  # device RNG for synthetic batches
+        self._loss_ref = None
+        self._pending = 0
+        self._t0 = 0.0
         # measured per-GPU footprint (reported once to the scheduler
         # for bin-packing refinement)
         self._mem_base = (torch.cuda.memory_allocated(device)
@@ -155,10 +158,12 @@ class JobRunner:
         self._graph_warmup -= 1
         return self._eager_iter(x, y)
 
-    def step(self, n_iters: int = 1) -> tuple:
-        """Run up to n_iters training iterations; returns
-        (iters_done, wall_seconds)."""
-        t0 = time.perf_counter()
+    def step_async(self, n_iters: int = 1) -> None:
+        """Launch n_iters iterations on this job's stream WITHOUT the
+        final sync (lets co-packed jobs' kernels overlap);
+        :meth:`step_finish` completes the accounting."""
+        self._t0 = time.perf_counter()
+        self._pending = n_iters
         loss = None
         if self.stream is None:
             for _ in range(n_iters):
@@ -173,14 +178,27 @@ class JobRunner:
                     loss = None  # drop autograd refs before capture
                     loss = self._one_iter(x, y)
                     self.done_iters += 1
+        self._loss_ref = loss
+
+    def step_finish(self) -> tuple:
+        if self.stream is not None:
             self.stream.synchronize()
             if self.measured_mem == 0 and self.done_iters >= 4:
                 used = (torch.cuda.memory_allocated(self.device)
                         - self._mem_base)
                 self.measured_mem = max(used, 1)
-        if loss is not None:
-            self.last_loss = float(loss.detach())
-        return n_iters, time.perf_counter() - t0
+        if self._loss_ref is not None:
+            self.last_loss = float(self._loss_ref.detach())
+            self._loss_ref = None
+        n = self._pending
+        self._pending = 0
+        return n, time.perf_counter() - self._t0
+
+    def step(self, n_iters: int = 1) -> tuple:
+        """Run up to n_iters training iterations; returns
+        (iters_done, wall_seconds)."""
+        self.step_async(n_iters)
+        return self.step_finish()
 
     def release(self) -> None:
         """Drop GPU state (eviction or completion)."""

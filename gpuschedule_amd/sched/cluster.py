@@ -29,9 +29,16 @@ class GPUState:
     gpu_id: int                  # global index
     node_id: int
     hbm_bytes: int = MI355X_HBM_BYTES
-    running: Optional[str] = None          # job_id currently executing
+    # jobs currently EXECUTING here (normally one; Gandiva packing may
+    # co-run two on separate HIP streams)
+    running_set: Set[str] = field(default_factory=set)
     resident: Set[str] = field(default_factory=set)  # jobs with state in HBM
     resident_bytes: int = 0
+
+    @property
+    def running(self) -> Optional[str]:
+        """First running job (compat accessor; None when idle)."""
+        return next(iter(self.running_set), None)
 
     @property
     def free_bytes(self) -> int:
@@ -55,13 +62,19 @@ class Cluster:
         return self.gpus[i]
 
     def free_gpus(self) -> List[int]:
-        return [g.gpu_id for g in self.gpus if g.running is None]
+        return [g.gpu_id for g in self.gpus if not g.running_set]
 
     def n_free(self) -> int:
-        return sum(1 for g in self.gpus if g.running is None)
+        return sum(1 for g in self.gpus if not g.running_set)
 
     def running_jobs(self) -> Set[str]:
-        return {g.running for g in self.gpus if g.running is not None}
+        out: Set[str] = set()
+        for g in self.gpus:
+            out |= g.running_set
+        return out
+
+    def running_on(self, gpu_id: int) -> Set[str]:
+        return set(self.gpus[gpu_id].running_set)
 
     def utilization(self) -> float:
         return 1.0 - self.n_free() / max(1, self.n_gpus)
@@ -88,7 +101,7 @@ class Cluster:
         def window_ok(ids: Sequence[int]) -> bool:
             for i in ids:
                 g = self.gpus[i]
-                if g.running is not None:
+                if g.running_set:
                     return False
                 extra = sum(self._mem.get(j, 0) for j in g.resident
                             if j in resident_ok)
@@ -119,20 +132,22 @@ class Cluster:
         for node_start in range(0, self.n_gpus, self.gpus_per_node):
             node_end = min(node_start + self.gpus_per_node, self.n_gpus)
             ok = max(ok, sum(1 for i in range(node_start, node_end)
-                             if self.gpus[i].running is None
+                             if not self.gpus[i].running_set
                              and self.gpus[i].free_bytes >= mem_bytes))
         return ok >= n
 
     # ---- mutation ------------------------------------------------
 
-    def place(self, job: Job, gpus: List[int], mem_bytes: int) -> None:
-        """Mark job executing on ``gpus``; make it resident there."""
+    def place(self, job: Job, gpus: List[int], mem_bytes: int,
+              capacity: int = 1) -> None:
+        """Mark job executing on ``gpus``; make it resident there.
+        ``capacity``: max co-running jobs per GPU (Gandiva packing)."""
         self._mem[job.job_id] = mem_bytes
         for i in gpus:
             g = self.gpus[i]
-            assert g.running is None, \
-                f"GPU {i} busy with {g.running}, placing {job.job_id}"
-            g.running = job.job_id
+            assert len(g.running_set) < capacity, \
+                f"GPU {i} full ({g.running_set}), placing {job.job_id}"
+            g.running_set.add(job.job_id)
             if job.job_id not in g.resident:
                 g.resident.add(job.job_id)
                 g.resident_bytes += mem_bytes
@@ -143,8 +158,7 @@ class Cluster:
         """Stop executing; keep resident unless ``evict``."""
         for i in job.placement:
             g = self.gpus[i]
-            if g.running == job.job_id:
-                g.running = None
+            g.running_set.discard(job.job_id)
             if evict and job.job_id in g.resident:
                 g.resident.remove(job.job_id)
                 g.resident_bytes -= self._mem.get(job.job_id, 0)
@@ -152,7 +166,7 @@ class Cluster:
     def evict_resident(self, job_id: str) -> None:
         """Drop a (non-running) job's resident footprint everywhere."""
         for g in self.gpus:
-            if g.running != job_id and job_id in g.resident:
+            if job_id not in g.running_set and job_id in g.resident:
                 g.resident.remove(job_id)
                 g.resident_bytes -= self._mem.get(job_id, 0)
 

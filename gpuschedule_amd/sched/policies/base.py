@@ -28,6 +28,10 @@ class Policy:
     migration = False
     #: Elastic jobs may shrink below desired width to fit.
     elastic = False
+    #: Gandiva packing: allow co-running a second 1-GPU job on a busy
+    #: GPU (separate HIP streams) when no idle GPU exists.
+    packing = False
+    pack_capacity = 2
 
     def rank(self, now: float, jobs: Dict[str, Job]) -> List[Request]:
         """Return active jobs in descending priority with desired

@@ -25,8 +25,11 @@ class GandivaPolicy(Policy):
     preemptive = True
     migration = True
 
-    def __init__(self, quantum: float = 60.0):
+    def __init__(self, quantum: float = 60.0, packing: bool = False,
+                 pack_capacity: int = 2):
         self.quantum = quantum
+        self.packing = packing
+        self.pack_capacity = pack_capacity
         self._slice_start: Dict[str, float] = {}
         self._next_rotate = 0.0
 

@@ -147,7 +147,9 @@ class Scheduler:
     def _allocate_pass(self, requests, stable: bool):
         """One greedy pass in rank order over a shadow cluster."""
         C = self.cluster
-        busy: Set[int] = set()
+        cap = self.policy.pack_capacity if self.policy.packing else 1
+        slots: Dict[int, int] = {}  # gpu -> allocated run slots
+        busy = slots  # truthiness helper below uses slot counts
         # Shadow per-GPU memory delta: newly placed non-resident jobs
         # add, evictions subtract.
         extra: Dict[int, int] = {}
@@ -155,13 +157,17 @@ class Scheduler:
         alloc: Dict[str, List[int]] = {}
         failed_gang = False
 
+        def occupy(ids):
+            for i in ids:
+                slots[i] = slots.get(i, 0) + 1
+
         pinned: Set[str] = set()
         if not self.policy.preemptive:
             # Running jobs cannot be preempted: pin them first.
             for j in self.jobs.values():
                 if j.state == JobState.RUNNING:
                     alloc[j.job_id] = list(j.placement)
-                    busy.update(j.placement)
+                    occupy(j.placement)
                     pinned.add(j.job_id)
 
         def free_mem(g: int, job: Job) -> int:
@@ -187,8 +193,8 @@ class Scheduler:
             return out
 
         def try_window(ids: List[int], job: Job, mem: int,
-                       allow_evict: bool) -> bool:
-            if any(i in busy for i in ids):
+                       allow_evict: bool, max_slots: int = 1) -> bool:
+            if any(slots.get(i, 0) >= max_slots for i in ids):
                 return False
             need_evict: List[str] = []
             for i in ids:
@@ -224,13 +230,13 @@ class Scheduler:
                     and try_window(list(job.placement), job, mem,
                                    allow_evict)):
                 alloc[job.job_id] = list(job.placement)
-                busy.update(job.placement)
+                occupy(job.placement)
                 return True
             # Preference 2: resume where resident (suspended in HBM).
             if stable and job.placement and len(job.placement) == width:
                 if try_window(list(job.placement), job, mem, allow_evict):
                     alloc[job.job_id] = list(job.placement)
-                    busy.update(job.placement)
+                    occupy(job.placement)
                     return True
             # Preference 3: best-fit contiguous window (least free mem).
             best, best_score = None, None
@@ -244,24 +250,37 @@ class Scheduler:
                     if try_window(ids, job, mem, allow_evict=True):
                         best = ids
                         break
+            # Preference 4 (Gandiva packing): co-run a 1-GPU job on the
+            # least-loaded busy GPU with memory headroom.
+            if best is None and cap > 1 and width == 1:
+                cands = sorted(
+                    (i for i in range(C.n_gpus)
+                     if slots.get(i, 0) < cap),
+                    key=lambda i: (slots.get(i, 0), -free_mem(i, job)))
+                for i in cands:
+                    if try_window([i], job, mem, allow_evict=False,
+                                  max_slots=cap):
+                        best = [i]
+                        break
             if best is None:
                 return False
             alloc[job.job_id] = best
-            busy.update(best)
+            occupy(best)
             return True
 
         for job_id, width in requests:
             job = self.jobs.get(job_id)
             if job is None or not job.is_active or job_id in pinned:
                 continue
-            if len(busy) >= C.n_gpus:
+            if sum(slots.values()) >= C.n_gpus * cap:
                 break
             widths = [width]
             if self.policy.elastic and job.spec.elastic:
                 widths = list(range(width, job.spec.min_gpus - 1, -1))
             ok = False
+            n_full = sum(1 for v in slots.values() if v >= 1)
             for w in widths:
-                if w > C.n_gpus or len(busy) + w > C.n_gpus:
+                if w > C.n_gpus or (w > 1 and n_full + w > C.n_gpus):
                     continue
                 if place(job, w, allow_evict=True):
                     ok = True
@@ -334,7 +353,9 @@ class Scheduler:
                 job.resizes += 1
                 self.log.emit(now, "resize", p.job_id,
                               old_gpus=p.resize_from, new_gpus=len(p.gpus))
-            C.place(job, p.gpus, mem)
+            C.place(job, p.gpus, mem,
+                    capacity=(self.policy.pack_capacity
+                              if self.policy.packing else 1))
             job.placement = list(p.gpus)
             job.state = JobState.RUNNING
             job.last_sched_t = now

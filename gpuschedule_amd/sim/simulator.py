@@ -77,8 +77,15 @@ class Simulator:
         arr = sched.next_arrival()
         if arr is not None:
             horizon = min(horizon, max(arr, self.now + 1e-9))
-        for j in running:
+        def eff_iter_time(j):
             it = catalog.sim_iter_time(j.spec, len(j.placement))
+            if any(len(sched.cluster.gpu(g).running_set) > 1
+                   for g in j.placement):
+                it *= catalog.PACK_SLOWDOWN
+            return it
+
+        for j in running:
+            it = eff_iter_time(j)
             left = (j.remaining_iters - self._frac.get(j.job_id, 0.0)) * it
             fin = self.now + self._overhead.get(j.job_id, 0.0) + max(left, 0)
             horizon = min(horizon, fin)
@@ -100,7 +107,7 @@ class Simulator:
                 avail -= used
             if avail <= 0:
                 continue
-            it = catalog.sim_iter_time(j.spec, len(j.placement))
+            it = eff_iter_time(j)
             frac = self._frac.get(j.job_id, 0.0) + avail / it + 1e-9
             whole = min(int(frac), j.remaining_iters)
             self._frac[j.job_id] = max(frac - int(frac) - 1e-9, 0.0)

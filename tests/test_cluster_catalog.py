@@ -50,7 +50,7 @@ def test_contiguous_window_within_node():
     c = Cluster(16, 8)
     # a 4-gang must not straddle the node boundary (gpus 6..9 invalid)
     for i in (5, 6, 7, 10, 11, 12):
-        c.gpu(i).running = "x"
+        c.gpu(i).running_set.add("x")
     got = c.find_placement(4, 1 * GB)
     assert got is not None
     s = got[0]

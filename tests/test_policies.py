@@ -152,3 +152,28 @@ def test_cluster_overcommit_guard():
         assert False, "expected over-commit assertion"
     except AssertionError:
         pass
+
+
+def test_gandiva_packing_co_runs():
+    # With packing, two 1-GPU jobs co-run on the single GPU instead of
+    # time-slicing: makespan ~ PACK_SLOWDOWN * T, not 2T, and no
+    # preemptions are needed.
+    from gpuschedule_amd import catalog
+    pol = make_policy("gandiva", quantum=5.0, packing=True)
+    specs = [J("a", iters=20, hint=1.0), J("b", iters=20, hint=1.0)]
+    log, s = run(specs, pol, n_gpus=1)
+    assert int(s["n_complete"]) == 2
+    assert s["preemptions"] == 0
+    solo = 20 * 1.0
+    assert s["makespan"] < 1.9 * solo  # ~1.7x packed, not 2x serial
+    assert s["makespan"] > 1.4 * solo
+
+
+def test_packing_prefers_idle_gpus():
+    pol = make_policy("gandiva", quantum=5.0, packing=True)
+    specs = [J("a", iters=10, hint=1.0), J("b", iters=10, hint=1.0)]
+    log, s = run(specs, pol, n_gpus=2)
+    # two GPUs available: jobs must spread, not pack
+    starts = {e["job"]: e["gpus"] for e in log.events
+              if e["event"] == "start"}
+    assert starts["a"] != starts["b"]

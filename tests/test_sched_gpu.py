@@ -97,3 +97,27 @@ def test_eviction_restore_cycle_gpu(tmp_path):
               for j in ("e1", "e2")}
     assert sched.jobs["e1"].done_iters >= 10, (states, log.events)
     assert sched.jobs["e2"].done_iters >= 10, (states, log.events)
+
+
+def test_gandiva_packing_on_gpu(tmp_path):
+    # Two jobs co-running on one MI355X via per-job HIP streams.
+    sched = Scheduler(Cluster(1, 1),
+                      make_policy("gandiva", quantum=60.0, packing=True))
+    specs = [J("p1", iters=10, model="resnet18", batch=64),
+             J("p2", iters=10, model="bert-base", batch=8)]
+    sched.submit(specs)
+    be = TorchBackend(sched, rank=0, world=1, device_type="cuda",
+                      iters_per_tick=2, ckpt_root=str(tmp_path / "ck"))
+    log = be.run(max_seconds=600)
+    s = log.summary()
+    assert int(s["n_complete"]) == 2
+    assert s["preemptions"] == 0  # co-ran, never time-sliced
+    # both were running simultaneously at some point
+    running, overlap = set(), False
+    for e in log.events:
+        if e["event"] == "start":
+            running.add(e["job"])
+            overlap |= len(running & {"p1", "p2"}) == 2
+        elif e["event"] in ("complete", "preempt", "fail"):
+            running.discard(e.get("job"))
+    assert overlap
