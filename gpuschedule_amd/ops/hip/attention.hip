@@ -72,8 +72,12 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
   const int tid = threadIdx.x;
   char* my_p = p_lds + wid * 16 * KVB * 2;
 
-  const int bh = blockIdx.x / ((S + QB - 1) / QB);
-  const int qt = blockIdx.x % ((S + QB - 1) / QB);
+  const int n_qt_g = (S + QB - 1) / QB;
+  const int bh = blockIdx.x / n_qt_g;
+  // causal: later q-tiles do more kv work; launch them FIRST so the
+  // long blocks don't form the tail of the wave
+  int qt = blockIdx.x % n_qt_g;
+  if (CAUSAL) qt = n_qt_g - 1 - qt;
   const int b = bh / H, h = bh % H;
   const int q0 = qt * QB + wid * 16;      // this wave's first q row
 
