@@ -88,6 +88,15 @@ def cmd_gen_trace(args):
     print(f"wrote {len(specs)} jobs -> {args.output}")
 
 
+def cmd_export_trace(args):
+    from .sched.events import EventLog
+    log = EventLog.load(args.log)
+    log.save_chrome_trace(args.output)
+    n = len(log.to_chrome_trace()["traceEvents"])
+    print(f"wrote {n} trace events -> {args.output} "
+          "(open in chrome://tracing or ui.perfetto.dev)")
+
+
 def main(argv=None):
     ap = argparse.ArgumentParser(prog="gpuschedule_amd")
     sub = ap.add_subparsers(dest="cmd", required=True)
@@ -113,6 +122,12 @@ def main(argv=None):
     p.add_argument("--interarrival", type=float, default=10.0)
     p.add_argument("-o", "--output", required=True)
     p.set_defaults(fn=cmd_gen_trace)
+    p = sub.add_parser(
+        "export-trace",
+        help="convert an event-log JSONL to a Chrome/Perfetto trace")
+    p.add_argument("--log", required=True, help="event log JSONL in")
+    p.add_argument("-o", "--output", required=True)
+    p.set_defaults(fn=cmd_export_trace)
 
     args = ap.parse_args(argv)
     args.fn(args)

@@ -9,7 +9,8 @@ std::vector<torch::Tensor> norm_fwd(torch::Tensor x,
 std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                                     c10::optional<torch::Tensor> mean,
                                     torch::Tensor rstd, torch::Tensor gamma,
-                                    bool rms);
+                                    bool rms,
+                                    c10::optional<torch::Tensor> dadd);
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor transpose2d(torch::Tensor x);
@@ -59,11 +60,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "fused LayerNorm/RMSNorm forward (bf16)");
-  m.def("norm_bwd", &norm_bwd, "fused LayerNorm/RMSNorm backward (bf16)");
+  m.def("norm_bwd", &norm_bwd,
+        "fused LayerNorm/RMSNorm backward (bf16); optional dadd is a "
+        "downstream residual-stream grad added into dx in-kernel",
+        py::arg("dy"), py::arg("x"), py::arg("mean"), py::arg("rstd"),
+        py::arg("gamma"), py::arg("rms"),
+        py::arg("dadd") = c10::optional<torch::Tensor>());
   m.def("gemm_nt", &gemm_nt, "MFMA bf16 GEMM C = A @ B^T (+bias, +relu)");
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
   m.def("transpose2d_colsum", &transpose2d_colsum,
-        "bf16 transpose + fp32 column sums (bias grad)");
+        "bf16 transpose + bf16 column sums (bias grad)");
   m.def("adamw_step", &adamw_step, "fused AdamW step");
   m.def("sgd_step", &sgd_step, "fused SGD-momentum step");
   m.def("attn_fwd", &attn_fwd, "flash-attention forward (bf16, online softmax); returns (o, lse)");

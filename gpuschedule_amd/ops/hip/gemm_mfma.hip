@@ -507,12 +507,15 @@ __global__ __launch_bounds__(256) void transpose_kernel(
   }
   __syncthreads();
   if (COLSUM) {
-    // threads 0..63 each own one tile column
+    // threads 0..63 each own one tile column; per-row-tile partials go
+    // to slot row blockIdx.y (plain store — no atomics, and no
+    // zero-fill of the accumulator before the launch); a tiny reduce
+    // kernel folds the R/64 slot rows into the bf16 bias grad.
     if (tid < TT && tc0 + tid < Cc) {
       float s = 0.f;
       int rmax = min(TT, R - tr0);
       for (int r = 0; r < rmax; ++r) s += bf16_to_f32(tile[r][tid]);
-      atomicAdd(colsum + (long)blockIdx.z * Cc + tc0 + tid, s);
+      colsum[(long)blockIdx.y * Cc + tc0 + tid] = s;
     }
   }
   // store: out[c][r]; thread writes 8 contiguous out-cols (= in-rows)
@@ -534,6 +537,17 @@ __global__ __launch_bounds__(256) void transpose_kernel(
       }
     }
   }
+}
+
+// slot partials [P][C] f32 -> bf16 column sums [C] (bias grad)
+__global__ void colsum_reduce_kernel(const float* __restrict__ slots,
+                                     short* __restrict__ db,
+                                     int P, int C) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= C) return;
+  float s = 0.f;
+  for (int p = 0; p < P; ++p) s += slots[(long)p * C + i];
+  db[i] = f32_to_bf16(s);
 }
 
 }  // namespace
@@ -609,18 +623,25 @@ torch::Tensor transpose2d(torch::Tensor x) {
   return out;
 }
 
-// transpose + fp32 column sums (dy -> dy^T and bias grad in one pass)
+// transpose + column sums (dy -> dy^T and bf16 bias grad in one pass)
 std::vector<torch::Tensor> transpose2d_colsum(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   const int R = x.size(0), C = x.size(1);
   auto out = torch::empty({C, R}, x.options());
-  auto cs = torch::zeros({C}, x.options().dtype(torch::kFloat32));
   dim3 grid((C + TT - 1) / TT, (R + TT - 1) / TT, 1);
+  auto slots = torch::empty({(int)grid.y, C},
+                            x.options().dtype(torch::kFloat32));
+  auto db = torch::empty({C}, x.options());
   auto stream = c10::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(transpose_kernel<true>, grid, dim3(256), 0,
                      stream.stream(), (const short*)x.data_ptr(),
-                     (short*)out.data_ptr(), cs.data_ptr<float>(), R, C);
+                     (short*)out.data_ptr(), slots.data_ptr<float>(),
+                     R, C);
+  hipLaunchKernelGGL(colsum_reduce_kernel, dim3((C + 255) / 256),
+                     dim3(256), 0, stream.stream(),
+                     slots.data_ptr<float>(), (short*)db.data_ptr(),
+                     (int)grid.y, C);
   HIP_CHECK_LAST();
-  return {out, cs};
+  return {out, db};
 }

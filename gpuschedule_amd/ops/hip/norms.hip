@@ -178,9 +178,10 @@ template <bool RMS>
 __global__ __launch_bounds__(NT) void norm_bwd_wave_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    const short* __restrict__ gamma, short* __restrict__ dx,
+    const short* __restrict__ gamma, const short* __restrict__ dadd,
+    short* __restrict__ dx,
     float* __restrict__ part_dg, float* __restrict__ part_db,
-    long n_rows, int D) {
+    long pstride, long n_rows, int D) {
   // per-thread dgamma/dbeta accumulators for its fixed columns; the
   // block's 4 waves cover the SAME columns -> merge through LDS once.
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -227,13 +228,17 @@ __global__ __launch_bounds__(NT) void norm_bwd_wave_kernel(
       int i = c * 512 + lane * VEC;
       if (i < D) {
         short8 out;
+        short8 dav;
+        if (dadd)  // fused downstream residual-stream grad add
+          dav = *reinterpret_cast<const short8*>(dadd + row * D + i);
 #pragma unroll
         for (int j = 0; j < VEC; ++j) {
           float d = bf16_to_f32(dyv[c][j]);
           float h = d * bf16_to_f32(gv[c][j]);
           float xn = (bf16_to_f32(xv[c][j]) - mu) * rs;
-          out[j] = f32_to_bf16(RMS ? rs * (h - xn * c2)
-                                   : rs * (h - c1 - xn * c2));
+          float v = RMS ? rs * (h - xn * c2) : rs * (h - c1 - xn * c2);
+          if (dadd) v += bf16_to_f32(dav[j]);
+          out[j] = f32_to_bf16(v);
           acc_dg[c][j] += d * xn;
           acc_db[c][j] += d;
         }
@@ -257,8 +262,8 @@ __global__ __launch_bounds__(NT) void norm_bwd_wave_kernel(
   }
   __syncthreads();
   for (int i = threadIdx.x; i < D; i += NT) {
-    part_dg[(long)blockIdx.x * D + i] = merge[i];
-    if (part_db) part_db[(long)blockIdx.x * D + i] = merge[D + i];
+    part_dg[(long)blockIdx.x * pstride + i] = merge[i];
+    if (part_db) part_db[(long)blockIdx.x * pstride + i] = merge[D + i];
   }
 }
 
@@ -267,9 +272,10 @@ template <int NCHUNK, bool RMS>
 __global__ __launch_bounds__(NT) void norm_bwd_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    const short* __restrict__ gamma, short* __restrict__ dx,
+    const short* __restrict__ gamma, const short* __restrict__ dadd,
+    short* __restrict__ dx,
     float* __restrict__ part_dg, float* __restrict__ part_db,
-    int n_rows, int D) {
+    long pstride, int n_rows, int D) {
   __shared__ float scratch[16];
   float acc_dg[NCHUNK][VEC];
   float acc_db[NCHUNK][VEC];
@@ -312,6 +318,10 @@ __global__ __launch_bounds__(NT) void norm_bwd_kernel(
       int i = c * CHUNK + threadIdx.x * VEC;
       if (i < D) {
         short8 out;
+        short8 dav;
+        if (dadd)  // fused downstream residual-stream grad add
+          dav = *reinterpret_cast<const short8*>(
+              dadd + (long)row * D + i);
 #pragma unroll
         for (int j = 0; j < VEC; ++j) {
           float d = bf16_to_f32(dyv[c][j]);
@@ -319,6 +329,7 @@ __global__ __launch_bounds__(NT) void norm_bwd_kernel(
           float xn = (bf16_to_f32(xv[c][j]) - mu) * rs;
           float v = RMS ? rs * (h - xn * c2)
                         : rs * (h - c1 - xn * c2);
+          if (dadd) v += bf16_to_f32(dav[j]);
           out[j] = f32_to_bf16(v);
           acc_dg[c][j] += d * xn;
           acc_db[c][j] += d;
@@ -333,11 +344,11 @@ __global__ __launch_bounds__(NT) void norm_bwd_kernel(
   for (int c = 0; c < NCHUNK; ++c) {
     int i = c * CHUNK + threadIdx.x * VEC;
     if (i < D) {
-      float* pg = part_dg + (long)blockIdx.x * D + i;
+      float* pg = part_dg + (long)blockIdx.x * pstride + i;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) pg[j] = acc_dg[c][j];
       if (part_db) {
-        float* pb = part_db + (long)blockIdx.x * D + i;
+        float* pb = part_db + (long)blockIdx.x * pstride + i;
 #pragma unroll
         for (int j = 0; j < VEC; ++j) pb[j] = acc_db[c][j];
       }
@@ -345,27 +356,36 @@ __global__ __launch_bounds__(NT) void norm_bwd_kernel(
   }
 }
 
-// partials [P][D] f32 -> f32 accumulator [D] via device-scope atomics
-// (grid covers D x P/PCHUNK: enough blocks to fill the chip; the old
+// partials [P][D2] f32 -> [P2][D2] stage sums -> bf16 [D2] final.
+// Two plain-store passes instead of atomicAdd into a zero-filled
+// accumulator: deterministic, needs no fp32 fill kernel per call, and
+// the final cast fuses the last reduction level.  D2 = 2*D for
+// layernorm (dgamma and dbeta packed side by side in one buffer so the
+// whole reduction is ONE launch pair instead of two).
+// Stage grid covers D2 x P/PCHUNK: enough blocks to fill the chip; a
 // single-pass D/256-block version left 250+ CUs idle and was ~100x
-// slower than the bwd kernel it served).
+// slower than the bwd kernel it served.
 constexpr int PCHUNK = 16;
-__global__ void reduce_partials_kernel(const float* __restrict__ part,
-                                       float* __restrict__ acc,
-                                       int P, int D) {
+__global__ void reduce_partials_stage_kernel(
+    const float* __restrict__ part, float* __restrict__ pp,
+    int P, int D2) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= D) return;
+  if (i >= D2) return;
   int p0 = blockIdx.y * PCHUNK;
   int p1 = min(p0 + PCHUNK, P);
   float s = 0.f;
-  for (int p = p0; p < p1; ++p) s += part[(long)p * D + i];
-  atomicAdd(acc + i, s);
+  for (int p = p0; p < p1; ++p) s += part[(long)p * D2 + i];
+  pp[(long)blockIdx.y * D2 + i] = s;
 }
 
-__global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
-                                     short* __restrict__ out, int n) {
+__global__ void sum_cast_bf16_kernel(const float* __restrict__ pp,
+                                     short* __restrict__ out,
+                                     int P2, int D2) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) out[i] = f32_to_bf16(in[i]);
+  if (i >= D2) return;
+  float s = 0.f;
+  for (int p = 0; p < P2; ++p) s += pp[(long)p * D2 + i];
+  out[i] = f32_to_bf16(s);
 }
 
 int pick_nchunk(int D) {
@@ -475,9 +495,17 @@ std::vector<torch::Tensor> norm_fwd(torch::Tensor x,
 std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                                     c10::optional<torch::Tensor> mean,
                                     torch::Tensor rstd, torch::Tensor gamma,
-                                    bool rms) {
+                                    bool rms,
+                                    c10::optional<torch::Tensor> dadd) {
   check_norm_inputs(x, gamma);
   TORCH_CHECK(dy.is_contiguous() && dy.scalar_type() == torch::kBFloat16);
+  const short* dadd_p = nullptr;
+  if (dadd.has_value() && dadd->defined()) {
+    TORCH_CHECK(dadd->is_contiguous() &&
+                dadd->scalar_type() == torch::kBFloat16 &&
+                dadd->numel() == x.numel());
+    dadd_p = (const short*)dadd->data_ptr();
+  }
   const int D = x.size(-1);
   const long n_rows = x.numel() / D;
   auto dx = torch::empty_like(x);
@@ -486,8 +514,12 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
   if (D <= 1024)
     grid = (int)std::min<long>((n_rows + 3) / 4, 512);
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto part_dg = torch::empty({grid, D}, fopt);
-  auto part_db = rms ? torch::Tensor() : torch::empty({grid, D}, fopt);
+  // dgamma partials at column 0..D-1, dbeta partials (LN) at D..2D-1:
+  // one buffer, one reduction pass for both.
+  const long D2 = rms ? D : 2L * D;
+  auto part = torch::empty({grid, D2}, fopt);
+  float* part_dg_p = part.data_ptr<float>();
+  float* part_db_p = rms ? nullptr : part_dg_p + D;
   auto stream = c10::cuda::getCurrentCUDAStream();
   if (D <= 1024) {
     if (rms) {
@@ -496,9 +528,9 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                          (const short*)dy.data_ptr(),
                          (const short*)x.data_ptr(), nullptr,
                          rstd.data_ptr<float>(),
-                         (const short*)gamma.data_ptr(),
+                         (const short*)gamma.data_ptr(), dadd_p,
                          (short*)dx.data_ptr(),
-                         part_dg.data_ptr<float>(), nullptr, n_rows, D);
+                         part_dg_p, nullptr, D2, n_rows, D);
     } else {
       hipLaunchKernelGGL((norm_bwd_wave_kernel<false>), dim3(grid),
                          dim3(NT), 2 * D * 4, stream.stream(),
@@ -506,10 +538,9 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                          (const short*)x.data_ptr(),
                          mean->data_ptr<float>(),
                          rstd.data_ptr<float>(),
-                         (const short*)gamma.data_ptr(),
+                         (const short*)gamma.data_ptr(), dadd_p,
                          (short*)dx.data_ptr(),
-                         part_dg.data_ptr<float>(),
-                         part_db.data_ptr<float>(), n_rows, D);
+                         part_dg_p, part_db_p, D2, n_rows, D);
     }
   } else
   DISPATCH_NCHUNK(nc, rms, {
@@ -518,40 +549,31 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                          0, stream.stream(), (const short*)dy.data_ptr(),
                          (const short*)x.data_ptr(), nullptr,
                          rstd.data_ptr<float>(),
-                         (const short*)gamma.data_ptr(),
-                         (short*)dx.data_ptr(), part_dg.data_ptr<float>(),
-                         nullptr, (int)n_rows, D);
+                         (const short*)gamma.data_ptr(), dadd_p,
+                         (short*)dx.data_ptr(), part_dg_p,
+                         nullptr, D2, (int)n_rows, D);
     } else {
       hipLaunchKernelGGL((norm_bwd_kernel<kNC, false>), dim3(grid), dim3(NT),
                          0, stream.stream(), (const short*)dy.data_ptr(),
                          (const short*)x.data_ptr(),
                          mean->data_ptr<float>(), rstd.data_ptr<float>(),
-                         (const short*)gamma.data_ptr(),
-                         (short*)dx.data_ptr(), part_dg.data_ptr<float>(),
-                         part_db.data_ptr<float>(), (int)n_rows, D);
+                         (const short*)gamma.data_ptr(), dadd_p,
+                         (short*)dx.data_ptr(), part_dg_p,
+                         part_db_p, D2, (int)n_rows, D);
     }
   });
-  auto dgamma = torch::empty({D}, x.options());
-  auto acc_g = torch::zeros({D}, fopt);
-  dim3 rgrid((D + NT - 1) / NT, (grid + PCHUNK - 1) / PCHUNK);
-  int cgrid = (D + NT - 1) / NT;
-  hipLaunchKernelGGL(reduce_partials_kernel, rgrid, dim3(NT), 0,
-                     stream.stream(), part_dg.data_ptr<float>(),
-                     acc_g.data_ptr<float>(), grid, D);
-  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(cgrid), dim3(NT), 0,
-                     stream.stream(), acc_g.data_ptr<float>(),
-                     (short*)dgamma.data_ptr(), D);
-  torch::Tensor dbeta;
-  if (!rms) {
-    dbeta = torch::empty({D}, x.options());
-    auto acc_b = torch::zeros({D}, fopt);
-    hipLaunchKernelGGL(reduce_partials_kernel, rgrid, dim3(NT), 0,
-                       stream.stream(), part_db.data_ptr<float>(),
-                       acc_b.data_ptr<float>(), grid, D);
-    hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(cgrid), dim3(NT), 0,
-                       stream.stream(), acc_b.data_ptr<float>(),
-                       (short*)dbeta.data_ptr(), D);
-  }
+  const int P2 = (grid + PCHUNK - 1) / PCHUNK;
+  auto pp = torch::empty({P2, D2}, fopt);
+  auto gb = torch::empty({D2}, x.options());  // [dgamma | dbeta] bf16
+  dim3 rgrid((int)((D2 + NT - 1) / NT), P2);
+  hipLaunchKernelGGL(reduce_partials_stage_kernel, rgrid, dim3(NT), 0,
+                     stream.stream(), part.data_ptr<float>(),
+                     pp.data_ptr<float>(), grid, (int)D2);
+  hipLaunchKernelGGL(sum_cast_bf16_kernel,
+                     dim3((int)((D2 + NT - 1) / NT)), dim3(NT), 0,
+                     stream.stream(), pp.data_ptr<float>(),
+                     (short*)gb.data_ptr(), P2, (int)D2);
+  auto dgamma = gb.narrow(0, 0, D);
   HIP_CHECK_LAST();
-  return {dx, dgamma, rms ? dgamma : dbeta};
+  return {dx, dgamma, rms ? dgamma : gb.narrow(0, D, D)};
 }

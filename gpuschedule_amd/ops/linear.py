@@ -64,8 +64,7 @@ class _MFMALinearFn(torch.autograd.Function):
         dyp = _pad_rows64(dy)
         xp = _pad_rows64(x2d)
         if ctx.has_bias:
-            dyt, db32 = ext.transpose2d_colsum(dyp)  # [N, Mp] + colsums
-            db = db32.to(dy.dtype)
+            dyt, db = ext.transpose2d_colsum(dyp)  # [N, Mp] + bf16 db
         else:
             dyt = ext.transpose2d(dyp)
             db = None

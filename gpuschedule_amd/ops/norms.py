@@ -37,11 +37,13 @@ class _NormFn(torch.autograd.Function):
         summed, weight, mean, rstd = ctx.saved_tensors
         if dy is None:
             dy = torch.zeros_like(summed)
+        # downstream residual-stream grad (dsum) is added into dx
+        # INSIDE the bwd kernel — no separate elementwise add pass
         dx, dw, db = ext.norm_bwd(dy.contiguous(), summed,
                                   None if ctx.rms else mean, rstd,
-                                  weight, ctx.rms)
-        if dsum is not None:  # residual stream used downstream
-            dx = dx + dsum
+                                  weight, ctx.rms,
+                                  dsum.contiguous()
+                                  if dsum is not None else None)
         dres = dx if ctx.has_res else None
         return dx, dres, dw, (None if ctx.rms else db), None, None
 

@@ -56,6 +56,69 @@ class EventLog:
                     log.events.append(json.loads(line))
         return log
 
+    # ---- visualization -------------------------------------------
+
+    def to_chrome_trace(self) -> dict:
+        """Render the schedule as a Chrome/Perfetto trace (load in
+        chrome://tracing or ui.perfetto.dev): one track (tid) per GPU,
+        one slice per contiguous occupancy interval of a job, instant
+        markers for submits/evictions/migrations/resizes.  Times are
+        microseconds as the format requires."""
+        US = 1e6
+        ev: List[dict] = [
+            {"ph": "M", "name": "process_name", "pid": 0,
+             "args": {"name": "cluster"}},
+        ]
+        open_runs: Dict[str, tuple] = {}   # job -> (t0, gpus)
+        meta: Dict[str, dict] = {}
+        gpus_seen = set()
+        t_max = 0.0
+
+        def close(job: str, t1: float) -> None:
+            t0, gpus = open_runs.pop(job)
+            for g in gpus:
+                ev.append({"ph": "X", "name": job, "pid": 0, "tid": g,
+                           "ts": t0 * US, "dur": max(t1 - t0, 0) * US,
+                           "args": meta.get(job, {})})
+
+        for e in self.events:
+            t = e["t"]
+            t_max = max(t_max, t)
+            kind = e["event"]
+            job = e.get("job", "")
+            if kind == "submit":
+                meta[job] = {"model": e.get("model", ""),
+                             "n_gpus": e.get("n_gpus", 0),
+                             "iters": e.get("iters", 0)}
+                ev.append({"ph": "i", "name": f"submit {job}", "pid": 0,
+                           "tid": "sched", "ts": t * US, "s": "g"})
+            elif kind in ("start", "resume"):
+                if job in open_runs:
+                    close(job, t)
+                gpus = e.get("gpus", [])
+                gpus_seen.update(gpus)
+                open_runs[job] = (t, gpus)
+            elif kind in ("preempt", "complete", "fail"):
+                if job in open_runs:
+                    close(job, t)
+                if kind == "preempt" and e.get("evicted"):
+                    ev.append({"ph": "i", "name": f"evict {job}",
+                               "pid": 0, "tid": "sched", "ts": t * US,
+                               "s": "g"})
+            elif kind in ("migrate", "resize"):
+                ev.append({"ph": "i", "name": f"{kind} {job}", "pid": 0,
+                           "tid": "sched", "ts": t * US, "s": "g"})
+        for job in list(open_runs):
+            close(job, t_max)
+        for g in sorted(gpus_seen):
+            ev.append({"ph": "M", "name": "thread_name", "pid": 0,
+                       "tid": g, "args": {"name": f"GPU {g}"}})
+        return {"traceEvents": ev, "displayTimeUnit": "ms"}
+
+    def save_chrome_trace(self, path: str) -> None:
+        with open(path, "w") as f:
+            json.dump(self.to_chrome_trace(), f)
+
     # ---- metrics -------------------------------------------------
 
     def summary(self) -> Dict[str, float]:

@@ -47,6 +47,23 @@ def test_layernorm_fwd_bwd(shape):
 
 
 @pytest.mark.parametrize("shape", [(2048, 768), (512, 4096)])
+def test_norm_bwd_fused_dadd(shape):
+    # norm_bwd(dy, ..., dadd=t) == norm_bwd(dy, ...)[0] + t (the
+    # residual-stream grad add fused into the dx write)
+    torch.manual_seed(7)
+    N, D = shape
+    x = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+    t = torch.randn(N, D, device="cuda", dtype=torch.bfloat16)
+    _, _, _, rstd = ext().norm_fwd(x, None, g, None, 1e-6, True)
+    dx0, _, _ = ext().norm_bwd(dy, x, None, rstd, g, True)
+    dxa, _, _ = ext().norm_bwd(dy, x, None, rstd, g, True, t)
+    ref = (dx0.float() + t.float())
+    assert mx(dxa, ref) / (ref.abs().max().item() + 1e-3) < 8e-3
+
+
+@pytest.mark.parametrize("shape", [(2048, 768), (512, 4096)])
 def test_rmsnorm_fwd_bwd(shape):
     torch.manual_seed(1)
     N, D = shape

@@ -56,6 +56,28 @@ def test_event_log_jsonl_roundtrip(tmp_path):
     assert loaded.summary() == s
 
 
+def test_chrome_trace_export(tmp_path):
+    specs = make_trace(6, seed=3, mean_interarrival=2.0,
+                       iters_range=(5, 30), gang_sizes=[1, 1, 2])
+    log, s = simulate(specs, make_policy("gandiva", quantum=3.0),
+                      n_gpus=2)
+    assert int(s["n_complete"]) == 6
+    trace = log.to_chrome_trace()
+    ev = trace["traceEvents"]
+    slices = [e for e in ev if e["ph"] == "X"]
+    # every job appears as at least one occupancy slice on a GPU track
+    assert {e["name"] for e in slices} == {sp.job_id for sp in specs}
+    for e in slices:
+        assert e["dur"] >= 0 and e["tid"] in (0, 1)
+    # instant submit markers on the sched track
+    assert sum(1 for e in ev
+               if e["ph"] == "i" and e["name"].startswith("submit")) == 6
+    p = tmp_path / "trace.json"
+    log.save_chrome_trace(str(p))
+    with open(p) as f:
+        assert json.load(f)["traceEvents"]
+
+
 def test_all_jobs_complete_all_policies():
     specs = make_trace(24, mix="philly-slice", seed=11,
                        mean_interarrival=1.0, iters_range=(5, 60))
