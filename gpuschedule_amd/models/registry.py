@@ -21,8 +21,17 @@ from .resnet import resnet18_cifar, resnet50
 
 def _img_batch(shape, classes):
     def make(batch, device, gen):
-        x = torch.randn(batch, *shape, device=device, generator=gen,
-                        dtype=torch.float32)
+        if torch.device(device).type == "cpu":
+            x = torch.randn(batch, *shape, device=device, generator=gen,
+                            dtype=torch.float32)
+        else:
+            # generate DIRECTLY in the compute dtype and NHWC layout:
+            # the trainer's .to(bf16)/.contiguous(channels_last) then
+            # no-op instead of two full-batch copy kernels per iter
+            x = torch.empty(batch, *shape, device=device,
+                            dtype=torch.bfloat16,
+                            memory_format=torch.channels_last)
+            x.normal_(generator=gen)
         y = torch.randint(0, classes, (batch,), device=device,
                           generator=gen)
         return x, y
