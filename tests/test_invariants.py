@@ -1,0 +1,96 @@
+"""Black-box schedule invariants, reconstructed purely from event logs.
+
+For randomized traces across every policy, replay the event stream and
+assert cluster-level safety properties the scheduler must never
+violate, regardless of policy:
+
+* GPU occupancy never exceeds capacity (1 job/GPU; 2 when the policy
+  packs),
+* gang placements are contiguous GPU windows,
+* a job is never running twice concurrently,
+* every completed job ran exactly its specified iterations,
+* preempted/evicted jobs were actually running when preempted.
+"""
+import pytest
+
+from gpuschedule_amd import make_policy, make_trace, simulate
+
+
+POLICIES = {
+    "fifo": lambda: make_policy("fifo"),
+    "srtf": lambda: make_policy("srtf"),
+    "las": lambda: make_policy("las", thresholds=(20.0, 80.0)),
+    "gandiva": lambda: make_policy("gandiva", quantum=5.0),
+    "gandiva-pack": lambda: make_policy("gandiva", quantum=5.0,
+                                        packing=True),
+    "elastic": lambda: make_policy("elastic"),
+}
+
+
+def replay_and_check(log, capacity):
+    running = {}        # job -> list of gpus
+    occupancy = {}      # gpu -> set of jobs
+    for e in log.events:
+        kind, job = e["event"], e.get("job")
+        if kind in ("start", "resume"):
+            assert job not in running, \
+                f"{job} started while already running @ {e['t']}"
+            gpus = e["gpus"]
+            assert gpus == sorted(gpus)
+            # gang contiguity (single-node windows)
+            if len(gpus) > 1:
+                assert gpus == list(range(gpus[0], gpus[0] + len(gpus))), \
+                    f"non-contiguous gang {gpus} for {job}"
+            for g in gpus:
+                occ = occupancy.setdefault(g, set())
+                occ.add(job)
+                assert len(occ) <= capacity, \
+                    f"GPU {g} oversubscribed: {occ} @ {e['t']}"
+            running[job] = gpus
+        elif kind in ("preempt", "complete", "fail"):
+            if kind == "preempt":
+                assert job in running, \
+                    f"preempt of non-running {job} @ {e['t']}"
+            if job in running:
+                for g in running.pop(job):
+                    occupancy[g].discard(job)
+        elif kind == "migrate":
+            # emitted alongside preempt/resume pairs; nothing to track
+            pass
+    return running
+
+
+@pytest.mark.parametrize("policy", sorted(POLICIES))
+@pytest.mark.parametrize("seed", [0, 7])
+def test_schedule_invariants(policy, seed):
+    specs = make_trace(20, mix="philly-slice", seed=seed,
+                       mean_interarrival=2.0, iters_range=(5, 60),
+                       gang_sizes=[1, 1, 1, 2, 4])
+    pol = POLICIES[policy]()
+    log, s = simulate(specs, pol, n_gpus=4)
+    capacity = 2 if getattr(pol, "packing", False) else 1
+    leftover = replay_and_check(log, capacity)
+    assert not leftover, f"jobs still 'running' at end: {leftover}"
+    # all completions ran exactly the specified iterations
+    iters = {sp.job_id: sp.total_iters for sp in specs}
+    completes = [e for e in log.events if e["event"] == "complete"]
+    for e in completes:
+        assert e["iters"] == iters[e["job"]]
+    # nothing lost: every admitted job either completed or failed
+    done = {e["job"] for e in log.events
+            if e["event"] in ("complete", "fail")}
+    assert done == set(iters)
+
+
+def test_elastic_resize_events_consistent():
+    from gpuschedule_amd.trace.synth import make_elastic_trace
+    specs = make_elastic_trace(seed=3, total_iters=200)
+    log, s = simulate(specs, make_policy("elastic"), n_gpus=4)
+    assert int(s["n_complete"]) == len(specs)
+    ids = {sp.job_id for sp in specs}
+    resizes = [e for e in log.events if e["event"] == "resize"]
+    assert resizes, "elastic trace on idle capacity must resize"
+    for e in resizes:
+        assert e["job"] in ids
+        assert len(e["new_gpus"]) >= 1
+        assert e["new_gpus"] == sorted(e["new_gpus"])
