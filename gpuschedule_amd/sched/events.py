@@ -12,7 +12,7 @@ Event kinds and extra fields:
 * ``preempt``   — reason ("policy"|"quantum"|"resize"|"migrate"), evicted
 * ``resume``    — gpus
 * ``migrate``   — src, dst (lists)
-* ``resize``    — old_gpus, new_gpus (lists)
+* ``resize``    — old_gpus, new_gpus (GPU counts before/after)
 * ``complete``  — jct, iters
 * ``fail``      — error
 * ``tick``      — util, running, queued (cluster samples; optional)

@@ -90,7 +90,7 @@ def test_elastic_resize_events_consistent():
     ids = {sp.job_id for sp in specs}
     resizes = [e for e in log.events if e["event"] == "resize"]
     assert resizes, "elastic trace on idle capacity must resize"
-    for e in resizes:
+    for e in resizes:  # old_gpus/new_gpus are GPU counts
         assert e["job"] in ids
-        assert len(e["new_gpus"]) >= 1
-        assert e["new_gpus"] == sorted(e["new_gpus"])
+        assert e["new_gpus"] >= 1 and e["old_gpus"] >= 1
+        assert e["new_gpus"] != e["old_gpus"]
