@@ -539,15 +539,26 @@ __global__ __launch_bounds__(256) void transpose_kernel(
   }
 }
 
-// slot partials [P][C] f32 -> bf16 column sums [C] (bias grad)
+// slot partials [P][C] f32 -> bf16 column sums [C] (bias grad).
+// 64 columns x 16 slot-lanes per 1024-thread block: the P loop is
+// 16-way parallel per column (a serial P loop with C/256 blocks left
+// 290+ CUs idle and ran 4x slower than the transpose it serves).
 __global__ void colsum_reduce_kernel(const float* __restrict__ slots,
                                      short* __restrict__ db,
                                      int P, int C) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= C) return;
+  __shared__ float sh[64][16];
+  const int cl = threadIdx.x & 63, pl = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + cl;
   float s = 0.f;
-  for (int p = 0; p < P; ++p) s += slots[(long)p * C + i];
-  db[i] = f32_to_bf16(s);
+  if (c < C)
+    for (int p = pl; p < P; p += 16) s += slots[(long)p * C + c];
+  sh[cl][pl] = s;
+  __syncthreads();
+  if (pl != 0 || c >= C) return;
+  s = 0.f;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) s += sh[cl][j];
+  db[c] = f32_to_bf16(s);
 }
 
 }  // namespace
@@ -638,8 +649,8 @@ std::vector<torch::Tensor> transpose2d_colsum(torch::Tensor x) {
                      stream.stream(), (const short*)x.data_ptr(),
                      (short*)out.data_ptr(), slots.data_ptr<float>(),
                      R, C);
-  hipLaunchKernelGGL(colsum_reduce_kernel, dim3((C + 255) / 256),
-                     dim3(256), 0, stream.stream(),
+  hipLaunchKernelGGL(colsum_reduce_kernel, dim3((C + 63) / 64),
+                     dim3(1024), 0, stream.stream(),
                      slots.data_ptr<float>(), (short*)db.data_ptr(),
                      (int)grid.y, C);
   HIP_CHECK_LAST();

@@ -356,35 +356,29 @@ __global__ __launch_bounds__(NT) void norm_bwd_kernel(
   }
 }
 
-// partials [P][D2] f32 -> [P2][D2] stage sums -> bf16 [D2] final.
-// Two plain-store passes instead of atomicAdd into a zero-filled
-// accumulator: deterministic, needs no fp32 fill kernel per call, and
-// the final cast fuses the last reduction level.  D2 = 2*D for
-// layernorm (dgamma and dbeta packed side by side in one buffer so the
-// whole reduction is ONE launch pair instead of two).
-// Stage grid covers D2 x P/PCHUNK: enough blocks to fill the chip; a
-// single-pass D/256-block version left 250+ CUs idle and was ~100x
-// slower than the bwd kernel it served.
-constexpr int PCHUNK = 16;
-__global__ void reduce_partials_stage_kernel(
-    const float* __restrict__ part, float* __restrict__ pp,
-    int P, int D2) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= D2) return;
-  int p0 = blockIdx.y * PCHUNK;
-  int p1 = min(p0 + PCHUNK, P);
-  float s = 0.f;
-  for (int p = p0; p < p1; ++p) s += part[(long)p * D2 + i];
-  pp[(long)blockIdx.y * D2 + i] = s;
-}
-
-__global__ void sum_cast_bf16_kernel(const float* __restrict__ pp,
+// partials [P][D2] f32 -> bf16 [D2] in ONE launch (vs the original
+// zero-fill + device-atomic accumulate + cast trio): deterministic
+// plain reads, no fp32 fill kernel per call.  D2 = 2*D for layernorm
+// (dgamma and dbeta packed side by side in one buffer so both grads
+// come out of a single reduction).  64 columns x 16 row-lanes per
+// 1024-thread block keep the P loop 16-way parallel per column — a
+// serial-P version with D2/256 blocks left most CUs idle (same lesson
+// as the BN slot reduce).
+__global__ void sum_cast_bf16_kernel(const float* __restrict__ part,
                                      short* __restrict__ out,
-                                     int P2, int D2) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= D2) return;
+                                     int P, int D2) {
+  __shared__ float sh[64][16];
+  const int cl = threadIdx.x & 63, pl = threadIdx.x >> 6;
+  const int i = blockIdx.x * 64 + cl;
   float s = 0.f;
-  for (int p = 0; p < P2; ++p) s += pp[(long)p * D2 + i];
+  if (i < D2)
+    for (int p = pl; p < P; p += 16) s += part[(long)p * D2 + i];
+  sh[cl][pl] = s;
+  __syncthreads();
+  if (pl != 0 || i >= D2) return;
+  s = 0.f;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) s += sh[cl][j];
   out[i] = f32_to_bf16(s);
 }
 
@@ -562,17 +556,11 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                          part_db_p, D2, (int)n_rows, D);
     }
   });
-  const int P2 = (grid + PCHUNK - 1) / PCHUNK;
-  auto pp = torch::empty({P2, D2}, fopt);
   auto gb = torch::empty({D2}, x.options());  // [dgamma | dbeta] bf16
-  dim3 rgrid((int)((D2 + NT - 1) / NT), P2);
-  hipLaunchKernelGGL(reduce_partials_stage_kernel, rgrid, dim3(NT), 0,
-                     stream.stream(), part.data_ptr<float>(),
-                     pp.data_ptr<float>(), grid, (int)D2);
   hipLaunchKernelGGL(sum_cast_bf16_kernel,
-                     dim3((int)((D2 + NT - 1) / NT)), dim3(NT), 0,
-                     stream.stream(), pp.data_ptr<float>(),
-                     (short*)gb.data_ptr(), P2, (int)D2);
+                     dim3((int)((D2 + 63) / 64)), dim3(1024), 0,
+                     stream.stream(), part.data_ptr<float>(),
+                     (short*)gb.data_ptr(), grid, (int)D2);
   auto dgamma = gb.narrow(0, 0, D);
   HIP_CHECK_LAST();
   return {dx, dgamma, rms ? dgamma : gb.narrow(0, D, D)};

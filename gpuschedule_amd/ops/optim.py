@@ -33,6 +33,22 @@ class _MasterStateMixin:
                                          dtype=torch.float32)
 
 
+def _mt_eligible(p) -> bool:
+    """Param qualifies for the flat multi-tensor kernels: any DENSE
+    layout works (the kernels are elementwise over data_ptr+numel) as
+    long as param and grad share identical strides — channels_last
+    conv weights included (``is_contiguous()`` alone rejected every
+    channels_last ResNet weight, silently forcing the per-tensor
+    path: ~160 optimizer launches/iter instead of 2)."""
+    g = p.grad
+    if g is None or not p.is_cuda:
+        return False
+    if g.stride() != p.stride():
+        return False
+    return (p.is_contiguous() or
+            p.is_contiguous(memory_format=torch.channels_last))
+
+
 class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
     """Fused AdamW.  When every param of a group is a bf16 CUDA tensor
     with a grad, the MULTI-TENSOR path runs: ONE kernel updates all
@@ -69,9 +85,8 @@ class FusedAdamW(_MasterStateMixin, torch.optim.Optimizer):
     def _try_build_mt(self, group):
         params = [p for p in group["params"] if p.requires_grad]
         if not params or not all(
-                p.is_cuda and p.dtype == torch.bfloat16
-                and p.grad is not None and p.grad.is_contiguous()
-                and p.is_contiguous() for p in params):
+                p.dtype == torch.bfloat16 and _mt_eligible(p)
+                for p in params):
             return None
         dev = params[0].device
         ptrs, sizes, ct, co = [], [], [], []
@@ -173,10 +188,8 @@ class FusedSGD(_MasterStateMixin, torch.optim.Optimizer):
     def _try_build_mt(self, group):
         params = [p for p in group["params"] if p.requires_grad]
         if not params or not all(
-                p.is_cuda and p.grad is not None
-                and p.dtype in (torch.bfloat16, torch.float32)
-                and p.is_contiguous() and p.grad.is_contiguous()
-                for p in params):
+                p.dtype in (torch.bfloat16, torch.float32)
+                and _mt_eligible(p) for p in params):
             return None
         dev = params[0].device
         tables = []

@@ -1,23 +1,28 @@
-"""Attribute residual torch-glue kernels (fills/copies/adds) to op sites.
+"""Attribute residual torch-glue kernels (fills/copies/adds) to the
+training-step phase that launches them.
 
-Runs a few eager iterations (graphs disabled so the profiler keeps op
-correlation) of one job per family and prints the op-level aggregation
-of every kernel whose name matches the glue patterns, with source
-stacks.  Diagnostic only — not part of the test suite.
+Runs eager iterations (graphs disabled so op correlation survives)
+with record_function brackets around each phase, then attributes every
+aten::fill_/copy_/add_ call to its enclosing phase by CPU-interval
+containment.  Diagnostic only — not part of the test suite.
 """
 import os
 import sys
+from collections import defaultdict
 
 os.environ["GSAMD_NO_GRAPHS"] = "1"
 
 import torch  # noqa: E402
-from torch.profiler import ProfilerActivity, profile  # noqa: E402
+from torch.profiler import ProfilerActivity, profile, record_function  # noqa: E402
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))))
 from gpuschedule_amd.exec.trainer import JobRunner  # noqa: E402
+from gpuschedule_amd.models import compute_loss  # noqa: E402
 
-GLUE = ("fill", "copy", "Functor_add", "direct_copy")
+GLUE_OPS = ("aten::fill_", "aten::zero_", "aten::zeros", "aten::copy_",
+            "aten::add_", "aten::add", "aten::contiguous", "aten::cat",
+            "aten::to", "aten::_to_copy")
 
 
 def run(model, iters=4):
@@ -26,33 +31,55 @@ def run(model, iters=4):
     for _ in range(3):
         r.step(1)
     torch.cuda.synchronize()
+
+    def one_iter():
+        with record_function("PH_datagen"):
+            x, y = r._batch(r.done_iters)
+        with record_function("PH_forward"):
+            loss = compute_loss(r.model_name, r.model, x, y)
+        with record_function("PH_backward"):
+            loss.backward()
+        with record_function("PH_optim"):
+            r.optim.step()
+        r.done_iters += 1
+
     with profile(activities=[ProfilerActivity.CPU,
-                             ProfilerActivity.CUDA],
-                 with_stack=True) as p:
+                             ProfilerActivity.CUDA]) as p:
         for _ in range(iters):
-            r.step(1)
+            one_iter()
+        torch.cuda.synchronize()
+
+    # phase intervals on each CPU thread
+    phases = []
+    for e in p.events():
+        if e.name.startswith("PH_"):
+            phases.append((e.time_range.start, e.time_range.end,
+                           e.name, e.thread))
+    agg = defaultdict(lambda: [0, 0.0])
+
+    def phase_of(e):
+        for s, en, name, tid in phases:
+            if tid == e.thread and s <= e.time_range.start <= en:
+                return name
+        return "other"
+
+    for e in p.events():
+        if e.name in GLUE_OPS and not e.cpu_parent:
+            k = (phase_of(e), e.name)
+            agg[k][0] += 1
+            agg[k][1] += e.device_time_total
+        elif e.name in GLUE_OPS and e.cpu_parent \
+                and e.cpu_parent.name not in GLUE_OPS \
+                and not e.cpu_parent.name.startswith(("autograd",)):
+            k = (phase_of(e), f"{e.cpu_parent.name} -> {e.name}")
+            agg[k][0] += 1
+            agg[k][1] += e.device_time_total
+
     print(f"\n===== {model} ({iters} iters) =====")
-    ka = p.key_averages(group_by_stack_n=5)
-    rows = []
-    for e in ka:
-        dev_us = e.self_device_time_total
-        if dev_us <= 0:
+    for (ph, op), (n, us) in sorted(agg.items(), key=lambda kv: -kv[1][1]):
+        if us < 50:
             continue
-        rows.append((dev_us, e))
-    rows.sort(key=lambda t: -t[0])
-    total = sum(d for d, _ in rows)
-    print(f"total device time {total/1e3:.2f} ms; ops touching glue:")
-    for dev_us, e in rows:
-        name = e.key
-        if not any(g in name for g in
-                   ("zero", "fill", "copy", "add", "cat", "to")):
-            continue
-        if dev_us / total < 0.002:
-            continue
-        print(f"  {dev_us/1e3:8.2f} ms  n={e.count:5d}  {name[:70]}")
-        if e.stack:
-            for fr in e.stack[:4]:
-                print(f"      {fr[-100:]}")
+        print(f"  {us/1e3:8.2f} ms  n={n:5d}  {ph:12s} {op[:70]}")
     r.release()
     del r
     torch.cuda.empty_cache()

@@ -57,3 +57,19 @@ def test_bench_cpu_contract(tmp_path):
         assert key in res
     assert res["steps"] == 6 and res["n_gpus"] == 1
     assert res["higher_is_better"] is False
+
+
+def test_bench_trace_identical_across_n():
+    """Strong scaling: the trace-job set must not depend on N (only
+    the best-effort fill jobs do)."""
+    sys.path.insert(0, REPO)
+    from bench import build_trace
+    a1, f1 = build_trace(8, 40, 1, 2)
+    a8, f8 = build_trace(8, 40, 8, 2)
+    assert [(t, s.job_id, s.total_iters, s.model, s.n_gpus)
+            for t, s in a1] == \
+           [(t, s.job_id, s.total_iters, s.model, s.n_gpus)
+            for t, s in a8]
+    assert len(f1) == 1 and all(s.background for s in f1)
+    # at N>=2 the first fill is a 2-GPU gang (exercises RCCL all-reduce)
+    assert f8[0].n_gpus == 2 and sum(s.n_gpus for s in f8) == 8
