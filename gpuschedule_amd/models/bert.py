@@ -12,7 +12,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.attention import flash_attention
+from ..ops.attention import flash_attention_qkv
 from ..ops.ce import fused_cross_entropy
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedLayerNorm
@@ -42,9 +42,11 @@ class BertLayer(nn.Module):
 
     def forward(self, x):
         B, S, Hd = x.shape
+        # packed-QKV attention: the kernels consume the projection view
+        # directly (strided q/k/v slices) and the head-merge transpose
+        # below is a free view of the [B,S,H,D] output buffer
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.hd)
-        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        att = flash_attention(q, k, v, causal=False)
+        att = flash_attention_qkv(qkv, causal=False)
         att = att.transpose(1, 2).reshape(B, S, Hd)
         x = self.ln1(self.proj(att), residual=x)  # fused add+LN
         h = F.gelu(self.fc1(x), approximate="tanh")

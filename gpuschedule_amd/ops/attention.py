@@ -47,6 +47,48 @@ class _FlashAttnFn(torch.autograd.Function):
         return dq, dk, dv, None
 
 
+class _FlashAttnQKVFn(torch.autograd.Function):
+    """Packed-QKV self-attention: takes the [B,S,3,H,D] view of one
+    fused projection.  The kernels read the q/k/v slices through their
+    strides and write the three grads into ONE [B,S,3,H,D] buffer, so
+    the whole attention block adds ZERO layout copies (no per-slice
+    contiguous(), no select-backward zero+copy, no output re-pack)."""
+
+    @staticmethod
+    def forward(ctx, qkv, causal):
+        ext = get_ext(required=True)
+        q = qkv[:, :, 0].transpose(1, 2)  # [B,H,S,D] strided views
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        o, lse = ext.attn_fwd(q, k, v, causal)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.causal = causal
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        ext = get_ext(required=True)
+        qkv, o, lse = ctx.saved_tensors
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        if do.stride() != o.stride():  # usually a free view already
+            do = do.permute(0, 2, 1, 3).contiguous().permute(0, 2, 1, 3)
+        out = ext.attn_bwd(q, k, v, o, do, lse, ctx.causal)
+        return out[3], None  # the packed [B,S,3,H,D] grad
+
+
+def flash_attention_qkv(qkv: torch.Tensor,
+                        causal: bool = False) -> torch.Tensor:
+    """qkv: [B, S, 3, H, D] (contiguous, e.g. a fused-projection view).
+    Returns O as logical [B, H, S, D]; ``o.transpose(1, 2)`` is a free
+    view for the head-merge reshape."""
+    if qkv.is_cuda:
+        return _FlashAttnQKVFn.apply(qkv.contiguous(), causal)
+    q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+    return _ref_attention(q, k, v, causal)
+
+
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     causal: bool = False) -> torch.Tensor:
     """q,k,v: [B, H, S, D] bf16 (k/v may have fewer heads: GQA)."""

@@ -48,11 +48,18 @@ __device__ __forceinline__ float group_sum(float v) {
   return v;
 }
 
+// Q/K and O are addressed through (batch, head, row) stride triples so
+// strided views work directly — e.g. the packed-QKV path (q/k/v are
+// slices of one [B,S,3,H,D] projection, O written as a [B,S,H,D] view)
+// runs with ZERO layout copies.  Contiguous [B,H,S,D] is the special
+// case (sB=H*S*D, sH=S*D, sS=D).  Vt is an internal contiguous
+// scratch either way.
 template <int D, bool CAUSAL, int NW>
 __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ Vt, short* __restrict__ O,
-    float* __restrict__ LSE, int B, int H, int S, float scale) {
+    float* __restrict__ LSE, int B, int H, int S, float scale,
+    long qsB, long qsH, long qsS, long osB, long osH, long osS) {
   constexpr int NT = NW * 64;      // threads (NW waves x 16 q rows)
   constexpr int QB = NW * 16;      // q rows per block
   constexpr int DF = D / 16;       // PV col fragments
@@ -81,7 +88,8 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
   const int b = bh / H, h = bh % H;
   const int q0 = qt * QB + wid * 16;      // this wave's first q row
 
-  const long qoff = ((long)(b * H + h) * S) * D;
+  const long qb = (long)b * qsB + (long)h * qsH;
+  const long ob = (long)b * osB + (long)h * osH;
   const long voff = ((long)(b * H + h) * D) * S;  // Vt is [D][S]
 
   // ---- Q into registers (A-fragment layout), pre-scaled ----------
@@ -94,7 +102,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
       int d0 = c * 32 + (lane >> 4) * 8;
       if (ok) {
         bf16x8 v = *reinterpret_cast<const bf16x8*>(
-            Q + qoff + (long)row * D + d0);
+            Q + qb + (long)row * qsS + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           v[j] = (__bf16)((float)v[j] * scale);
@@ -133,7 +141,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
       int kvr = kv0 + r;
       if (kvr < S) {
         kst[p] = *reinterpret_cast<const short8*>(
-            K + qoff + (long)kvr * D + cb / 2);
+            K + qb + (long)kvr * qsS + cb / 2);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) kst[p][j] = 0;
@@ -280,7 +288,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
     int row = orow0 + r;
     if (row >= S) continue;
     float inv = (l[r] > 0.f) ? 1.0f / l[r] : 0.f;
-    short* orow = O + qoff + (long)row * D;
+    short* orow = O + ob + (long)row * osS;
 #pragma unroll
     for (int f = 0; f < DF; ++f)
       orow[f * 16 + (lane & 15)] = f32_to_bf16(o_acc[f][r] * inv);
@@ -334,18 +342,22 @@ __device__ __forceinline__ void glds_tile(
   }
 }
 
-// Drow = rowsum(dO * O): one 64-lane wave per row.
+// Drow = rowsum(dO * O): one 64-lane wave per row.  Drow stays in
+// (b, h, s) order regardless of the O/dO layout (stride triple).
 __global__ __launch_bounds__(256) void attn_drow_kernel(
     const short* __restrict__ dO, const short* __restrict__ O,
-    float* __restrict__ Drow, long n_rows, int D) {
+    float* __restrict__ Drow, long n_rows, int H, int S, int D,
+    long osB, long osH, long osS) {
   long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= n_rows) return;
   const int lane = threadIdx.x & 63;
+  const int s_i = (int)(row % S);
+  const long bh = row / S;
+  const long base = (bh / H) * osB + (bh % H) * osH + (long)s_i * osS;
   float s = 0.f;
   for (int i = lane * 2; i + 1 < D; i += 128) {
-    s += bf16_to_f32(dO[row * D + i]) * bf16_to_f32(O[row * D + i]);
-    s += bf16_to_f32(dO[row * D + i + 1]) *
-         bf16_to_f32(O[row * D + i + 1]);
+    s += bf16_to_f32(dO[base + i]) * bf16_to_f32(O[base + i]);
+    s += bf16_to_f32(dO[base + i + 1]) * bf16_to_f32(O[base + i + 1]);
   }
   s = wave_reduce_sum(s);
   if (lane == 0) Drow[row] = s;
@@ -358,7 +370,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     const short* __restrict__ Qt, const short* __restrict__ dOt,
     const float* __restrict__ LSE, const float* __restrict__ DR,
     short* __restrict__ dK, short* __restrict__ dV,
-    int B, int H, int S, float scale) {
+    int B, int H, int S, float scale,
+    long qsB, long qsH, long qsS, long osB, long osH, long osS) {
   constexpr int DF = D / 16;
   constexpr int KC = D / 32;
   constexpr int RB = D * 2;
@@ -388,7 +401,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   const int kv0 = kvt * KVB;
   const int krow = kv0 + wid * 16 + (lane & 15);
 
-  const long off = ((long)(b * H + h) * S) * D;
+  const long qvb = (long)b * qsB + (long)h * qsH;   // Q/K/V + d* base
+  const long dob = (long)b * osB + (long)h * osH;   // dO base
   const long offT = ((long)(b * H + h) * D) * S;
   const long roff = (long)(b * H + h) * S;
 
@@ -400,13 +414,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     for (int c = 0; c < KC; ++c) {
       int d0 = c * 32 + (lane >> 4) * 8;
       bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
-          K + off + (long)row * D + d0);
+          K + qvb + (long)row * qsS + d0);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         kv8[j] = (__bf16)((float)kv8[j] * scale);
       kreg[c] = kv8;
       vreg[c] = *reinterpret_cast<const bf16x8*>(
-          V + off + (long)row * D + d0);
+          V + qvb + (long)row * qsS + d0);
     }
   }
 
@@ -421,8 +435,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   const int n_qt = (S + 63) / 64;
   for (int qt = qt_start; qt < n_qt; ++qt) {
     const int q0 = qt * 64;
-    glds_tile<KVB, RB>(Q + off, D, q0, S, 0, D, q_lds, wid, lane);
-    glds_tile<KVB, RB>(dO + off, D, q0, S, 0, D, do_lds, wid, lane);
+    glds_tile<KVB, RB>(Q + qvb, qsS, q0, S, 0, D, q_lds, wid, lane);
+    glds_tile<KVB, RB>(dO + dob, osS, q0, S, 0, D, do_lds, wid, lane);
     glds_tile<D, KVB * 2>(Qt + offT, S, 0, D, q0, S, qt_lds, wid, lane);
     glds_tile<D, KVB * 2>(dOt + offT, S, 0, D, q0, S, dot_lds, wid, lane);
     for (int i = tid; i < KVB; i += NTHREADS) {
@@ -505,8 +519,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   for (int r = 0; r < 4; ++r) {
     int row = orow0 + r;
     if (row >= S) continue;
-    short* dkr = dK + off + (long)row * D;
-    short* dvr = dV + off + (long)row * D;
+    short* dkr = dK + qvb + (long)row * qsS;
+    short* dvr = dV + qvb + (long)row * qsS;
 #pragma unroll
     for (int f = 0; f < DF; ++f) {
       dkr[f * 16 + (lane & 15)] = f32_to_bf16(dk_acc[f][r]);
@@ -521,7 +535,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
     const short* __restrict__ V, const short* __restrict__ dO,
     const short* __restrict__ Kt, const float* __restrict__ LSE,
     const float* __restrict__ DR, short* __restrict__ dQ,
-    int B, int H, int S, float scale) {
+    int B, int H, int S, float scale,
+    long qsB, long qsH, long qsS, long osB, long osH, long osS) {
   constexpr int DF = D / 16;
   constexpr int KC = D / 32;
   constexpr int RB = D * 2;
@@ -544,7 +559,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
   const int b = bh / H, h = bh % H;
   const int q0 = qt * 64 + wid * 16;
 
-  const long off = ((long)(b * H + h) * S) * D;
+  const long qvb = (long)b * qsB + (long)h * qsH;
+  const long dob = (long)b * osB + (long)h * osH;
   const long offT = ((long)(b * H + h) * D) * S;
   const long roff = (long)(b * H + h) * S;
 
@@ -557,13 +573,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
     for (int c = 0; c < KC; ++c) {
       int d0 = c * 32 + (lane >> 4) * 8;
       bf16x8 v8 = *reinterpret_cast<const bf16x8*>(
-          Q + off + (long)row * D + d0);
+          Q + qvb + (long)row * qsS + d0);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         v8[j] = (__bf16)((float)v8[j] * scale);
       qreg[c] = v8;
       doreg[c] = *reinterpret_cast<const bf16x8*>(
-          dO + off + (long)row * D + d0);
+          dO + dob + (long)row * osS + d0);
     }
   }
   // per-lane lse/Drow for this lane's 4 C-rows
@@ -588,8 +604,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
       : (S + KVB - 1) / KVB;
   for (int t = 0; t < n_kv; ++t) {
     const int kv0 = t * KVB;
-    glds_tile<KVB, RB>(K + off, D, kv0, S, 0, D, k_lds, wid, lane);
-    glds_tile<KVB, RB>(V + off, D, kv0, S, 0, D, v_lds, wid, lane);
+    glds_tile<KVB, RB>(K + qvb, qsS, kv0, S, 0, D, k_lds, wid, lane);
+    glds_tile<KVB, RB>(V + qvb, qsS, kv0, S, 0, D, v_lds, wid, lane);
     glds_tile<D, KVB * 2>(Kt + offT, S, 0, D, kv0, S, kt_lds, wid, lane);
     __syncthreads();
 
@@ -652,7 +668,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
   for (int r = 0; r < 4; ++r) {
     int row = orow0 + r;
     if (row >= S) continue;
-    short* dqr = dQ + off + (long)row * D;
+    short* dqr = dQ + qvb + (long)row * qsS;
 #pragma unroll
     for (int f = 0; f < DF; ++f)
       dqr[f * 16 + (lane & 15)] = f32_to_bf16(dq_acc[f][r]);
@@ -662,23 +678,46 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
 }  // namespace
 
 torch::Tensor transpose2d(torch::Tensor x);  // gemm_mfma.hip (batched)
+torch::Tensor transpose2d_bhsd(torch::Tensor x);
+
+namespace {
+// q/k/v must be contiguous OR share one (b,h,s) stride triple with
+// unit d-stride and 16-byte-aligned rows — exactly what slices of a
+// packed [B,S,3,H,D] QKV projection satisfy.
+void check_qkv_strides(const torch::Tensor& q, const torch::Tensor& k,
+                       const torch::Tensor& v) {
+  TORCH_CHECK(q.stride(3) == 1 && k.strides() == q.strides() &&
+              v.strides() == q.strides(),
+              "attn: q/k/v must share one stride layout (d contiguous)");
+  TORCH_CHECK(q.stride(2) % 8 == 0 && q.stride(1) % 8 == 0 &&
+              q.stride(0) % 8 == 0,
+              "attn: strides must keep rows 16B-aligned");
+}
+}  // namespace
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.dim() == 4, "attn_fwd: [B,H,S,D]");
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes(),
               "attn_fwd: expand KV heads before calling (GQA)");
   TORCH_CHECK(D == 64 || D == 128, "attn_fwd: D must be 64 or 128");
   TORCH_CHECK(S % 8 == 0, "attn_fwd: S must be a multiple of 8");
+  const bool cont = q.is_contiguous() && k.is_contiguous() &&
+                    v.is_contiguous();
+  if (!cont) check_qkv_strides(q, k, v);
   // pre-transpose V: [B,H,S,D] -> [B,H,D,S] (one memory-bound pass on
   // the batched LDS-tiled transpose; lets the PV B-fragments read
   // 8 bf16 contiguous along kv)
-  auto vt = transpose2d(v.reshape({(long)B * H, S, D}))
-                .reshape({B, H, D, S});
-  auto o = torch::empty_like(q);
+  auto vt = cont ? transpose2d(v.reshape({(long)B * H, S, D}))
+                       .reshape({B, H, D, S})
+                 : transpose2d_bhsd(v);
+  // strided (packed-QKV) path: O goes out as a [B,S,H,D] buffer viewed
+  // [B,H,S,D], so the caller's head-merge reshape is a free view
+  auto o = cont ? torch::empty({B, H, S, D}, q.options())
+                : torch::empty({B, S, H, D}, q.options())
+                      .permute({0, 2, 1, 3});
   auto lse = torch::empty({B, H, S},
                           q.options().dtype(torch::kFloat32));
   float scale = 1.0f / sqrtf((float)D);
@@ -704,7 +743,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                      (const short*)k.data_ptr(),                         \
                      (const short*)vt.data_ptr(),                        \
                      (short*)o.data_ptr(), lse.data_ptr<float>(),        \
-                     B, H, S, scale)
+                     B, H, S, scale, q.stride(0), q.stride(1),           \
+                     q.stride(2), o.stride(0), o.stride(1), o.stride(2))
 #define LAUNCH_D(DD)                                                     \
   do {                                                                   \
     if (causal) { if (wide) LAUNCH(DD, true, 8);                         \
@@ -719,27 +759,51 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+// Contiguous q/k/v: returns {dq, dk, dv} (each contiguous).  Strided
+// (packed-QKV) q/k/v: allocates ONE [B,S,3,H,D] buffer, writes the
+// three grads into it as strided views, and returns
+// {dq, dk, dv, dqkv} — the caller hands dqkv straight back as the
+// packed projection's grad (no select-backward zeros/copies, no cat).
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     torch::Tensor dout,
                                     torch::Tensor lse, bool causal) {
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
-  TORCH_CHECK(dout.is_contiguous() && o.is_contiguous());
+  const bool cont = q.is_contiguous() && k.is_contiguous() &&
+                    v.is_contiguous();
+  if (!cont) check_qkv_strides(q, k, v);
+  TORCH_CHECK(o.stride(3) == 1 && dout.strides() == o.strides(),
+              "attn_bwd: dout must match o's layout");
   auto stream = c10::cuda::getCurrentCUDAStream();
   // pre-transposes (batched LDS-tiled kernel) + Drow
-  auto qt = transpose2d(q.reshape({(long)B * H, S, D}));
-  auto kt = transpose2d(k.reshape({(long)B * H, S, D}));
-  auto dot = transpose2d(dout.reshape({(long)B * H, S, D}));
+  auto qt = cont ? transpose2d(q.reshape({(long)B * H, S, D}))
+                 : transpose2d_bhsd(q);
+  auto kt = cont ? transpose2d(k.reshape({(long)B * H, S, D}))
+                 : transpose2d_bhsd(k);
+  auto dot = dout.is_contiguous()
+                 ? transpose2d(dout.reshape({(long)B * H, S, D}))
+                 : transpose2d_bhsd(dout);
   long n_rows = (long)B * H * S;
   auto drow = torch::empty({n_rows}, q.options().dtype(torch::kFloat32));
   hipLaunchKernelGGL(attn_drow_kernel,
                      dim3((int)((n_rows + 3) / 4)), dim3(256), 0,
                      stream.stream(), (const short*)dout.data_ptr(),
                      (const short*)o.data_ptr(), drow.data_ptr<float>(),
-                     n_rows, D);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+                     n_rows, H, S, D, o.stride(0), o.stride(1),
+                     o.stride(2));
+  torch::Tensor dq, dk, dv, dqkv;
+  if (cont) {
+    dq = torch::empty({B, H, S, D}, q.options());
+    dk = torch::empty({B, H, S, D}, q.options());
+    dv = torch::empty({B, H, S, D}, q.options());
+  } else {
+    dqkv = torch::empty({B, S, 3, H, D}, q.options());
+    dq = dqkv.select(2, 0).permute({0, 2, 1, 3});
+    dk = dqkv.select(2, 1).permute({0, 2, 1, 3});
+    dv = dqkv.select(2, 2).permute({0, 2, 1, 3});
+    TORCH_CHECK(dq.strides() == q.strides(),
+                "attn_bwd: packed grads must mirror the input layout");
+  }
   float scale = 1.0f / sqrtf((float)D);
   int grid_kv = B * H * ((S + KVB - 1) / KVB);
   int grid_q = B * H * ((S + 63) / 64);
@@ -755,7 +819,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        (const short*)dot.data_ptr(),                    \
                        lse.data_ptr<float>(), drow.data_ptr<float>(),   \
                        (short*)dk.data_ptr(), (short*)dv.data_ptr(),    \
-                       B, H, S, scale);                                 \
+                       B, H, S, scale, q.stride(0), q.stride(1),        \
+                       q.stride(2), o.stride(0), o.stride(1),           \
+                       o.stride(2));                                    \
     hipLaunchKernelGGL((attn_bwd_q_kernel<DD, CC>), dim3(grid_q),       \
                        dim3(NTHREADS), 0, stream.stream(),              \
                        (const short*)q.data_ptr(),                      \
@@ -764,7 +830,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        (const short*)dout.data_ptr(),                   \
                        (const short*)kt.data_ptr(),                     \
                        lse.data_ptr<float>(), drow.data_ptr<float>(),   \
-                       (short*)dq.data_ptr(), B, H, S, scale);          \
+                       (short*)dq.data_ptr(), B, H, S, scale,           \
+                       q.stride(0), q.stride(1), q.stride(2),           \
+                       o.stride(0), o.stride(1), o.stride(2));          \
   } while (0)
   if (D == 64) {
     if (causal) LAUNCH_BWD(64, true); else LAUNCH_BWD(64, false);
@@ -773,5 +841,6 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   }
 #undef LAUNCH_BWD
   HIP_CHECK_LAST();
-  return {dq, dk, dv};
+  if (cont) return {dq, dk, dv};
+  return {dq, dk, dv, dqkv};
 }

@@ -14,6 +14,7 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor transpose2d(torch::Tensor x);
+torch::Tensor transpose2d_bhsd(torch::Tensor x);
 std::vector<torch::Tensor> transpose2d_colsum(torch::Tensor x);
 void adamw_step(torch::Tensor p_bf16, torch::Tensor p32, torch::Tensor grad,
                 torch::Tensor m, torch::Tensor v, double lr, double beta1,
@@ -68,6 +69,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dadd") = c10::optional<torch::Tensor>());
   m.def("gemm_nt", &gemm_nt, "MFMA bf16 GEMM C = A @ B^T (+bias, +relu)");
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
+  m.def("transpose2d_bhsd", &transpose2d_bhsd,
+        "strided [B,H,S,D] -> contiguous [B,H,D,S] bf16 transpose");
   m.def("transpose2d_colsum", &transpose2d_colsum,
         "bf16 transpose + bf16 column sums (bias grad)");
   m.def("adamw_step", &adamw_step, "fused AdamW step");

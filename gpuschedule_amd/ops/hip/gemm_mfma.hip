@@ -539,6 +539,59 @@ __global__ __launch_bounds__(256) void transpose_kernel(
   }
 }
 
+// strided-input batched transpose: logical [B,H,R,C] tensor with
+// arbitrary (b,h,row) strides (col stride 1) -> contiguous
+// [B*H, C, R].  Lets attention pre-transpose Q/K/V/dO slices of a
+// packed projection without materializing contiguous copies first.
+__global__ __launch_bounds__(256) void transpose_bhsd_kernel(
+    const short* __restrict__ in0, short* __restrict__ out0,
+    int R, int Cc, int H, long sB, long sH, long sR) {
+  const long z = blockIdx.z;
+  const short* in = in0 + (z / H) * sB + (z % H) * sH;
+  short* out = out0 + z * (long)R * Cc;
+  __shared__ short tile[TT][TT + 8];
+  int tr0 = blockIdx.y * TT, tc0 = blockIdx.x * TT;
+  int tid = threadIdx.x;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int r = p * 32 + (tid >> 3);
+    int c = (tid & 7) * 8;
+    int gr = tr0 + r, gc = tc0 + c;
+    if (gr < R) {
+      if (gc + 7 < Cc) {
+        short8 v = *reinterpret_cast<const short8*>(
+            in + (long)gr * sR + gc);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) tile[r][c + j] = v[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          tile[r][c + j] = (gc + j < Cc)
+              ? in[(long)gr * sR + gc + j] : 0;
+      }
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int c = p * 32 + (tid >> 3);
+    int r = (tid & 7) * 8;
+    int gor = tc0 + c, goc = tr0 + r;
+    if (gor < Cc) {
+      if (goc + 7 < R) {
+        short8 v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = tile[r + j][c];
+        *reinterpret_cast<short8*>(out + (long)gor * R + goc) = v;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (goc + j < R) out[(long)gor * R + goc + j] = tile[r + j][c];
+      }
+    }
+  }
+}
+
 // slot partials [P][C] f32 -> bf16 column sums [C] (bias grad).
 // 64 columns x 16 slot-lanes per 1024-thread block: the P loop is
 // 16-way parallel per column (a serial P loop with C/256 blocks left
@@ -630,6 +683,23 @@ torch::Tensor transpose2d(torch::Tensor x) {
   hipLaunchKernelGGL(transpose_kernel<false>, grid, dim3(256), 0,
                      stream.stream(), (const short*)x.data_ptr(),
                      (short*)out.data_ptr(), nullptr, R, C);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// strided 4-D transpose: x logical [B,H,S,D] (any strides, D-dim
+// contiguous) -> contiguous [B,H,D,S]
+torch::Tensor transpose2d_bhsd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.stride(3) == 1);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  const int B = x.size(0), H = x.size(1), S = x.size(2), D = x.size(3);
+  auto out = torch::empty({B, H, D, S}, x.options());
+  dim3 grid((D + TT - 1) / TT, (S + TT - 1) / TT, B * H);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(transpose_bhsd_kernel, grid, dim3(256), 0,
+                     stream.stream(), (const short*)x.data_ptr(),
+                     (short*)out.data_ptr(), S, D, H,
+                     x.stride(0), x.stride(1), x.stride(2));
   HIP_CHECK_LAST();
   return out;
 }

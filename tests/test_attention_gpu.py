@@ -93,6 +93,45 @@ def test_attn_autograd_path():
         assert (got.float() - want).abs().max().item() / scale < 0.06
 
 
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("shape", [(2, 12, 128, 64), (1, 4, 256, 128),
+                                   (1, 2, 200, 64)])
+def test_attn_qkv_packed(shape, causal):
+    """Packed-QKV (strided) path vs the fp32 reference, fwd + bwd."""
+    from gpuschedule_amd.ops.attention import flash_attention_qkv
+    B, H, S, D = shape
+    torch.manual_seed(5)
+    qkv = torch.randn(B, S, 3, H, D, device="cuda",
+                      dtype=torch.bfloat16, requires_grad=True)
+    o = flash_attention_qkv(qkv, causal=causal)
+    assert o.shape == (B, H, S, D)
+    loss = o.float().square().mean()
+    loss.backward()
+    assert qkv.grad is not None and qkv.grad.shape == qkv.shape
+
+    qkvf = qkv.detach().float().requires_grad_(True)
+    q, k, v = (qkvf[:, :, i].transpose(1, 2) for i in range(3))
+    ref = ref_attn(q, k, v, causal)
+    assert (o.float() - ref).abs().max().item() / \
+        (ref.abs().max().item() + 1e-6) < 0.03
+    ref.square().mean().backward()
+    scale = qkvf.grad.abs().max().item() + 1e-6
+    assert (qkv.grad.float() - qkvf.grad).abs().max().item() / scale \
+        < 0.06
+
+
+def test_transpose_bhsd_strided():
+    from gpuschedule_amd.ops import get_ext
+    ext = get_ext(required=True)
+    torch.manual_seed(6)
+    qkv = torch.randn(2, 96, 3, 4, 64, device="cuda",
+                      dtype=torch.bfloat16)
+    v = qkv[:, :, 2].transpose(1, 2)  # strided [B,H,S,D]
+    vt = ext.transpose2d_bhsd(v)
+    ref = v.contiguous().transpose(2, 3).contiguous()
+    assert torch.equal(vt, ref)
+
+
 def test_bert_job_step_on_gpu():
     from gpuschedule_amd.exec import JobRunner
     r = JobRunner("bertgpu", "bert-base", 8, "cuda:0", torch.bfloat16,
