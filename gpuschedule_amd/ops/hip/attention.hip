@@ -54,12 +54,16 @@ __device__ __forceinline__ float group_sum(float v) {
 // runs with ZERO layout copies.  Contiguous [B,H,S,D] is the special
 // case (sB=H*S*D, sH=S*D, sS=D).  Vt is an internal contiguous
 // scratch either way.
-template <int D, bool CAUSAL, int NW>
+template <int D, bool CAUSAL, int NW, bool STRIDED>
 __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ Vt, short* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int S, float scale,
-    long qsB, long qsH, long qsS, long osB, long osH, long osS) {
+    long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
+  // !STRIDED folds the row strides to the compile-time D (measured:
+  // runtime-stride address math cost 8% on the 8-wave long-seq path)
+  const long qsS = STRIDED ? qsS_ : (long)D;
+  const long osS = STRIDED ? osS_ : (long)D;
   constexpr int NT = NW * 64;      // threads (NW waves x 16 q rows)
   constexpr int QB = NW * 16;      // q rows per block
   constexpr int DF = D / 16;       // PV col fragments
@@ -133,15 +137,27 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
   constexpr int CPRV = KVB * 2 / 16;
   short8 kst[PIECES_K], vst[PIECES_V];
 
+  // per-thread K piece offsets hoisted out of the staging loop: the
+  // per-tile address is k_off[p] + kv0*qsS (ONE uniform multiply per
+  // tile) — a per-piece 64-bit kvr*qsS in the issue path cost ~8% on
+  // the 8-wave long-seq kernel when qsS became a runtime stride
+  long k_off[PIECES_K];
+#pragma unroll
+  for (int p = 0; p < PIECES_K; ++p) {
+    int idx = p * NT + tid;
+    k_off[p] = qb + (long)(idx / CPR) * qsS + (idx % CPR) * 8;
+  }
+
   auto stage_load = [&](int kv0) {
+    const long kv_add = (long)kv0 * qsS;
 #pragma unroll
     for (int p = 0; p < PIECES_K; ++p) {
       int idx = p * NT + tid;
-      int r = idx / CPR, cb = (idx % CPR) * 16;
+      int r = idx / CPR;
       int kvr = kv0 + r;
       if (kvr < S) {
         kst[p] = *reinterpret_cast<const short8*>(
-            K + qb + (long)kvr * qsS + cb / 2);
+            K + k_off[p] + kv_add);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) kst[p][j] = 0;
@@ -326,14 +342,21 @@ __device__ __forceinline__ void glds_tile(
   constexpr int NI = (ROWS / 4) / RPI;     // instructions per wave
   const int rl = lane / CPR;
   const int cb = (lane % CPR) * 16;
+  // hoist the 64-bit row*ld multiplies out of the issue loop (ld is a
+  // runtime stride now): one mul up front, adds of RPI*ld per piece,
+  // and a precomputed uniform clamp offset for the rare tail rows
+  const int rp0 = wid * (ROWS / 4) + rl;
+  const long addr0 = (long)(row0 + rp0) * ld;
+  const long step = (long)RPI * ld;
+  const long clampoff = (long)(rows_max - 1) * ld;
 #pragma unroll
   for (int p = 0; p < NI; ++p) {
-    int rp = wid * (ROWS / 4) + p * RPI + rl;
+    int rp = rp0 + p * RPI;
     int gr = row0 + rp;
-    if (gr >= rows_max) gr = rows_max - 1;
+    long a = (gr >= rows_max) ? clampoff : addr0 + p * step;
     int scol = (cb ^ ((rp & 7) << 4)) / 2 + col0;
     if (scol > cols_max - 8) scol = cols_max - 8;
-    const void* g = src + (long)gr * ld + scol;
+    const void* g = src + a + scol;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)g,
         (__attribute__((address_space(3))) unsigned int*)
@@ -363,7 +386,7 @@ __global__ __launch_bounds__(256) void attn_drow_kernel(
   if (lane == 0) Drow[row] = s;
 }
 
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool STRIDED>
 __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
@@ -371,7 +394,9 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     const float* __restrict__ LSE, const float* __restrict__ DR,
     short* __restrict__ dK, short* __restrict__ dV,
     int B, int H, int S, float scale,
-    long qsB, long qsH, long qsS, long osB, long osH, long osS) {
+    long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
+  const long qsS = STRIDED ? qsS_ : (long)D;
+  const long osS = STRIDED ? osS_ : (long)D;
   constexpr int DF = D / 16;
   constexpr int KC = D / 32;
   constexpr int RB = D * 2;
@@ -529,14 +554,16 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   }
 }
 
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool STRIDED>
 __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
     const short* __restrict__ Kt, const float* __restrict__ LSE,
     const float* __restrict__ DR, short* __restrict__ dQ,
     int B, int H, int S, float scale,
-    long qsB, long qsH, long qsS, long osB, long osH, long osS) {
+    long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
+  const long qsS = STRIDED ? qsS_ : (long)D;
+  const long osS = STRIDED ? osS_ : (long)D;
   constexpr int DF = D / 16;
   constexpr int KC = D / 32;
   constexpr int RB = D * 2;
@@ -737,7 +764,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   int grid = B * H * ((S + QB - 1) / QB);
   auto stream = c10::cuda::getCurrentCUDAStream();
 #define LAUNCH(DD, CC, NW)                                               \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, NW>), dim3(grid),          \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, NW, SR>), dim3(grid),          \
                      dim3(NW * 64), 0, stream.stream(),                  \
                      (const short*)q.data_ptr(),                         \
                      (const short*)k.data_ptr(),                         \
@@ -745,14 +772,20 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                      (short*)o.data_ptr(), lse.data_ptr<float>(),        \
                      B, H, S, scale, q.stride(0), q.stride(1),           \
                      q.stride(2), o.stride(0), o.stride(1), o.stride(2))
-#define LAUNCH_D(DD)                                                     \
+#define LAUNCH_D(DD, SR)                                                 \
   do {                                                                   \
     if (causal) { if (wide) LAUNCH(DD, true, 8);                         \
                   else LAUNCH(DD, true, 4); }                            \
     else { if (wide) LAUNCH(DD, false, 8);                               \
            else LAUNCH(DD, false, 4); }                                  \
   } while (0)
-  if (D == 64) LAUNCH_D(64); else LAUNCH_D(128);
+  if (cont) {
+    if (D == 64) { constexpr bool SR = false; LAUNCH_D(64, SR); }
+    else { constexpr bool SR = false; LAUNCH_D(128, SR); }
+  } else {
+    if (D == 64) { constexpr bool SR = true; LAUNCH_D(64, SR); }
+    else { constexpr bool SR = true; LAUNCH_D(128, SR); }
+  }
 #undef LAUNCH_D
 #undef LAUNCH
   HIP_CHECK_LAST();
@@ -809,7 +842,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   int grid_q = B * H * ((S + 63) / 64);
 #define LAUNCH_BWD(DD, CC)                                              \
   do {                                                                  \
-    hipLaunchKernelGGL((attn_bwd_kv_kernel<DD, CC>), dim3(grid_kv),     \
+    hipLaunchKernelGGL((attn_bwd_kv_kernel<DD, CC, SR>), dim3(grid_kv), \
                        dim3(NTHREADS), 0, stream.stream(),              \
                        (const short*)q.data_ptr(),                      \
                        (const short*)k.data_ptr(),                      \
@@ -822,7 +855,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        B, H, S, scale, q.stride(0), q.stride(1),        \
                        q.stride(2), o.stride(0), o.stride(1),           \
                        o.stride(2));                                    \
-    hipLaunchKernelGGL((attn_bwd_q_kernel<DD, CC>), dim3(grid_q),       \
+    hipLaunchKernelGGL((attn_bwd_q_kernel<DD, CC, SR>), dim3(grid_q),   \
                        dim3(NTHREADS), 0, stream.stream(),              \
                        (const short*)q.data_ptr(),                      \
                        (const short*)k.data_ptr(),                      \
@@ -834,10 +867,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        q.stride(0), q.stride(1), q.stride(2),           \
                        o.stride(0), o.stride(1), o.stride(2));          \
   } while (0)
-  if (D == 64) {
-    if (causal) LAUNCH_BWD(64, true); else LAUNCH_BWD(64, false);
+  if (cont) {
+    constexpr bool SR = false;
+    if (D == 64) {
+      if (causal) LAUNCH_BWD(64, true); else LAUNCH_BWD(64, false);
+    } else {
+      if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false);
+    }
   } else {
-    if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false);
+    constexpr bool SR = true;
+    if (D == 64) {
+      if (causal) LAUNCH_BWD(64, true); else LAUNCH_BWD(64, false);
+    } else {
+      if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false);
+    }
   }
 #undef LAUNCH_BWD
   HIP_CHECK_LAST();
