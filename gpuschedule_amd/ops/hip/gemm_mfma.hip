@@ -475,6 +475,15 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_kernel(
 // ---- bf16 2-D transpose (LDS-tiled, vectorized loads) --------------
 // grid.z = batch index (stride R*C per matrix) for batched use.
 constexpr int TT = 64;  // tile
+// LDS tile index with XOR bank swizzle: the former [TT][TT+8] pad was
+// useless for this kernel's stride-8-row column reads (8 rows x 72
+// shorts = 288 dwords = 0 mod 32 banks -> 8-way conflicts, 22% of
+// GPUTime per rocprofv3 --pmc LDSBankConflict); xoring the column by
+// (row&7)*8 keeps 16B store blocks intact and spreads the 8-row read
+// column across 8 distinct banks.
+__device__ __forceinline__ int tsw(int r, int c) {
+  return r * TT + (c ^ ((r & 7) << 3));
+}
 // colsum (optional): fp32 per-input-column sums accumulated while the
 // tile is already in LDS — gives nn.Linear's bias grad (dy.sum(0)) for
 // free instead of a separate full-tensor torch reduce pass.
@@ -484,7 +493,7 @@ __global__ __launch_bounds__(256) void transpose_kernel(
     float* __restrict__ colsum, int R, int Cc) {
   const short* in = in0 + (long)blockIdx.z * R * Cc;
   short* out = out0 + (long)blockIdx.z * R * Cc;
-  __shared__ short tile[TT][TT + 8];
+  __shared__ short tile[TT * TT];
   int tr0 = blockIdx.y * TT, tc0 = blockIdx.x * TT;
   int tid = threadIdx.x;
   // load: 256 threads x 8 cols = 2048 elems/pass; 2 passes for 64x64
@@ -496,12 +505,12 @@ __global__ __launch_bounds__(256) void transpose_kernel(
     if (gr < R) {
       if (gc + 7 < Cc) {
         short8 v = *reinterpret_cast<const short8*>(in + (long)gr * Cc + gc);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) tile[r][c + j] = v[j];
+        *reinterpret_cast<short8*>(tile + tsw(r, c)) = v;
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          tile[r][c + j] = (gc + j < Cc) ? in[(long)gr * Cc + gc + j] : 0;
+          tile[tsw(r, c + j)] = (gc + j < Cc)
+              ? in[(long)gr * Cc + gc + j] : 0;
       }
     }
   }
@@ -514,7 +523,7 @@ __global__ __launch_bounds__(256) void transpose_kernel(
     if (tid < TT && tc0 + tid < Cc) {
       float s = 0.f;
       int rmax = min(TT, R - tr0);
-      for (int r = 0; r < rmax; ++r) s += bf16_to_f32(tile[r][tid]);
+      for (int r = 0; r < rmax; ++r) s += bf16_to_f32(tile[tsw(r, tid)]);
       colsum[(long)blockIdx.y * Cc + tc0 + tid] = s;
     }
   }
@@ -528,12 +537,13 @@ __global__ __launch_bounds__(256) void transpose_kernel(
       if (goc + 7 < R) {
         short8 v;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) v[j] = tile[r + j][c];
+        for (int j = 0; j < 8; ++j) v[j] = tile[tsw(r + j, c)];
         *reinterpret_cast<short8*>(out + (long)gor * R + goc) = v;
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          if (goc + j < R) out[(long)gor * R + goc + j] = tile[r + j][c];
+          if (goc + j < R)
+            out[(long)gor * R + goc + j] = tile[tsw(r + j, c)];
       }
     }
   }
@@ -549,7 +559,7 @@ __global__ __launch_bounds__(256) void transpose_bhsd_kernel(
   const long z = blockIdx.z;
   const short* in = in0 + (z / H) * sB + (z % H) * sH;
   short* out = out0 + z * (long)R * Cc;
-  __shared__ short tile[TT][TT + 8];
+  __shared__ short tile[TT * TT];
   int tr0 = blockIdx.y * TT, tc0 = blockIdx.x * TT;
   int tid = threadIdx.x;
 #pragma unroll
@@ -561,12 +571,11 @@ __global__ __launch_bounds__(256) void transpose_bhsd_kernel(
       if (gc + 7 < Cc) {
         short8 v = *reinterpret_cast<const short8*>(
             in + (long)gr * sR + gc);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) tile[r][c + j] = v[j];
+        *reinterpret_cast<short8*>(tile + tsw(r, c)) = v;
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          tile[r][c + j] = (gc + j < Cc)
+          tile[tsw(r, c + j)] = (gc + j < Cc)
               ? in[(long)gr * sR + gc + j] : 0;
       }
     }
@@ -581,12 +590,13 @@ __global__ __launch_bounds__(256) void transpose_bhsd_kernel(
       if (goc + 7 < R) {
         short8 v;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) v[j] = tile[r + j][c];
+        for (int j = 0; j < 8; ++j) v[j] = tile[tsw(r + j, c)];
         *reinterpret_cast<short8*>(out + (long)gor * R + goc) = v;
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          if (goc + j < R) out[(long)gor * R + goc + j] = tile[r + j][c];
+          if (goc + j < R)
+            out[(long)gor * R + goc + j] = tile[tsw(r + j, c)];
       }
     }
   }
