@@ -88,6 +88,23 @@ def cmd_gen_trace(args):
     print(f"wrote {len(specs)} jobs -> {args.output}")
 
 
+def cmd_compare(args):
+    """Replay one trace through several policies in the simulator and
+    print a comparison table (avg/p95 JCT, makespan, slowdown,
+    preemptions)."""
+    cfg = _load_config(args.config) if args.config else {}
+    specs = _specs_from_args(args, cfg)
+    n_gpus = cfg.get("gpus", args.gpus)
+    names = [p.strip() for p in args.policies.split(",")]
+    cols = ("avg_jct", "p95_jct", "makespan", "avg_slowdown",
+            "p95_slowdown", "avg_queue_delay", "preemptions")
+    print(f"{'policy':10s} " + " ".join(f"{c:>14s}" for c in cols))
+    for name in names:
+        _, s = simulate(specs, make_policy(name), n_gpus=n_gpus)
+        print(f"{name:10s} " +
+              " ".join(f"{s[c]:14.2f}" for c in cols))
+
+
 def cmd_export_trace(args):
     from .sched.events import EventLog
     log = EventLog.load(args.log)
@@ -122,6 +139,11 @@ def main(argv=None):
     p.add_argument("--interarrival", type=float, default=10.0)
     p.add_argument("-o", "--output", required=True)
     p.set_defaults(fn=cmd_gen_trace)
+    p = sub.add_parser("compare", parents=[common],
+                       help="replay one trace under several policies "
+                            "(simulator) and print a metric table")
+    p.add_argument("--policies", default="fifo,srtf,las,gandiva,elastic")
+    p.set_defaults(fn=cmd_compare)
     p = sub.add_parser(
         "export-trace",
         help="convert an event-log JSONL to a Chrome/Perfetto trace")

@@ -143,7 +143,19 @@ class EventLog:
             makespan = (max(e["t"] for e in completes.values())
                         - min(submits.values()))
         qdelays = [starts[j] - submits[j] for j in starts if j in submits]
+        # finish-time slowdown = JCT / contention-free solo runtime
+        # (Tiresias/Themis-style fairness metric); ideal_s comes from
+        # the submit events
+        ideals = {e["job"]: e["ideal_s"] for e in self.events
+                  if e["event"] == "submit" and e.get("ideal_s")}
+        slowdowns = sorted(e["jct"] / ideals[j]
+                           for j, e in completes.items() if j in ideals)
+        ns = len(slowdowns)
         return {
+            "avg_slowdown": sum(slowdowns) / ns if ns else 0.0,
+            "p95_slowdown": (slowdowns[min(ns - 1, int(0.95 * ns))]
+                             if ns else 0.0),
+            "max_slowdown": slowdowns[-1] if ns else 0.0,
             "n_jobs": float(len(submits)),
             "n_complete": float(n),
             "avg_jct": sum(jcts) / n if n else 0.0,

@@ -71,9 +71,14 @@ class Scheduler:
     def _admit(self, spec: JobSpec, now: float) -> None:
         job = Job(spec=spec, submit_t=max(now, spec.submit_time))
         self.jobs[spec.job_id] = job
+        # ideal_s: the job's contention-free solo runtime (for the
+        # finish-time-slowdown metric in EventLog.summary)
+        ideal = spec.total_iters * catalog.sim_iter_time(spec,
+                                                         spec.n_gpus)
         self.log.emit(job.submit_t, "submit", spec.job_id,
                       n_gpus=spec.n_gpus, model=spec.model,
-                      iters=spec.total_iters)
+                      iters=spec.total_iters,
+                      ideal_s=round(ideal, 6))
         # Reject unsatisfiable jobs at admission (they would otherwise
         # starve in the queue forever).
         width = spec.min_gpus if spec.elastic else spec.n_gpus

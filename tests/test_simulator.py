@@ -56,6 +56,17 @@ def test_event_log_jsonl_roundtrip(tmp_path):
     assert loaded.summary() == s
 
 
+def test_slowdown_metrics():
+    # solo job on an idle cluster: slowdown ~1; a queued job behind it
+    # on 1 GPU has slowdown > 1
+    specs = [J("a", t=0.0, iters=20, hint=1.0),
+             J("b", t=0.0, iters=20, hint=1.0)]
+    log, s = simulate(specs, make_policy("fifo"), n_gpus=1)
+    assert 0.9 < s["avg_slowdown"] and s["max_slowdown"] >= 1.8
+    _, s1 = simulate(specs[:1], make_policy("fifo"), n_gpus=1)
+    assert 0.9 < s1["avg_slowdown"] < 1.2
+
+
 def test_chrome_trace_export(tmp_path):
     specs = make_trace(6, seed=3, mean_interarrival=2.0,
                        iters_range=(5, 30), gang_sizes=[1, 1, 2])
