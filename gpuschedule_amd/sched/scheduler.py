@@ -72,9 +72,13 @@ class Scheduler:
         job = Job(spec=spec, submit_t=max(now, spec.submit_time))
         self.jobs[spec.job_id] = job
         # ideal_s: the job's contention-free solo runtime (for the
-        # finish-time-slowdown metric in EventLog.summary)
-        ideal = spec.total_iters * catalog.sim_iter_time(spec,
-                                                         spec.n_gpus)
+        # finish-time-slowdown metric in EventLog.summary); unknown
+        # models are rejected just below — log their submit without it
+        try:
+            ideal = spec.total_iters * catalog.sim_iter_time(
+                spec, spec.n_gpus)
+        except ValueError:
+            ideal = 0.0
         self.log.emit(job.submit_t, "submit", spec.job_id,
                       n_gpus=spec.n_gpus, model=spec.model,
                       iters=spec.total_iters,
