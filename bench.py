@@ -13,10 +13,11 @@ HIP kernel path), then the policy re-evaluates.  The timed region is
 EXACTLY K ticks, bracketed by barrier + torch.cuda.synchronize on both
 sides.  The trace's job budgets are sized in *tick-slot* space so every
 trace job completes inside the window independent of wall-clock
-iteration times (sum over jobs of running-ticks <= 70% of K*N slots);
-per-GPU best-effort fill jobs (excluded from JCT/makespan, real
-training work) keep every GPU busy for all K ticks, so no timed tick
-does less than N GPUs of real work.
+iteration times (sum over jobs of running-ticks <= 70% of K slots —
+ONE GPU's window, so the identical trace replays at every N and the
+strong-scaling metric is honest); per-GPU best-effort fill jobs
+(excluded from JCT/makespan, real training work) keep every GPU busy
+for all K ticks, so no timed tick does less than N GPUs of real work.
 
 value = avg JCT + makespan over the trace jobs, in wall seconds
 (lower is better).  Strong scaling: the SAME trace (single-GPU jobs,
