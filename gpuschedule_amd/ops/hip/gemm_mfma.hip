@@ -472,6 +472,194 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_kernel(
   }
 }
 
+// ==== 256-tile, consumption-scheduled staging (counted vmcnt) ======
+// A/B RESULT (MI355X, GSAMD_GEMM_CS=1 vs default): wash to -3% at
+// 4k/8k square, ~+2% llama-ffn — i.e. the K-tile-boundary vmcnt(0)
+// drain this schedule eliminates was NOT a real stall (pieces are >=1
+// phase ≈500+ cycles old and HBM latency is already covered).  Kept
+// as a gated variant: the counted-wait machinery below is the
+// starting point if round 2 raises prefetch distance (3 buffers /
+// smaller BK), where counted waits become load-bearing.
+//
+// The plain 256-tile kernel drains ALL in-flight DMA at each K-tile
+// boundary (__syncthreads => s_waitcnt vmcnt(0)), stalling on pieces
+// staged as recently as one phase earlier.  Here every wave stages
+// EXACTLY the rows it (or its column-sharing partner) consumes, in
+// its own consumption order — one A piece + one B piece per phase:
+//
+//   A slot p: rows  half*128 + (p>>1)*64 + (p&1)*32 + (wid&3)*8
+//             (half = wid>>2, the wave's OWN A half; phase p consumes
+//              quadrant jr=p>>1 = slots {2jr, 2jr+1})
+//   B slot p: rows  (wid&3)*64 + (p>>1)*32 + (p&1)*16 + (wid>>2)*8
+//             (the wave's OWN B columns, shared with its wid^4
+//              partner; phase p consumes jc=p&1 = slots {2jc, 2jc+1})
+//
+// With that per-wave FIFO (A0,B0,A1,B1,A2,B2,A3,B3 per tile) the
+// phase waits become counted constants instead of full drains:
+//   ph0 needs A0,A1,B0,B1 (oldest 4 of the previous tile's 8)
+//       -> s_waitcnt vmcnt(4)
+//   ph1 needs B2,B3 (forces the rest of the previous tile; the
+//       youngest piece is ~2 phase-gaps old by then)
+//       -> s_waitcnt vmcnt(2)
+//   ph2/ph3 need nothing new -> vmcnt(4)/vmcnt(6) are no-ops.
+// Issue happens AFTER the phase barrier, so tile t+2's DMA (into the
+// buffer tile t read) cannot start until every wave has left tile t's
+// MFMAs.  No __syncthreads inside the K loop at all.
+__device__ __forceinline__ void glds_cs_piece(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int k0, char* lds_img, int rp_base, int lane) {
+  const int rl = lane >> 3;
+  const int gcolb = ((lane & 7) * 16) ^ (rl << 4);
+  int gr = row0 + rp_base + rl;
+  if (gr >= rows_max) gr = rows_max - 1;
+  const void* g = src + (long)gr * ld + k0 + gcolb / 2;
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)
+          (lds_img + rp_base * ROWB),
+      16, 0, 0);
+}
+
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_cs_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+  char* Ac = lds;
+  char* Bc = lds + BM2 * ROWB;
+  char* An = lds + 2 * BM2 * ROWB;
+  char* Bn = lds + 3 * BM2 * ROWB;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  // per-wave staged-piece row bases (slot p)
+  const int a_rp0 = (wid >> 2) * 128 + (wid & 3) * 8;  // + p-dependent
+  const int b_rp0 = (wid & 3) * 64 + (wid >> 2) * 8;
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // prologue: tile 0 staged in slot order, full drain once
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    glds_cs_piece(A, K, bm0, M, 0, Ac,
+                  a_rp0 + (p >> 1) * 64 + (p & 1) * 32, lane);
+    glds_cs_piece(B, K, bn0, N, 0, Bc,
+                  b_rp0 + (p >> 1) * 32 + (p & 1) * 16, lane);
+  }
+  __syncthreads();
+
+  const int nk = K / BK2;
+  const int arow = (lane & 15);
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK2;
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int jr = ph >> 1, jc = ph & 1;
+      // counted wait for THIS phase's quadrant, then the phase barrier
+      // simm16: vmcnt[3:0] | expcnt(7)<<4 | lgkmcnt(15)<<8 (no
+      // exp/lgkm wait); vmcnt-only counted waits
+      if (ph == 0) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+      else if (ph == 1) __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+      // ph2/ph3: everything needed already forced at ph1
+      __builtin_amdgcn_s_barrier();
+      if (pre) {
+        glds_cs_piece(A, K, bm0, M, k1, An,
+                      a_rp0 + jr * 64 + jc * 32, lane);
+        glds_cs_piece(B, K, bn0, N, k1, Bn,
+                      b_rp0 + jr * 32 + jc * 16, lane);
+      }
+      bf16x8 af[4][2], bf[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int fi = 0; fi < 4; ++fi)
+          af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+              Ac + swz(wr + jr * 64 + fi * 16 + arow, kk * 64 + kb));
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          bf[fj][kk] = *reinterpret_cast<const bf16x8*>(
+              Bc + swz(wc + jc * 32 + fj * 16 + arow, kk * 64 + kb));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+          for (int fj = 0; fj < 2; ++fj)
+            acc[jr * 4 + fi][jc * 2 + fj] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[fi][kk], bf[fj][kk],
+                    acc[jr * 4 + fi][jc * 2 + fj], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (pre) {
+      char* tmp = Ac; Ac = An; An = tmp;
+      tmp = Bc; Bc = Bn; Bn = tmp;
+    }
+  }
+  __builtin_amdgcn_s_waitcnt(0xf70);  // drain DMA before LDS reuse
+  __builtin_amdgcn_s_barrier();
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __syncthreads();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
 // ---- bf16 2-D transpose (LDS-tiled, vectorized loads) --------------
 // grid.z = batch index (stride R*C per matrix) for batched use.
 constexpr int TT = 64;  // tile
@@ -649,10 +837,23 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
   long blocks256 = (long)((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
   if (M >= 256 && N >= 256 && blocks256 >= 200) {
     int grid = (int)blocks256;
-    hipLaunchKernelGGL(gemm_nt256_kernel, dim3(grid), dim3(NT2), 0,
-                       stream.stream(), (const short*)a.data_ptr(),
-                       (const short*)b.data_ptr(), (short*)c.data_ptr(),
-                       bias_p, M, N, K, relu ? 1 : 0);
+    // consumption-scheduled staging variant (counted vmcnt waits, no
+    // boundary drain) — A/B gate while it is being qualified
+    static const bool use_cs = [] {
+      const char* e = getenv("GSAMD_GEMM_CS");
+      return e && e[0] == '1';
+    }();
+    if (use_cs) {
+      hipLaunchKernelGGL(gemm_nt256_cs_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+    } else {
+      hipLaunchKernelGGL(gemm_nt256_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+    }
   } else {
     // A/B result: the 32x32 MFMA variant ties at 4k/8k and loses on
     // small grids (the structure is stage/barrier-bound, not
