@@ -130,82 +130,63 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
       ? min((q_tile_end / KVB) + 1, (S + KVB - 1) / KVB)
       : (S + KVB - 1) / KVB;
 
-  // ---- staging helpers (register path: T14 async-split) ---------
+  // ---- staging: direct HBM->LDS DMA (global_load_lds, 16 B/lane) --
+  // Replaces the earlier register-staged T14 path: same async timing
+  // (issue early, completion drained by the end-of-tile barrier) but
+  // no kst/vst registers and no ds_write pass.  The XOR bank swizzle
+  // moves to the SOURCE column (lane-linear LDS dest), exactly as in
+  // the GEMM staging.  Tail rows/cols are CLAMPED to the last valid
+  // position instead of zeroed: clamped K rows produce finite garbage
+  // scores on columns the softmax kills with -inf, and clamped V
+  // columns are multiplied by those zero probabilities.
   constexpr int PIECES_K = KVB * D * 2 / (NT * 16);
   constexpr int CPR = D * 2 / 16;     // 16B chunks per K row
   constexpr int PIECES_V = D * KVB * 2 / (NT * 16);
   constexpr int CPRV = KVB * 2 / 16;
-  short8 kst[PIECES_K], vst[PIECES_V];
 
-  // per-thread K piece offsets hoisted out of the staging loop: the
-  // per-tile address is k_off[p] + kv0*qsS (ONE uniform multiply per
-  // tile) — a per-piece 64-bit kvr*qsS in the issue path cost ~8% on
-  // the 8-wave long-seq kernel when qsS became a runtime stride
-  long k_off[PIECES_K];
-#pragma unroll
-  for (int p = 0; p < PIECES_K; ++p) {
-    int idx = p * NT + tid;
-    k_off[p] = qb + (long)(idx / CPR) * qsS + (idx % CPR) * 8;
-  }
-
-  auto stage_load = [&](int kv0) {
+  auto stage_dma = [&](int kv0, char* kbuf, char* vbuf) {
     const long kv_add = (long)kv0 * qsS;
 #pragma unroll
     for (int p = 0; p < PIECES_K; ++p) {
       int idx = p * NT + tid;
       int r = idx / CPR;
+      int cbs = ((idx % CPR) * 16) ^ ((r & 7) << 4);  // src-swizzled
       int kvr = kv0 + r;
-      if (kvr < S) {
-        kst[p] = *reinterpret_cast<const short8*>(
-            K + k_off[p] + kv_add);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) kst[p][j] = 0;
-      }
+      long roff = (kvr < S) ? kv_add + (long)r * qsS
+                            : (long)(S - 1) * qsS;
+      const void* g = K + qb + roff + cbs / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)
+              (kbuf + (p * NT + (tid & ~63)) * 16),
+          16, 0, 0);
     }
 #pragma unroll
     for (int p = 0; p < PIECES_V; ++p) {
       int idx = p * NT + tid;
-      int r = idx / CPRV, cb = (idx % CPRV) * 16;
-      int kvc = kv0 + cb / 2;
-      if (kvc + 7 < S) {
-        vst[p] = *reinterpret_cast<const short8*>(
-            Vt + voff + (long)r * S + kvc);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vst[p][j] = (kvc + j < S)
-              ? (Vt + voff + (long)r * S)[kvc + j] : (short)0;
-      }
-    }
-  };
-  auto stage_write = [&](char* kbuf, char* vbuf) {
-#pragma unroll
-    for (int p = 0; p < PIECES_K; ++p) {
-      int idx = p * NT + tid;
-      int r = idx / CPR, cb = (idx % CPR) * 16;
-      *reinterpret_cast<short8*>(kbuf + swz(r, cb, ROWB)) = kst[p];
-    }
-#pragma unroll
-    for (int p = 0; p < PIECES_V; ++p) {
-      int idx = p * NT + tid;
-      int r = idx / CPRV, cb = (idx % CPRV) * 16;
-      *reinterpret_cast<short8*>(vbuf + swz(r, cb, KVB * 2)) = vst[p];
+      int r = idx / CPRV;
+      int cbs = ((idx % CPRV) * 16) ^ ((r & 7) << 4);
+      int kvc = kv0 + cbs / 2;
+      if (kvc + 7 >= S) kvc = S - 8;
+      const void* g = Vt + voff + (long)r * S + kvc;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)
+              (vbuf + (p * NT + (tid & ~63)) * 16),
+          16, 0, 0);
     }
   };
 
-  stage_load(0);
-  stage_write(k_cur, v_cur);
+  stage_dma(0, k_cur, v_cur);
   __syncthreads();
 
   for (int t = 0; t < n_kv; ++t) {
     const int kv0 = t * KVB;
     char* k_lds = k_cur;
     char* v_lds = v_cur;
-    // T14 async-split: issue next tile's global loads NOW; their
-    // HBM latency hides under QK^T + softmax + PV; the ds_write's
-    // implicit wait lands just before the end-of-tile barrier.
-    if (t + 1 < n_kv) stage_load((t + 1) * KVB);
+    // issue next tile's HBM->LDS DMA NOW; its latency hides under
+    // QK^T + softmax + PV; the end-of-tile __syncthreads drains it
+    if (t + 1 < n_kv) stage_dma((t + 1) * KVB, k_nxt, v_nxt);
 
     // ---- S = QK^T: 4 col-frags x KC k-steps ---------------------
     float4v s_acc[4];
@@ -290,7 +271,6 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
       }
     }
     if (t + 1 < n_kv) {
-      stage_write(k_nxt, v_nxt);
       char* tmp = k_cur; k_cur = k_nxt; k_nxt = tmp;
       tmp = v_cur; v_cur = v_nxt; v_nxt = tmp;
     }
