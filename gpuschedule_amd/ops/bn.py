@@ -26,10 +26,14 @@ class _BNFn(torch.autograd.Function):
         else:
             y, _, _ = ext.bn_fwd(x2d, weight, bias, mean, rstd, None,
                                  None, res2d, momentum, eps, relu)
-        ctx.save_for_backward(x2d, y if relu else None, mean, rstd,
-                              weight)
+        # relu WITHOUT residual: backward recomputes the mask from
+        # x*scale+shift (needs beta) instead of re-reading y — one
+        # fewer activation stream in both backward passes
+        has_res = res2d is not None
+        ctx.save_for_backward(x2d, y if (relu and has_res) else None,
+                              mean, rstd, weight, bias)
         ctx.relu = relu
-        ctx.has_res = res2d is not None
+        ctx.has_res = has_res
         # y ONLY: returning mean/rstd made autograd materialize two
         # zero-grad tensors per BN backward (106 fp32 fills/iter on
         # ResNet-50)
@@ -38,9 +42,10 @@ class _BNFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = get_ext(required=True)
-        x2d, y, mean, rstd, weight = ctx.saved_tensors
+        x2d, y, mean, rstd, weight, bias = ctx.saved_tensors
         out = ext.bn_bwd(dy.contiguous(), x2d, y, mean,
-                         rstd, weight, ctx.relu, ctx.has_res)
+                         rstd, weight, ctx.relu, ctx.has_res,
+                         beta=None if y is not None else bias)
         dx, dbeta, dgamma = out[0], out[1], out[2]
         dres = out[3] if ctx.has_res else None
         return (dx, dres, dgamma, dbeta, None, None, None, None,

@@ -57,7 +57,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   torch::Tensor gamma, bool relu,
-                                  bool want_dres);
+                                  bool want_dres,
+                                  c10::optional<torch::Tensor> beta);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "fused LayerNorm/RMSNorm forward (bf16)");
@@ -86,5 +87,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_multi", &adamw_multi, "multi-tensor fused AdamW (one launch, zeroes grads)");
   m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");
   m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");
-  m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+ReLU) backward, channels_last");
+  m.def("bn_bwd", &bn_bwd,
+        "fused BatchNorm(+ReLU) backward, channels_last; pass beta "
+        "instead of y to recompute the relu mask from x",
+        py::arg("dy"), py::arg("x"), py::arg("y"), py::arg("mean"),
+        py::arg("rstd"), py::arg("gamma"), py::arg("relu"),
+        py::arg("want_dres"),
+        py::arg("beta") = c10::optional<torch::Tensor>());
 }

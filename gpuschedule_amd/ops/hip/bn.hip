@@ -26,11 +26,15 @@ constexpr int PASS = NT * 8;  // elements per block pass (2048)
 // ---- per-channel sum/sumsq (or sum_dy/sum_dy_xhat) ----------------
 // REGPATH requires PASS % C == 0 (all ResNet widths); BWD computes
 // backward sums instead; RELU masks dy by (y > 0).
+// RELU backward mask: read from y when given, else RECOMPUTED as
+// (x*scale + shift) > 0 (valid only without a fused residual) — drops
+// a full activation stream from the backward stats/apply reads.
 template <bool BWD, bool RELU>
 __global__ __launch_bounds__(NT) void bn_stats_kernel(
     const short* __restrict__ x, const short* __restrict__ dy,
     const short* __restrict__ y, const float* __restrict__ mean,
-    const float* __restrict__ rstd, float* __restrict__ out_a,
+    const float* __restrict__ rstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ out_a,
     float* __restrict__ out_b, long E, int C) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* lds_a = (float*)smem;          // [C]
@@ -43,12 +47,19 @@ __global__ __launch_bounds__(NT) void bn_stats_kernel(
   float acc_a[8], acc_b[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) { acc_a[j] = 0.f; acc_b[j] = 0.f; }
-  float mu[8], rs[8];
+  float mu[8], rs[8], msc[8], msh[8];
   if (BWD) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       mu[j] = mean[ch0 + j];
       rs[j] = rstd[ch0 + j];
+    }
+    if (RELU && !y) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        msc[j] = gamma[ch0 + j] * rs[j];
+        msh[j] = beta[ch0 + j] - mu[j] * msc[j];
+      }
     }
   }
 
@@ -66,12 +77,17 @@ __global__ __launch_bounds__(NT) void bn_stats_kernel(
     } else {
       short8 dv = *reinterpret_cast<const short8*>(dy + e);
       short8 yv;
-      if (RELU) yv = *reinterpret_cast<const short8*>(y + e);
+      if (RELU && y) yv = *reinterpret_cast<const short8*>(y + e);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float d = bf16_to_f32(dv[j]);
-        if (RELU && bf16_to_f32(yv[j]) <= 0.f) d = 0.f;
-        float xh = (bf16_to_f32(xv[j]) - mu[j]) * rs[j];
+        float xf = bf16_to_f32(xv[j]);
+        if (RELU) {
+          bool dead = y ? (bf16_to_f32(yv[j]) <= 0.f)
+                        : (xf * msc[j] + msh[j] <= 0.f);
+          if (dead) d = 0.f;
+        }
+        float xh = (xf - mu[j]) * rs[j];
         acc_a[j] += d;
         acc_b[j] += d * xh;
       }
@@ -132,22 +148,27 @@ __global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
     const short* __restrict__ y, short* __restrict__ dx,
     short* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    const float* __restrict__ gamma, const float* __restrict__ sum_dy,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ sum_dy,
     const float* __restrict__ sum_dyxh, long E, int C, float inv_cnt) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // per-channel constants: k1 = g*rstd, c1 = sum_dy/Nc,
-  // c2 = sum_dyxh/Nc, mu, rs
+  // c2 = sum_dyxh/Nc, mu, rs (+ shift when the relu mask is
+  // recomputed from x instead of read from y)
   float* s_k1 = (float*)smem;
   float* s_c1 = s_k1 + C;
   float* s_c2 = s_c1 + C;
   float* s_mu = s_c2 + C;
   float* s_rs = s_mu + C;
+  float* s_sh = s_rs + C;
   for (int i = threadIdx.x; i < C; i += NT) {
-    s_k1[i] = gamma[i] * rstd[i];
+    float k1 = gamma[i] * rstd[i];
+    s_k1[i] = k1;
     s_c1[i] = sum_dy[i] * inv_cnt;
     s_c2[i] = sum_dyxh[i] * inv_cnt;
     s_mu[i] = mean[i];
     s_rs[i] = rstd[i];
+    if (RELU && beta) s_sh[i] = beta[i] - mean[i] * k1;
   }
   __syncthreads();
   const int t8 = threadIdx.x * 8;
@@ -158,13 +179,17 @@ __global__ __launch_bounds__(NT) void bn_bwd_apply_kernel(
     short8 dv = *reinterpret_cast<const short8*>(dy + e);
     short8 xv = *reinterpret_cast<const short8*>(x + e);
     short8 yv;
-    if (RELU) yv = *reinterpret_cast<const short8*>(y + e);
+    if (RELU && y) yv = *reinterpret_cast<const short8*>(y + e);
     short8 out, rout;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = ch0 + j;
       float d = bf16_to_f32(dv[j]);
-      if (RELU && bf16_to_f32(yv[j]) <= 0.f) d = 0.f;
+      if (RELU) {
+        bool dead = y ? (bf16_to_f32(yv[j]) <= 0.f)
+                      : (bf16_to_f32(xv[j]) * s_k1[c] + s_sh[c] <= 0.f);
+        if (dead) d = 0.f;
+      }
       if (RES) rout[j] = f32_to_bf16(d);
       float xh = (bf16_to_f32(xv[j]) - s_mu[c]) * s_rs[c];
       out[j] = f32_to_bf16(s_k1[c] * (d - s_c1[c] - xh * s_c2[c]));
@@ -295,8 +320,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     hipLaunchKernelGGL((bn_stats_kernel<false, false>),
                        dim3(sgrid), dim3(NT), 2 * C * 4,
                        stream.stream(), (const short*)x.data_ptr(),
-                       nullptr, nullptr, nullptr, nullptr,
-                       part.data_ptr<float>(), nullptr, E, C);
+                       nullptr, nullptr, nullptr, nullptr, nullptr,
+                       nullptr, part.data_ptr<float>(), nullptr, E, C);
     long cnt = E / C;
     mean = torch::empty({C}, fopt);
     rstd = torch::empty({C}, fopt);
@@ -331,12 +356,25 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
 }
 
 // returns (dx, sum_dy (=dbeta), sum_dyxh (=dgamma)); y needed iff relu
+// relu backward needs the forward mask: pass y, OR (without a fused
+// residual) pass beta instead and the kernels recompute the mask from
+// x — one fewer full activation stream read in both backward passes.
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   torch::Tensor gamma, bool relu,
-                                  bool want_dres) {
+                                  bool want_dres,
+                                  c10::optional<torch::Tensor> beta) {
   const int C = x.size(-1);
+  const float* beta_p = nullptr;
+  if (beta.has_value() && beta->defined())
+    beta_p = beta->data_ptr<float>();
+  const short* y_mask = nullptr;  // forward activations for the mask
+  if (relu) {
+    TORCH_CHECK(y.has_value() || beta_p,
+                "bn_bwd: relu needs y or beta (mask recompute)");
+    if (y.has_value()) y_mask = (const short*)y->data_ptr();
+  }
   check_bn(x, C);
   const long E = x.numel();
   auto fopt = x.options().dtype(torch::kFloat32);
@@ -346,13 +384,14 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   auto sums = torch::empty({2 * C}, fopt);
   auto sum_dy = sums.narrow(0, 0, C);
   auto sum_dyxh = sums.narrow(0, C, C);
-  const short* y_p = relu ? (const short*)y->data_ptr() : nullptr;
+  const short* y_p = y_mask;
   if (relu) {
     hipLaunchKernelGGL((bn_stats_kernel<true, true>), dim3(sgrid),
                        dim3(NT), 2 * C * 4, stream.stream(),
                        (const short*)x.data_ptr(),
                        (const short*)dy.data_ptr(), y_p,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta_p,
                        part.data_ptr<float>(), nullptr, E, C);
   } else {
     hipLaunchKernelGGL((bn_stats_kernel<true, false>), dim3(sgrid),
@@ -360,6 +399,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        (const short*)x.data_ptr(),
                        (const short*)dy.data_ptr(), nullptr,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       nullptr, nullptr,
                        part.data_ptr<float>(), nullptr, E, C);
   }
   hipLaunchKernelGGL(bn_reduce_slots_kernel,
@@ -376,13 +416,13 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   float inv_cnt = 1.0f / (float)(E / C);
 #define BN_BWD_APPLY(RL, RS, YP)                                         \
   hipLaunchKernelGGL((bn_bwd_apply_kernel<RL, RS>), dim3(grid_for(E)),   \
-                     dim3(NT), 5 * C * 4, stream.stream(),               \
+                     dim3(NT), 6 * C * 4, stream.stream(),               \
                      (const short*)dy.data_ptr(),                        \
                      (const short*)x.data_ptr(), YP,                     \
                      (short*)dx.data_ptr(), dres_p,                      \
                      mean.data_ptr<float>(),                             \
                      rstd.data_ptr<float>(), gamma.data_ptr<float>(),    \
-                     sum_dy.data_ptr<float>(),                           \
+                     beta_p, sum_dy.data_ptr<float>(),                   \
                      sum_dyxh.data_ptr<float>(), E, C, inv_cnt)
   if (relu) {
     if (want_dres) BN_BWD_APPLY(true, true, y_p);

@@ -235,6 +235,14 @@ def test_fused_bn_fwd_bwd(shape, relu):
     assert mx(dgamma, gf.grad) / s < 2e-2
     s = bf.grad.abs().max().item() + 1e-3
     assert mx(dbeta, bf.grad) / s < 2e-2
+    if relu:
+        # mask-recompute path (no y, beta instead) must agree
+        dx2, db2, dg2 = ext().bn_bwd(dy, x, None, mean, rstd, g, relu,
+                                     False, beta=b)
+        assert mx(dx2, dx.float()) / (dx.float().abs().max().item()
+                                      + 1e-4) < 1e-3
+        assert torch.allclose(db2, dbeta, rtol=1e-2, atol=1e-2)
+        assert torch.allclose(dg2, dgamma, rtol=1e-2, atol=1e-2)
 
 
 def test_fused_bn_module_matches_torch_bn():
