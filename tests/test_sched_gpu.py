@@ -55,14 +55,29 @@ def test_checkpoint_restore_gpu(tmp_path):
 
 
 def test_srtf_preempt_resume_gpu(tmp_path):
-    specs = [J("long", iters=10, iter_time_hint=10.0),
-             J("short", t=0.1, iters=2, iter_time_hint=0.001,
-               model="resnet18", batch=32)]
-    log = run_jobs(specs, "srtf", tmp_path)
+    # Deterministic by construction: the short job is injected only
+    # AFTER the long one has measurably progressed (a wall-clock
+    # arrival raced model-build time and occasionally admitted both
+    # in the same tick, where SRTF simply runs short first with no
+    # preemption to observe).
+    sched = Scheduler(Cluster(1, 1), make_policy("srtf"))
+    sched.submit([J("long", iters=40, iter_time_hint=10.0)])
+    be = TorchBackend(sched, rank=0, world=1, device_type="cuda",
+                      iters_per_tick=2, ckpt_root=str(tmp_path / "ck"))
+    for _ in range(50):
+        be.tick_once()
+        if sched.jobs["long"].done_iters >= 2:
+            break
+    assert sched.jobs["long"].done_iters >= 2
+    sched.submit_now(J("short", iters=2, iter_time_hint=0.001,
+                       model="resnet18", batch=32), be.now())
+    log = be.run(max_seconds=600)
     s = log.summary()
     assert int(s["n_complete"]) == 2
     order = [e["job"] for e in log.events if e["event"] == "complete"]
     assert order[0] == "short"
+    assert any(e["event"] == "preempt" and e["job"] == "long"
+               for e in log.events)
 
 
 def test_eviction_restore_cycle_gpu(tmp_path):
