@@ -59,6 +59,29 @@ def test_bench_cpu_contract(tmp_path):
     assert res["higher_is_better"] is False
 
 
+def test_compare_command_runs():
+    out = subprocess.run(
+        [sys.executable, "-m", "gpuschedule_amd", "compare",
+         "--jobs", "8", "--gpus", "2", "--policies", "fifo,srtf"],
+        cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    lines = [l for l in out.stdout.splitlines() if l.strip()]
+    assert lines[0].startswith("policy")
+    assert {l.split()[0] for l in lines[1:]} == {"fifo", "srtf"}
+
+
+def test_bench_trace_budget_fits_one_gpu():
+    """Trace demand must fit ~70% of ONE GPU's tick slots so the same
+    trace completes at every N (strong scaling) within K ticks."""
+    sys.path.insert(0, REPO)
+    from bench import build_trace
+    steps, ipt = 40, 2
+    arrivals, _ = build_trace(8, steps, 1, ipt)
+    total = sum(s.total_iters for _, s in arrivals)
+    assert total <= 0.75 * steps * ipt + 8 * ipt  # budget + rounding
+    assert all(t <= steps // 4 for t, _ in arrivals)
+
+
 def test_bench_trace_identical_across_n():
     """Strong scaling: the trace-job set must not depend on N (only
     the best-effort fill jobs do)."""
