@@ -1,10 +1,13 @@
-"""Attention dispatch: hand-written flash-forward kernel on GPU.
+"""Attention dispatch: hand-written flash kernels on GPU.
 
-``flash_attention(q, k, v, causal)`` takes [B, H, S, D] bf16 tensors.
-GPU forward runs the gfx950 flash kernel (ops/hip/attention.hip,
-online-softmax MFMA); backward recomputes through the exact composite
-(matmul-softmax-matmul) on the MFMA GEMM path via torch matmuls.
-CPU path: fp32 reference composite (tests).
+``flash_attention(q, k, v, causal)`` takes [B, H, S, D] bf16 tensors
+(k/v may have fewer heads: GQA).  ``flash_attention_qkv(qkv, causal)``
+takes the packed [B, S, 3, H, D] view of a fused self-attention
+projection and runs the zero-copy strided path.  GPU forward runs the
+gfx950 flash kernel (ops/hip/attention.hip, online-softmax MFMA);
+backward runs the FA2-style two-kernel HIP backward (dK/dV then dQ —
+no S x S activation is ever materialized).  CPU path: fp32 reference
+composite (tests).
 """
 from __future__ import annotations
 
