@@ -4,8 +4,12 @@
 // verified by the GEMM kernel):
 //  * Q stays in registers: each wave owns 16 query rows; A-fragments
 //    read straight from global (row = lane&15, 8 bf16 along D).
-//  * K tile [KVB=64][D] staged in LDS with the GEMM XOR swizzle; QK^T
-//    B-fragments are the same addressing as the GEMM B operand.
+//  * K tile [KVB=64][D] staged into LDS by direct HBM->LDS DMA
+//    (global_load_lds, source-swizzled columns) with the GEMM XOR
+//    swizzle; QK^T B-fragments are the same addressing as the GEMM B
+//    operand.  Double-buffered: the next tile's DMA issues before the
+//    current tile's compute (async-split), drained at the tile
+//    barrier.
 //  * V is PRE-TRANSPOSED once per call (fast LDS-tiled transpose2d
 //    kernel) to [B,H,D,S], so the PV B-operand also reads 8 bf16
 //    contiguous along the reduction (kv) dim — no ds transpose-read
@@ -16,9 +20,13 @@
 //  * Online softmax: per-lane running m/l for 4 rows; row reductions
 //    are 4 in-register values + a 16-lane xor-shuffle tree.
 //
-// One workgroup = 4 independent waves = 64 query rows of one (b, h).
+// One workgroup = 4 (or 8, long-seq gate) independent waves = 64/128
+// query rows of one (b, h).
 // Causal masking: whole KV tiles after the diagonal are skipped;
 // diagonal tiles mask per element.  Sequence tails are masked to -inf.
+// Q/K and O go through (batch, head, row) stride triples so packed-QKV
+// projection slices run with zero layout copies; the contiguous case
+// constant-folds the strides via the STRIDED template split.
 #include <torch/extension.h>
 #include <c10/cuda/CUDAStream.h>
 #include "common.h"
