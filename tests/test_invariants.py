@@ -82,6 +82,37 @@ def test_schedule_invariants(policy, seed):
     assert done == set(iters)
 
 
+def test_gangs_never_straddle_nodes():
+    """16-GPU, 2-node cluster: every gang start in the actual schedule
+    stays within one 8-GPU node (xGMI is intra-node)."""
+    from gpuschedule_amd import Cluster, Scheduler
+    from gpuschedule_amd.sim.simulator import Simulator
+    specs = make_trace(24, mix="philly-slice", seed=5,
+                       mean_interarrival=1.0, iters_range=(5, 40),
+                       gang_sizes=[1, 2, 4, 8])
+    sched = Scheduler(Cluster(16, 8), make_policy("las"))
+    sched.submit(specs)
+    sim = Simulator(sched)
+    sim.run()
+    s = sched.log.summary()
+    assert int(s["n_complete"]) == 24
+    for e in sched.log.events:
+        if e["event"] in ("start", "resume"):
+            nodes = {g // 8 for g in e["gpus"]}
+            assert len(nodes) == 1, f"gang straddles nodes: {e}"
+
+
+def test_simulator_deterministic_replay():
+    """Same trace + policy twice => bitwise-identical event streams."""
+    specs = make_trace(16, mix="philly-slice", seed=9,
+                       mean_interarrival=1.5, iters_range=(5, 50),
+                       gang_sizes=[1, 2])
+    log1, s1 = simulate(specs, make_policy("las"), n_gpus=4)
+    log2, s2 = simulate(specs, make_policy("las"), n_gpus=4)
+    assert log1.events == log2.events
+    assert s1 == s2
+
+
 def test_elastic_resize_events_consistent():
     from gpuschedule_amd.trace.synth import make_elastic_trace
     specs = make_elastic_trace(seed=3, total_iters=200)
