@@ -115,6 +115,19 @@ class Simulator:
             done.append((j.job_id, whole, service, it))
         self.now += dt
         for job_id, whole, service, it in done:
+            job = sched.jobs[job_id]
+            fail_at = job.spec.fail_at_iter
+            if fail_at and job.done_iters + whole >= fail_at:
+                # fault injection: partial progress up to the failure
+                # point, then the job dies mid-run
+                sched.on_progress(job_id,
+                                  max(fail_at - job.done_iters - 1, 0),
+                                  service, self.now, iter_time=it)
+                sched.on_fail(job_id, "injected fault "
+                              f"(fail_at_iter={fail_at})", self.now)
+                self._frac.pop(job_id, None)
+                self._overhead.pop(job_id, None)
+                continue
             finished = sched.on_progress(job_id, whole, service,
                                          self.now, iter_time=it)
             if finished:

@@ -54,6 +54,10 @@ class JobSpec:
     # this tick (the scheduler core ranks background jobs last under
     # every policy).  Used for fill/background workloads.
     background: bool = False
+    # Fault injection (simulator): the job fails once its progress
+    # reaches this iteration — exercises mid-run failure handling
+    # (resource release, cluster continues) in scheduler studies.
+    fail_at_iter: int = 0       # 0 = never
 
     def __post_init__(self) -> None:
         if self.n_gpus < 1:

@@ -4,7 +4,7 @@ Trace schema (CSV, header required; this is the framework's documented
 input format — a subset of the fields in Microsoft's Philly cluster
 trace, named compatibly)::
 
-    job_id,submit_time,num_gpus,iterations,model,batch_size,mem_bytes,elastic,min_gpus,max_gpus,iter_time_hint
+    job_id,submit_time,num_gpus,iterations,model,batch_size,mem_bytes,elastic,min_gpus,max_gpus,iter_time_hint,background,fail_at_iter
 
 Required columns: ``job_id, submit_time, num_gpus, iterations, model``.
 Optional columns default to 0/false and model defaults.
@@ -21,7 +21,7 @@ from .job import JobSpec
 
 REQUIRED = ("job_id", "submit_time", "num_gpus", "iterations", "model")
 OPTIONAL = ("batch_size", "mem_bytes", "elastic", "min_gpus", "max_gpus",
-            "iter_time_hint", "background")
+            "iter_time_hint", "background", "fail_at_iter")
 
 
 def _parse_bool(s: str) -> bool:
@@ -58,6 +58,7 @@ def load_trace(src: Union[str, io.TextIOBase]) -> List[JobSpec]:
                 max_gpus=int(row.get("max_gpus") or 8),
                 iter_time_hint=float(row.get("iter_time_hint") or 0.0),
                 background=_parse_bool(row.get("background") or ""),
+                fail_at_iter=int(row.get("fail_at_iter") or 0),
             ))
         specs.sort(key=lambda s: (s.submit_time, s.job_id))
         ids = [s.job_id for s in specs]
@@ -82,7 +83,8 @@ def save_trace(specs: Iterable[JobSpec], dst: Union[str, io.TextIOBase]) -> None
             w.writerow([s.job_id, s.submit_time, s.n_gpus, s.total_iters,
                         s.model, s.batch_size, s.mem_bytes,
                         int(s.elastic), s.min_gpus, s.max_gpus,
-                        s.iter_time_hint, int(s.background)])
+                        s.iter_time_hint, int(s.background),
+                        s.fail_at_iter])
     finally:
         if close:
             dst.close()

@@ -4,9 +4,9 @@ import json
 from gpuschedule_amd import EventLog, JobSpec, make_policy, make_trace, simulate
 
 
-def J(jid, t=0.0, n=1, iters=10, hint=1.0, model="resnet18"):
+def J(jid, t=0.0, n=1, iters=10, hint=1.0, model="resnet18", **kw):
     return JobSpec(job_id=jid, submit_time=t, n_gpus=n, total_iters=iters,
-                   model=model, iter_time_hint=hint)
+                   model=model, iter_time_hint=hint, **kw)
 
 
 def test_single_job_jct():
@@ -104,6 +104,20 @@ def test_contention_separates_policies():
     _, fifo = simulate(specs, make_policy("fifo"), n_gpus=1)
     _, srtf = simulate(specs, make_policy("srtf"), n_gpus=1)
     assert srtf["avg_jct"] < 0.5 * fifo["avg_jct"]
+
+
+def test_fault_injection_mid_run():
+    # Job "dies" at iteration 5 of 50: it fails, frees its GPU, and the
+    # queued job behind it still completes.
+    specs = [J("dies", iters=50, fail_at_iter=5),
+             J("ok", t=0.5, iters=5)]
+    log, s = simulate(specs, make_policy("fifo"), n_gpus=1)
+    fails = [e for e in log.events if e["event"] == "fail"]
+    assert len(fails) == 1 and fails[0]["job"] == "dies"
+    assert "injected fault" in fails[0]["error"]
+    assert int(s["n_complete"]) == 1
+    completes = [e["job"] for e in log.events if e["event"] == "complete"]
+    assert completes == ["ok"]
 
 
 def test_unsatisfiable_job_fails_at_admission():
