@@ -109,6 +109,25 @@ class TorchBackend:
                 r.release()
                 del self.runners[job_id]
             # reason "policy": suspend-in-HBM — runner stays resident.
+        # SUSPENDED (HBM-resident) jobs that the scheduler migrates or
+        # resizes never appear in plan.preempts (only RUNNING jobs do);
+        # they arrive as placements with migrate_from/resize_from set.
+        # Their resident runner holds the ONLY copy of the training
+        # state — persist it before phase B drops the stale runner,
+        # otherwise the destination rank rebuilds from initial weights.
+        preempted = {job_id for job_id, _ in plan.preempts}
+        for p in plan.places:
+            if p.migrate_from is None and p.resize_from is None:
+                continue
+            if p.job_id in preempted:
+                continue  # already checkpointed above
+            r = self.runners.get(p.job_id)
+            if r is None:
+                continue
+            if r.gang_rank == 0 and not p.from_evicted:
+                self.ckpt.save(p.job_id, r.model, r.optim, r.done_iters)
+            r.release()
+            del self.runners[p.job_id]
         for job_id in plan.evictions:
             r = self.runners.get(job_id)
             if r is not None:

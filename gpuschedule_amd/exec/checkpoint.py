@@ -15,7 +15,11 @@ MI355X-first design:
 
 Checkpoint file format (documented, stable): ``torch.save`` of
 ``{"job_id", "done_iters", "model": state_dict, "optim": state_dict,
-"torch_rng"}`` — all tensors on CPU.
+"torch_rng", "cuda_rng"}`` — all tensors on CPU.  The RNG states are
+the saving (gang-leader) rank's global torch CPU/device generator
+states, restored on load so stochastic ops (dropout) resume their
+stream; per-rank synthetic-data streams are independently seeded by
+(job, rank, iteration) and need no saved state.
 """
 from __future__ import annotations
 
@@ -26,19 +30,37 @@ import torch
 
 
 class PinnedPool:
-    """Reusable pinned staging buffers, keyed by size class."""
+    """Reusable pinned staging buffers, keyed by size class.
+
+    ``acquire`` hands out a free buffer (allocating when none is
+    cached) so several D2H copies can be in flight at once; the saver
+    calls ``release`` after the single final sync, returning buffers
+    for reuse by the next checkpoint."""
 
     def __init__(self):
-        self._bufs: Dict[int, torch.Tensor] = {}
+        self._free: Dict[int, list] = {}
 
-    def get(self, nbytes: int) -> torch.Tensor:
+    def acquire(self, nbytes: int):
         # round to 1 MiB size classes
-        size = max(1 << 20, 1 << (nbytes - 1).bit_length())
-        buf = self._bufs.get(size)
-        if buf is None:
-            buf = torch.empty(size, dtype=torch.uint8, pin_memory=True)
-            self._bufs[size] = buf
-        return buf
+        size = max(1 << 20, 1 << max((nbytes - 1).bit_length(), 0))
+        lst = self._free.get(size)
+        if lst:
+            return size, lst.pop()
+        return size, torch.empty(size, dtype=torch.uint8, pin_memory=True)
+
+    def release(self, size: int, buf: torch.Tensor) -> None:
+        self._free.setdefault(size, []).append(buf)
+
+
+class _Staged:
+    """A tensor whose D2H copy is in flight on the side stream."""
+    __slots__ = ("view", "shape", "size", "buf")
+
+    def __init__(self, view, shape, size, buf):
+        self.view = view
+        self.shape = shape
+        self.size = size
+        self.buf = buf
 
 
 class CheckpointStore:
@@ -57,21 +79,24 @@ class CheckpointStore:
     def exists(self, job_id: str) -> bool:
         return os.path.exists(self.path(job_id))
 
-    def _stage_to_cpu(self, t: torch.Tensor) -> torch.Tensor:
-        """D2H via pinned buffer on the side stream."""
+    def _stage_to_cpu(self, t: torch.Tensor):
+        """Issue the D2H copy on the side stream; NO per-tensor sync.
+        Returns a :class:`_Staged` placeholder that :meth:`_finalize`
+        resolves after ONE sync covering every in-flight copy — the
+        copies of a multi-GB state_dict pipeline back-to-back instead
+        of paying launch+sync latency per tensor."""
         if not t.is_cuda:
             return t.detach().clone()
         if self._side is None:
             self._side = torch.cuda.Stream()
         flat = t.detach().reshape(-1)
         nbytes = flat.numel() * flat.element_size()
-        pinned = self._pool.get(nbytes)
-        view = pinned[:nbytes].view(t.dtype)[:flat.numel()]
+        size, buf = self._pool.acquire(nbytes)
+        view = buf[:nbytes].view(t.dtype)[:flat.numel()]
         with torch.cuda.stream(self._side):
             self._side.wait_stream(torch.cuda.current_stream())
             view.copy_(flat, non_blocking=True)
-        self._side.synchronize()
-        return view.clone().reshape(t.shape)
+        return _Staged(view, t.shape, size, buf)
 
     def _tree_to_cpu(self, obj):
         if isinstance(obj, torch.Tensor):
@@ -80,6 +105,21 @@ class CheckpointStore:
             return {k: self._tree_to_cpu(v) for k, v in obj.items()}
         if isinstance(obj, (list, tuple)):
             t = [self._tree_to_cpu(v) for v in obj]
+            return type(obj)(t) if not isinstance(obj, tuple) else tuple(t)
+        return obj
+
+    def _finalize(self, obj):
+        """Resolve _Staged placeholders after the single side-stream
+        sync: clone out of the pinned buffer and return it to the
+        pool."""
+        if isinstance(obj, _Staged):
+            out = obj.view.clone().reshape(obj.shape)
+            self._pool.release(obj.size, obj.buf)
+            return out
+        if isinstance(obj, dict):
+            return {k: self._finalize(v) for k, v in obj.items()}
+        if isinstance(obj, (list, tuple)):
+            t = [self._finalize(v) for v in obj]
             return type(obj)(t) if not isinstance(obj, tuple) else tuple(t)
         return obj
 
@@ -92,7 +132,13 @@ class CheckpointStore:
             "model": self._tree_to_cpu(model.state_dict()),
             "optim": self._tree_to_cpu(optim.state_dict())
             if optim is not None else None,
+            "torch_rng": torch.get_rng_state(),
+            "cuda_rng": (torch.cuda.get_rng_state()
+                         if torch.cuda.is_available() else None),
         }
+        if self._side is not None:
+            self._side.synchronize()  # one sync for ALL staged copies
+        state = self._finalize(state)
         p = self.path(job_id)
         tmp = p + ".tmp"
         torch.save(state, tmp)

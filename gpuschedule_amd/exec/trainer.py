@@ -95,6 +95,15 @@ class JobRunner:
         self.model.load_state_dict(state["model"])
         self.optim.load_state_dict(state["optim"])
         self.done_iters = state.get("done_iters", self.done_iters)
+        # Restore the saver's global RNG streams so stochastic ops
+        # (dropout) resume where the checkpoint left off; synthetic
+        # data streams are independently (job, rank, iter)-seeded.
+        rng = state.get("torch_rng")
+        if rng is not None:
+            torch.set_rng_state(rng)
+        crng = state.get("cuda_rng")
+        if crng is not None and torch.cuda.is_available():
+            torch.cuda.set_rng_state(crng, self.device)
 
     # ---- execution ----------------------------------------------
 

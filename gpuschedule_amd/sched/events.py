@@ -7,9 +7,12 @@ object per line)::
 
 Event kinds and extra fields:
 
-* ``submit``    — n_gpus, model, iters
+* ``submit``    — n_gpus, model, iters, ideal_s (contention-free solo
+  runtime, used by the finish-time-slowdown metric)
 * ``start``     — gpus (list), first (bool: first ever start)
-* ``preempt``   — reason ("policy"|"quantum"|"resize"|"migrate"), evicted
+* ``preempt``   — reason ("policy"|"resize"|"migrate")
+* ``evict``     — job's state left HBM for a host checkpoint
+  (standalone event; follows a preempt when the job was running)
 * ``resume``    — gpus
 * ``migrate``   — src, dst (lists)
 * ``resize``    — old_gpus, new_gpus (GPU counts before/after)
@@ -101,10 +104,10 @@ class EventLog:
             elif kind in ("preempt", "complete", "fail"):
                 if job in open_runs:
                     close(job, t)
-                if kind == "preempt" and e.get("evicted"):
-                    ev.append({"ph": "i", "name": f"evict {job}",
-                               "pid": 0, "tid": "sched", "ts": t * US,
-                               "s": "g"})
+            elif kind == "evict":
+                ev.append({"ph": "i", "name": f"evict {job}",
+                           "pid": 0, "tid": "sched", "ts": t * US,
+                           "s": "g"})
             elif kind in ("migrate", "resize"):
                 ev.append({"ph": "i", "name": f"{kind} {job}", "pid": 0,
                            "tid": "sched", "ts": t * US, "s": "g"})

@@ -275,6 +275,13 @@ class Scheduler:
                 return False
             alloc[job.job_id] = best
             occupy(best)
+            # Record the shadow memory delta: jobs placed later this
+            # tick must see this job's footprint on GPUs where it is
+            # not already resident (free_mem only reads the cluster's
+            # committed residency + `extra`).
+            for i in best:
+                if job.job_id not in C.gpu(i).resident:
+                    extra[i] = extra.get(i, 0) + mem
             return True
 
         for job_id, width in requests:

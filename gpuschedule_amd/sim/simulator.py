@@ -119,10 +119,14 @@ class Simulator:
             fail_at = job.spec.fail_at_iter
             if fail_at and job.done_iters + whole >= fail_at:
                 # fault injection: partial progress up to the failure
-                # point, then the job dies mid-run
-                sched.on_progress(job_id,
-                                  max(fail_at - job.done_iters - 1, 0),
-                                  service, self.now, iter_time=it)
+                # point, then the job dies mid-run.  Attained service
+                # is scaled to the fraction of this slice's iterations
+                # actually completed before the fault.
+                completed = max(fail_at - job.done_iters - 1, 0)
+                frac_done = completed / whole if whole > 0 else 0.0
+                sched.on_progress(job_id, completed,
+                                  service * frac_done, self.now,
+                                  iter_time=it)
                 sched.on_fail(job_id, "injected fault "
                               f"(fail_at_iter={fail_at})", self.now)
                 self._frac.pop(job_id, None)

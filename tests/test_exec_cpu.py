@@ -107,6 +107,62 @@ def test_runner_checkpoint_determinism(tmp_path):
             f"restore diverged at {n1}"
 
 
+def test_suspended_job_migration_preserves_state(tmp_path):
+    """A SUSPENDED (HBM-resident, never-evicted) job that the scheduler
+    migrates appears only as a Placement with migrate_from set (no
+    preempt entry).  Phase A must persist its state before the stale
+    runner is dropped, or training silently restarts from initial
+    weights on the destination rank."""
+    from gpuschedule_amd.sched.scheduler import Plan, Placement
+
+    specs = [J("j", iters=100)]
+    be = make_backend(specs, ckpt_root=str(tmp_path / "ck"))
+    # Place + step the job so its state diverges from the seeded init.
+    be.tick_once()
+    assert "j" in be.runners
+    r = be.runners["j"]
+    stepped = {k: v.clone() for k, v in r.model.state_dict().items()}
+    done = r.done_iters
+    assert done > 0
+
+    # Scheduler suspends it (reason "policy": stays HBM-resident) ...
+    be._apply_phase_a(Plan(preempts=[("j", "policy")]), {})
+    assert "j" in be.runners and not be.ckpt.exists("j")
+
+    # ... then migrates it to another GPU.  This rank (0) is NOT in the
+    # destination; the only state copy is the resident runner.
+    plan = Plan(places=[Placement(job_id="j", gpus=[1], resume=True,
+                                  migrate_from=[0])])
+    meta = {"j": ("llama3-tiny", 2, [1], done)}
+    be._apply_phase_a(plan, meta)
+    assert "j" not in be.runners, "stale runner must be dropped"
+    assert be.ckpt.exists("j"), "state must be checkpointed in phase A"
+    state = be.ckpt.load("j")
+    assert state["done_iters"] == done
+    for k, v in stepped.items():
+        assert torch.allclose(state["model"][k].float(), v.float(),
+                              atol=1e-6), f"state lost at {k}"
+
+
+def test_packing_never_overcommits_memory():
+    """Two llama3-8b jobs (150 GB each) must not be packed onto one
+    288 GB GPU: the shadow memory map has to see the first job's
+    footprint when the second is considered in the same tick."""
+    from gpuschedule_amd.sched.policies.gandiva import GandivaPolicy
+
+    sched = Scheduler(Cluster(1, 1),
+                      GandivaPolicy(quantum=60.0, packing=True,
+                                    pack_capacity=2))
+    sched.submit([J("a", model="llama3-8b", iters=10, batch=2),
+                  J("b", model="llama3-8b", iters=10, batch=2)])
+    plan = sched.tick(0.0)  # must not raise (Cluster.place assert)
+    placed = {p.job_id for p in plan.places}
+    assert placed == {"a"}, placed
+    running = [j.job_id for j in sched.jobs.values()
+               if j.state.name == "RUNNING"]
+    assert running == ["a"]
+
+
 # ---- gloo world-2 ---------------------------------------------------
 
 def _worker(rank, world, port, scenario, out_file):
