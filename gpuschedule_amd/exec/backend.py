@@ -38,7 +38,7 @@ from .. import catalog
 from ..sched.scheduler import Plan, Scheduler
 from .checkpoint import CheckpointStore
 from .gangs import GangGroups
-from .trainer import JobRunner
+from .trainer import RUNNER_POOL, JobRunner
 
 
 @dataclass
@@ -154,10 +154,78 @@ class TorchBackend:
             if self.ckpt.exists(p.job_id):
                 state = self.ckpt.load(p.job_id)
             group = self.gangs.get(gpus)
-            self.runners[p.job_id] = JobRunner(
-                p.job_id, model_name, batch, self.device, self.dtype,
+            key = (model_name, batch, tuple(gpus))
+            if state is None and done_iters == 0:
+                # fresh job: adopt a warm pooled runner (same model /
+                # batch / gang ⇒ same graph, optimizer tables, stream)
+                pooled = RUNNER_POOL.get(key)
+                if pooled is not None:
+                    pooled.adopt(p.job_id)
+                    self.runners[p.job_id] = pooled
+                    continue
+            self.runners[p.job_id] = self._build_runner(
+                p.job_id, model_name, batch, group, gpus, done_iters,
+                state, key)
+
+    def _build_runner(self, job_id, model_name, batch, group, gpus,
+                      done_iters, state, key) -> JobRunner:
+        try:
+            return JobRunner(
+                job_id, model_name, batch, self.device, self.dtype,
                 group, gpus.index(self.rank), len(gpus),
-                done_iters=done_iters, state=state)
+                done_iters=done_iters, state=state, pool_key=key)
+        except torch.cuda.OutOfMemoryError:
+            # pooled runners hold HBM the scheduler's cluster model
+            # does not account for — reclaim it all and retry once
+            RUNNER_POOL.flush()
+            torch.cuda.empty_cache()
+            return JobRunner(
+                job_id, model_name, batch, self.device, self.dtype,
+                group, gpus.index(self.rank), len(gpus),
+                done_iters=done_iters, state=state, pool_key=key)
+
+    # ---- warm pool ----------------------------------------------
+
+    def prewarm(self, families: List[str], count: int = 1,
+                iters: int = 6) -> None:
+        """Pre-build `count` single-GPU runners per model family on
+        this rank, run `iters` training iterations each (eager warmup
+        + hipGraph capture + first-touch MIOpen find + allocator
+        growth), and park them in the process-level runner pool.  Jobs
+        the scheduler later places on this rank adopt a warm runner
+        instead of paying cold construction inside the timed path."""
+        import itertools
+        for i, fam in itertools.product(range(count), families):
+            batch = catalog.model_meta(fam).default_batch
+            key = (fam, batch, (self.rank,))
+            r = self._build_runner(f"_warm_{fam}_{i}", fam, batch,
+                                   None, [self.rank], 0, None, key)
+            r.step(iters)
+            if not RUNNER_POOL.put(r):
+                r.release()
+
+    def prewarm_gang(self, family: str, gpus: List[int],
+                     iters: int = 6) -> None:
+        """Gang-runner prewarm: every rank in `gpus` must call this at
+        the same point (the warmup iterations run the gang's bucketed
+        all-reduce collectively, warming the RCCL communicator)."""
+        if self.rank not in gpus:
+            return
+        batch = catalog.model_meta(family).default_batch
+        key = (family, batch, tuple(gpus))
+        r = self._build_runner(f"_warm_gang_{family}", family, batch,
+                               self.gangs.get(gpus), gpus, 0, None, key)
+        r.step(iters)
+        if not RUNNER_POOL.put(r):
+            r.release()
+
+    def retire_all(self) -> None:
+        """Release every hosted runner into the pool (end of a replay
+        phase; pooled runners stay warm for the next backend)."""
+        for r in self.runners.values():
+            if not RUNNER_POOL.put(r):
+                r.release()
+        self.runners.clear()
 
     # ---- one tick -----------------------------------------------
 
@@ -185,7 +253,7 @@ class TorchBackend:
 
         for job_id in msg.cleanup:
             r = self.runners.pop(job_id, None)
-            if r is not None:
+            if r is not None and not RUNNER_POOL.put(r):
                 r.release()
             if self.rank == 0:
                 self.ckpt.drop(job_id)
@@ -209,6 +277,7 @@ class TorchBackend:
                 r.step_async(n)
                 launched.append((job_id, r, n))
             except Exception as e:  # job-level fault isolation
+                r.poolable = False  # possibly corrupt: never reuse
                 report[job_id] = ("fail", 0, 0.0, f"{type(e).__name__}: {e}")
         for job_id, r, n in launched:
             try:
@@ -220,6 +289,7 @@ class TorchBackend:
                         r._mem_reported = True
                     report[job_id] = ("ok", done, secs, r.last_loss, mem)
             except Exception as e:
+                r.poolable = False  # possibly corrupt: never reuse
                 report[job_id] = ("fail", 0, 0.0, f"{type(e).__name__}: {e}")
 
         reports = self._gather(report)

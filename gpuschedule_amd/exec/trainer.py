@@ -2,10 +2,21 @@
 
 One JobRunner per (job, rank).  The model lives on this rank's GPU (or
 CPU in tests); gang members share a pre-created RCCL group and reduce
-grads through parallel/ddp.GradReducer.  Weights are seeded by job_id
-so every gang member constructs an identical replica without any
-broadcast; data is seeded by (job, rank, iteration) so DP shards
-differ and a migrated/restored job replays a deterministic stream.
+grads through parallel/ddp.GradReducer.  Weights are seeded by MODEL
+FAMILY so every gang member constructs an identical replica without
+any broadcast AND a pooled runner re-seeded in place is bit-identical
+to a fresh build; data is seeded by (job, rank, iteration) so DP
+shards differ and a migrated/restored job replays a deterministic
+stream.
+
+Warm runner pool: building a runner on-GPU costs model construction +
+eager warmup + hipGraph capture + first-touch MIOpen find — hundreds
+of ms that would otherwise land inside the scheduler's timed path for
+every arriving job.  RUNNER_POOL keeps released runners alive, keyed
+by (model, batch, gang); :meth:`JobRunner.adopt` rebinds one to a new
+job by copying the family's pristine-init template back into the SAME
+parameter/optimizer-state tensors, so the captured hipGraph (whose
+baked device pointers reference those tensors) replays unchanged.
 """
 from __future__ import annotations
 
@@ -25,11 +36,81 @@ def _job_seed(job_id: str) -> int:
     return zlib.crc32(job_id.encode()) & 0x7FFFFFFF
 
 
+# Pristine-init templates per (family, device, dtype): a clone of the
+# state_dict right after seeded construction.  Because construction is
+# seeded by FAMILY, the template equals what any fresh build produces,
+# so JobRunner.adopt(template copy) == fresh JobRunner, bit for bit.
+# Families above the size gate (Llama-8B) skip caching (and pooling).
+_TEMPLATE_MAX_BYTES = 4 << 30
+_INIT_TEMPLATES: dict = {}
+
+
+def _template_key(model_name: str, device, dtype) -> tuple:
+    return (model_name, str(device), str(dtype))
+
+
+class RunnerPool:
+    """Process-level cache of released JobRunners, keyed by
+    (model, batch, gang-gpus).  Budgeted by the runners' measured
+    per-job memory (the scheduler's cluster model does not account for
+    pooled memory, so the budget keeps hoarding bounded; the backend
+    flushes the pool and retries on OOM)."""
+
+    def __init__(self, max_bytes: Optional[int] = None):
+        if max_bytes is None:
+            max_bytes = int(os.environ.get("GSAMD_POOL_BYTES",
+                                           32 << 30))
+        self.max_bytes = max_bytes
+        self._pool: dict = {}      # key -> list[JobRunner]
+        self._order: list = []     # LRU of (key, runner)
+        self._bytes = 0
+
+    def _cost(self, r) -> int:
+        return max(r.measured_mem, 1 << 20)
+
+    def put(self, runner) -> bool:
+        if runner is None or runner.pool_key is None \
+                or runner.model is None or not runner.poolable:
+            return False
+        cost = self._cost(runner)
+        if cost > self.max_bytes:
+            return False
+        while self._bytes + cost > self.max_bytes and self._order:
+            k, old = self._order.pop(0)
+            self._pool[k].remove(old)
+            self._bytes -= self._cost(old)
+            old.release()
+        self._pool.setdefault(runner.pool_key, []).append(runner)
+        self._order.append((runner.pool_key, runner))
+        self._bytes += cost
+        return True
+
+    def get(self, key):
+        lst = self._pool.get(key)
+        if not lst:
+            return None
+        r = lst.pop()
+        self._order.remove((key, r))
+        self._bytes -= self._cost(r)
+        return r
+
+    def flush(self) -> None:
+        for k, r in self._order:
+            r.release()
+        self._pool.clear()
+        self._order.clear()
+        self._bytes = 0
+
+
+RUNNER_POOL = RunnerPool()
+
+
 class JobRunner:
     def __init__(self, job_id: str, model_name: str, batch_size: int,
                  device, dtype, gang_group, gang_rank: int,
                  gang_world: int, done_iters: int = 0,
-                 state: Optional[dict] = None):
+                 state: Optional[dict] = None,
+                 pool_key: Optional[tuple] = None):
         self.job_id = job_id
         self.model_name = model_name
         self.batch_size = batch_size
@@ -39,13 +120,28 @@ class JobRunner:
         self.gang_rank = gang_rank
         self.gang_world = gang_world
         self.done_iters = done_iters
+        self.pool_key = pool_key
 
+        # Seeded by FAMILY (not job) so a pooled runner re-seeded from
+        # the family template is identical to a fresh build on every
+        # gang member — see module docstring.
         self.model, self.make_batch, default_bs = build_model(
             model_name, device=device, dtype=dtype,
-            seed=_job_seed(job_id))
+            seed=_job_seed(model_name))
         if device == "cpu" or str(device) == "cpu":
             self.model = self.model.float()
         self.batch_size = batch_size or default_bs
+        eff_dtype = (torch.float32 if str(device) == "cpu" else dtype)
+        tkey = _template_key(model_name, device, eff_dtype)
+        if tkey not in _INIT_TEMPLATES:
+            sd = self.model.state_dict()
+            nbytes = sum(v.numel() * v.element_size()
+                         for v in sd.values())
+            _INIT_TEMPLATES[tkey] = (
+                {k: v.detach().clone() for k, v in sd.items()}
+                if nbytes <= _TEMPLATE_MAX_BYTES else None)
+        self.poolable = _INIT_TEMPLATES[tkey] is not None
+        self._tkey = tkey
         if model_name.startswith("resnet"):
             self.optim = FusedSGD(self.model.parameters(), lr=0.05,
                                   momentum=0.9, weight_decay=5e-4)
@@ -104,6 +200,29 @@ class JobRunner:
         crng = state.get("cuda_rng")
         if crng is not None and torch.cuda.is_available():
             torch.cuda.set_rng_state(crng, self.device)
+
+    @torch.no_grad()
+    def adopt(self, job_id: str, done_iters: int = 0) -> None:
+        """Rebind this (pooled) runner to a new job: copy the family's
+        pristine-init template back into the SAME parameter tensors
+        and reset the optimizer state in place, so the captured
+        hipGraph and the multi-tensor optimizer pointer tables stay
+        valid.  The result is bit-identical to a fresh build (family-
+        seeded init); data streams re-seed from the new job_id."""
+        tpl = _INIT_TEMPLATES[self._tkey]
+        sd = self.model.state_dict()
+        for k, v in tpl.items():
+            sd[k].copy_(v)
+        for p in self.model.parameters():
+            if p.grad is not None:
+                p.grad.zero_()
+        self.optim.reset_state()
+        self.job_id = job_id
+        self.done_iters = done_iters
+        self.last_loss = float("nan")
+        self._loss_ref = None
+        self._pending = 0
+        self._mem_reported = False
 
     # ---- execution ----------------------------------------------
 

@@ -32,6 +32,29 @@ class _MasterStateMixin:
                         st[k] = st[k].to(device=p.device,
                                          dtype=torch.float32)
 
+    @torch.no_grad()
+    def reset_state(self):
+        """Reset optimizer state IN PLACE (fresh-job semantics) without
+        replacing any state tensor: the multi-tensor pointer tables and
+        any hipGraph that baked them stay valid.  Masters re-snapshot
+        the (re-initialized) params; moments/momentum zero; the device
+        step buffer rewinds to 0."""
+        for group in self.param_groups:
+            for p in group["params"]:
+                st = self.state.get(p)
+                if not st:
+                    continue
+                if "step" in st:
+                    st["step"] = 0
+                if "p32" in st:
+                    st["p32"].copy_(p.detach().float())
+                for k in ("m", "v", "buf"):
+                    if k in st:
+                        st[k].zero_()
+        mt = getattr(self, "_mt", None)
+        if isinstance(mt, tuple):   # AdamW: (..., stepbuf, params)
+            mt[4].zero_()
+
 
 def _mt_eligible(p) -> bool:
     """Param qualifies for the flat multi-tensor kernels: any DENSE

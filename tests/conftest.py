@@ -6,6 +6,15 @@ def pytest_configure(config):
         "markers", "gpu: test requires a real MI355X GPU (run via gpurun)")
 
 
+@pytest.fixture(autouse=True)
+def _fresh_runner_pool():
+    """Isolate tests from the process-global warm runner pool."""
+    from gpuschedule_amd.exec.trainer import RUNNER_POOL
+    RUNNER_POOL.flush()
+    yield
+    RUNNER_POOL.flush()
+
+
 def pytest_collection_modifyitems(config, items):
     try:
         import torch

@@ -107,6 +107,49 @@ def test_runner_checkpoint_determinism(tmp_path):
             f"restore diverged at {n1}"
 
 
+def test_runner_pool_adopt_matches_fresh_build():
+    """An adopted pooled runner must be bit-identical to a fresh build
+    (same family-seeded init, same optimizer trajectory)."""
+    from gpuschedule_amd.exec.trainer import RUNNER_POOL
+
+    r1 = JobRunner("first", "llama3-tiny", 2, "cpu", torch.float32,
+                   None, 0, 1, pool_key=("llama3-tiny", 2, (0,)))
+    r1.step(3)  # diverge from init
+    assert RUNNER_POOL.put(r1)
+    pooled = RUNNER_POOL.get(("llama3-tiny", 2, (0,)))
+    assert pooled is r1
+    pooled.adopt("second")
+    assert pooled.done_iters == 0
+
+    fresh = JobRunner("second", "llama3-tiny", 2, "cpu", torch.float32,
+                      None, 0, 1)
+    pooled.step(3)
+    fresh.step(3)
+    assert pooled.last_loss == pytest.approx(fresh.last_loss, rel=1e-6)
+    for (n1, p1), (n2, p2) in zip(pooled.model.named_parameters(),
+                                  fresh.model.named_parameters()):
+        assert torch.equal(p1, p2), f"adopt diverged from fresh at {n1}"
+
+
+def test_backend_reuses_pooled_runner(tmp_path):
+    """Two same-family jobs in sequence: the second must adopt the
+    first's runner instead of building cold."""
+    specs = [J("a", iters=2), J("b", t=0.1, iters=2)]
+    be = make_backend(specs, ckpt_root=str(tmp_path / "ck"))
+    ran = {}
+    orig_phase_b = be._apply_phase_b
+
+    def spy(plan, meta):
+        orig_phase_b(plan, meta)
+        for p in plan.places:
+            if p.job_id in be.runners:
+                ran[p.job_id] = id(be.runners[p.job_id])
+    be._apply_phase_b = spy
+    log = be.run(max_seconds=300)
+    assert int(log.summary()["n_complete"]) == 2
+    assert ran["a"] == ran["b"], "second job should adopt pooled runner"
+
+
 def test_suspended_job_migration_preserves_state(tmp_path):
     """A SUSPENDED (HBM-resident, never-evicted) job that the scheduler
     migrates appears only as a Placement with migrate_from set (no
