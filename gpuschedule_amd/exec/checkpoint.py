@@ -53,14 +53,18 @@ class PinnedPool:
 
 
 class _Staged:
-    """A tensor whose D2H copy is in flight on the side stream."""
-    __slots__ = ("view", "shape", "size", "buf")
+    """A tensor whose D2H copy is in flight on the side stream.  Holds
+    a reference to the SOURCE (which may be a reshape temporary) so the
+    allocator cannot recycle its memory before the copy lands — the
+    per-tensor sync that used to provide that guarantee is gone."""
+    __slots__ = ("view", "shape", "size", "buf", "src")
 
-    def __init__(self, view, shape, size, buf):
+    def __init__(self, view, shape, size, buf, src):
         self.view = view
         self.shape = shape
         self.size = size
         self.buf = buf
+        self.src = src
 
 
 class CheckpointStore:
@@ -96,7 +100,7 @@ class CheckpointStore:
         with torch.cuda.stream(self._side):
             self._side.wait_stream(torch.cuda.current_stream())
             view.copy_(flat, non_blocking=True)
-        return _Staged(view, t.shape, size, buf)
+        return _Staged(view, t.shape, size, buf, flat)
 
     def _tree_to_cpu(self, obj):
         if isinstance(obj, torch.Tensor):

@@ -14,6 +14,7 @@ import torch.nn.functional as F
 
 from ..ops.attention import flash_attention_qkv
 from ..ops.ce import fused_cross_entropy
+from ..ops.embedding import FusedEmbedding
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedLayerNorm
 
@@ -58,8 +59,8 @@ class BertMLM(nn.Module):
     def __init__(self, cfg: BertConfig = BertConfig()):
         super().__init__()
         self.cfg = cfg
-        self.tok = nn.Embedding(cfg.vocab_size, cfg.hidden)
-        self.pos = nn.Embedding(cfg.max_seq, cfg.hidden)
+        self.tok = FusedEmbedding(cfg.vocab_size, cfg.hidden)
+        self.pos = FusedEmbedding(cfg.max_seq, cfg.hidden)
         self.ln = FusedLayerNorm(cfg.hidden)
         self.layers = nn.ModuleList(BertLayer(cfg)
                                     for _ in range(cfg.layers))
@@ -68,7 +69,7 @@ class BertMLM(nn.Module):
 
     @staticmethod
     def _init(m):
-        if isinstance(m, (nn.Embedding,)):
+        if isinstance(m, (nn.Embedding, FusedEmbedding)):
             nn.init.normal_(m.weight, std=0.02)
 
     def forward(self, tokens):

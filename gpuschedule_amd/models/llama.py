@@ -17,6 +17,7 @@ import torch.nn as nn
 from ..ops.attention import flash_attention
 from ..ops.ce import fused_cross_entropy
 from ..ops.elementwise import fused_rope, silu_mul
+from ..ops.embedding import FusedEmbedding
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedRMSNorm
 
@@ -93,7 +94,7 @@ class LlamaLM(nn.Module):
     def __init__(self, cfg: LlamaConfig = LLAMA3_TINY):
         super().__init__()
         self.cfg = cfg
-        self.tok = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.tok = FusedEmbedding(cfg.vocab_size, cfg.hidden)
         nn.init.normal_(self.tok.weight, std=0.02)
         self.layers = nn.ModuleList(LlamaLayer(cfg)
                                     for _ in range(cfg.layers))

@@ -17,7 +17,7 @@ BUILD_DIR = os.path.join(HERE, "_build")
 SOURCES = [os.path.join(HERE, "hip", f)
            for f in ("bindings.cpp", "norms.hip", "gemm_mfma.hip",
                      "optim.hip", "attention.hip", "bn.hip", "ce.hip",
-                     "elementwise.hip")]
+                     "elementwise.hip", "embedding.hip")]
 EXT_NAME = "_hip_ops"
 
 

@@ -53,6 +53,8 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits,
                                   torch::Tensor labels);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
                      torch::Tensor lse, torch::Tensor gscale);
+torch::Tensor embed_fwd(torch::Tensor weight, torch::Tensor idx);
+torch::Tensor embed_bwd(torch::Tensor dy, torch::Tensor idx, long V);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> y,
                                   torch::Tensor mean, torch::Tensor rstd,
@@ -86,6 +88,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_multi", &sgd_multi, "multi-tensor fused SGD (one launch/dtype, zeroes grads)");
   m.def("adamw_multi", &adamw_multi, "multi-tensor fused AdamW (one launch, zeroes grads)");
   m.def("sgd_step_fp32", &sgd_step_fp32, "fused fp32 SGD step");
+  m.def("embed_fwd", &embed_fwd, "embedding gather (bf16)");
+  m.def("embed_bwd", &embed_bwd,
+        "embedding dense grad: fp32 atomic scatter + bf16 cast");
   m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+ReLU) forward, channels_last");
   m.def("bn_bwd", &bn_bwd,
         "fused BatchNorm(+ReLU) backward, channels_last; pass beta "

@@ -445,3 +445,49 @@ def test_multi_tensor_sgd_matches_torch():
                       list(b_bf.parameters()) + list(b_fp.parameters())):
         err = (pa.float() - pb.float()).abs().max().item()
         assert err < 3e-2, f"MT sgd drift {err}"
+
+
+# ---- fused embedding ------------------------------------------------
+
+@pytest.mark.parametrize("vh", [(30528, 768), (1024, 256), (997, 64)])
+def test_fused_embedding_fwd_bwd(vh):
+    """Gather fwd + fp32-atomic-scatter bwd vs fp32 F.embedding,
+    including repeated tokens (accumulation in the dense grad)."""
+    torch.manual_seed(3)
+    V, H = vh
+    n = 4096
+    w = torch.randn(V, H, device="cuda", dtype=torch.bfloat16)
+    # force collisions: draw from a narrow band so rows repeat
+    idx = torch.randint(0, min(V, 512), (n,), device="cuda")
+    dy = torch.randn(n, H, device="cuda", dtype=torch.bfloat16)
+
+    out = ext().embed_fwd(w, idx)
+    ref = torch.nn.functional.embedding(idx, w.float())
+    assert torch.equal(out.float(), ref.to(torch.bfloat16).float())
+
+    dw = ext().embed_bwd(dy, idx, V)
+    wf = w.float().requires_grad_(True)
+    torch.nn.functional.embedding(idx, wf).backward(dy.float())
+    scale = wf.grad.abs().max().item() + 1e-3
+    assert mx(dw, wf.grad) / scale < 1.2e-2
+    # untouched rows must be exactly zero
+    mask = torch.ones(V, dtype=torch.bool, device="cuda")
+    mask[idx] = False
+    assert dw[mask].abs().max().item() == 0.0
+
+
+def test_fused_embedding_module_autograd():
+    from gpuschedule_amd.ops.embedding import FusedEmbedding
+    torch.manual_seed(4)
+    m = FusedEmbedding(256, 128).to("cuda", torch.bfloat16)
+    idx = torch.randint(0, 256, (8, 32), device="cuda")
+    y = m(idx)
+    assert y.shape == (8, 32, 128) and y.dtype == torch.bfloat16
+    y.float().pow(2).mean().backward()
+    assert m.weight.grad is not None
+    ref = torch.nn.functional.embedding(idx, m.weight.detach().float())
+    ref.requires_grad_(False)
+    wf = m.weight.detach().float().requires_grad_(True)
+    torch.nn.functional.embedding(idx, wf).float().pow(2).mean().backward()
+    scale = wf.grad.abs().max().item() + 1e-4
+    assert mx(m.weight.grad, wf.grad) / scale < 2e-2
