@@ -28,7 +28,6 @@ from __future__ import annotations
 import argparse
 import json
 import os
-import random
 import sys
 import time
 
@@ -43,29 +42,41 @@ def build_trace(n_jobs: int, steps: int, n_gpus: int, iters_per_tick: int,
                 seed: int = 1234):
     """Single-GPU job mix whose total tick-slot demand fits 70% of the
     K x N slot budget (identical job set at every N for strong
-    scaling).  Arrivals are TICK-INDEXED (injected by the bench loop
-    over the first 25% of the K ticks) so the arrival process lands
-    inside the measurement window regardless of hardware speed.
+    scaling).  Job SERVICE DEMANDS and ARRIVAL ORDER come from the
+    Philly-trace sample (data/philly_sample.json, the msr-fiddle
+    cluster_job_log schema — see trace/philly_log.py): each trace job's
+    logged GPU-time becomes its share of the tick-slot budget, and its
+    submit offset lands proportionally inside the first 25% of the K
+    ticks, so the replayed subset keeps the trace's heavy-tailed size
+    distribution and burstiness at any hardware speed.
 
     Returns (arrivals, fill_specs): arrivals is a list of
     (arrival_tick, JobSpec)."""
-    rng = random.Random(seed)
+    from gpuschedule_amd.trace.philly_log import load_philly_log
+    fixture = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "data", "philly_sample.json")
+    singles = [s for s in load_philly_log(fixture, time_scale=1 / 3600,
+                                          seed=seed) if s.n_gpus == 1]
+    philly = [singles[i % len(singles)] for i in range(n_jobs)]
     # STRONG scaling: the trace is IDENTICAL at every N, sized so the
     # total tick-slot demand fits one GPU's window (the smallest
     # configuration) — at larger N the same jobs spread across GPUs and
     # avg JCT + makespan genuinely shrink.
     budget = int(0.70 * steps) * iters_per_tick
     models = ["resnet50", "bert-base"]
-    # log-uniform weights, normalized to the budget
-    raw = [10 ** rng.uniform(0, 1.2) for _ in range(n_jobs)]
+    raw = [max(s.total_iters, 1) for s in philly]      # logged demand
+    subs = [s.submit_time for s in philly]
+    span = (max(subs) - min(subs)) or 1.0
     scale = budget / sum(raw)
     arrivals = []
-    for i, r in enumerate(raw):
-        iters = max(iters_per_tick, int(r * scale))
-        tick = int(i * (0.25 * steps) / max(1, n_jobs - 1))
+    for i, s in enumerate(philly):
+        iters = max(iters_per_tick, int(raw[i] * scale))
+        tick = min(int((subs[i] - min(subs)) / span * 0.25 * steps),
+                   max(int(0.25 * steps) - 1, 0))
         arrivals.append((tick, JobSpec(
             job_id=f"job{i:03d}", submit_time=0.0, n_gpus=1,
             total_iters=iters, model=models[i % 2])))
+    arrivals.sort(key=lambda a: a[0])
     # per-GPU fill jobs: best-effort, never complete inside the window.
     # At N>=2 the first fill is a 2-GPU DDP gang so the scale bench
     # exercises gang scheduling + bucketed RCCL all-reduce over xGMI

@@ -32,6 +32,13 @@ def _policy_from_args(args, cfg):
 
 
 def _specs_from_args(args, cfg):
+    if cfg.get("philly_log"):
+        from .trace.philly_log import load_philly_log
+        pl = cfg["philly_log"]
+        return load_philly_log(
+            pl["file"], time_scale=pl.get("time_scale", 1.0),
+            seed=pl.get("seed", 0), limit=pl.get("limit"),
+            clamp_gpus=pl.get("clamp_gpus", 8))
     if cfg.get("trace_file") or args.trace:
         return load_trace(cfg.get("trace_file") or args.trace)
     t = cfg.get("synthetic", {})

@@ -161,6 +161,8 @@ class EventLog:
             "max_slowdown": slowdowns[-1] if ns else 0.0,
             "n_jobs": float(len(submits)),
             "n_complete": float(n),
+            "n_failed": float(sum(1 for e in self.events
+                                  if e["event"] == "fail")),
             "avg_jct": sum(jcts) / n if n else 0.0,
             "p50_jct": pct(0.50),
             "p95_jct": pct(0.95),

@@ -22,7 +22,10 @@ def run_cli(*args):
 @pytest.mark.parametrize("cfg", CONFIGS)
 def test_simulate_all_shipped_configs(cfg):
     res = json.loads(run_cli("simulate", "--config", cfg).strip())
-    assert res["n_complete"] == res["n_jobs"] > 0
+    # every job either completes or (config 4: Philly Killed/Failed
+    # rows, replayed via fail_at_iter) fails mid-run by design
+    assert res["n_complete"] + res.get("n_failed", 0) == res["n_jobs"] > 0
+    assert res["n_complete"] > 0
 
 
 def test_gen_trace_and_simulate(tmp_path):

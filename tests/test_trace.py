@@ -47,3 +47,64 @@ def test_jobspec_validation():
     with pytest.raises(ValueError):
         JobSpec(job_id="x", submit_time=0, n_gpus=0, total_iters=1,
                 model="resnet18")
+
+
+# ---- real Philly cluster_job_log schema -----------------------------
+
+def test_philly_log_fixture_loads():
+    from gpuschedule_amd.trace.philly_log import load_philly_log
+    specs = load_philly_log("data/philly_sample.json",
+                            time_scale=1 / 3600, seed=4)
+    assert len(specs) == 64  # 3 never-started records skipped
+    assert all(1 <= s.n_gpus <= 8 for s in specs)  # 16/24-GPU clamped
+    assert specs == sorted(specs, key=lambda s: s.submit_time)
+    assert specs[0].submit_time == 0.0
+    assert any(s.fail_at_iter > 0 for s in specs)  # Killed/Failed map
+    for s in specs:
+        if s.fail_at_iter:
+            assert s.fail_at_iter < s.total_iters
+    # deterministic across loads
+    again = load_philly_log("data/philly_sample.json",
+                            time_scale=1 / 3600, seed=4)
+    assert [(s.job_id, s.model, s.total_iters) for s in specs] == \
+        [(s.job_id, s.model, s.total_iters) for s in again]
+
+
+def test_philly_log_schema_edge_cases(tmp_path):
+    import json
+    from gpuschedule_amd.trace.philly_log import load_philly_log
+    rows = [
+        # normal 2-server 10-GPU job, clamped to 8
+        {"status": "Pass", "jobid": "a", "vc": "v", "user": "u",
+         "submitted_time": "2017-10-03 05:00:00",
+         "attempts": [{"start_time": "2017-10-03 05:01:00",
+                       "end_time": "2017-10-03 06:01:00",
+                       "detail": [{"ip": "1", "gpus": ["g"] * 8},
+                                  {"ip": "2", "gpus": ["g"] * 2}]}]},
+        # killed mid-run with a retry attempt: durations sum
+        {"status": "Killed", "jobid": "b", "vc": "v", "user": "u",
+         "submitted_time": "2017-10-03 05:10:00",
+         "attempts": [{"start_time": "2017-10-03 05:11:00",
+                       "end_time": "2017-10-03 05:41:00",
+                       "detail": [{"ip": "1", "gpus": ["g"]}]},
+                      {"start_time": "2017-10-03 05:45:00",
+                       "end_time": "2017-10-03 06:15:00",
+                       "detail": [{"ip": "1", "gpus": ["g"]}]}]},
+        # never started -> skipped
+        {"status": "Killed", "jobid": "c", "vc": "v", "user": "u",
+         "submitted_time": "2017-10-03 05:20:00", "attempts": []},
+        # malformed timestamps -> skipped
+        {"status": "Pass", "jobid": "d", "vc": "v", "user": "u",
+         "submitted_time": "2017-10-03 05:30:00",
+         "attempts": [{"start_time": "n/a", "end_time": None,
+                       "detail": [{"ip": "1", "gpus": ["g"]}]}]},
+    ]
+    p = tmp_path / "log.json"
+    p.write_text(json.dumps(rows))
+    specs = load_philly_log(str(p), time_scale=1.0)
+    assert [s.job_id for s in specs] == ["a", "b"]
+    assert specs[0].n_gpus == 8
+    b = specs[1]
+    assert b.fail_at_iter > 0 and b.total_iters == b.fail_at_iter + 1
+    # b ran 60 min total across two attempts
+    assert b.submit_time == 600.0
