@@ -98,11 +98,26 @@ class TorchBackend:
 
     # ---- plan application ---------------------------------------
 
+    @staticmethod
+    def _bcast_resizes(plan: Plan) -> Dict[str, "Placement"]:
+        """Resizes whose old placement overlaps the new one: state
+        moves by RCCL broadcast from a surviving rank (no checkpoint
+        file) — see parallel/ddp.broadcast_state."""
+        out = {}
+        for p in plan.places:
+            if (p.resize_from_gpus and not p.from_evicted
+                    and set(p.resize_from_gpus) & set(p.gpus)):
+                out[p.job_id] = p
+        return out
+
     def _apply_phase_a(self, plan: Plan, meta: Dict[str, tuple]) -> None:
+        bcast = self._bcast_resizes(plan)
         for job_id, reason in plan.preempts:
             r = self.runners.get(job_id)
             if r is None:
                 continue
+            if reason == "resize" and job_id in bcast:
+                continue  # survivors keep state; joiners get broadcast
             if reason in ("migrate", "resize"):
                 if r.gang_rank == 0:
                     self.ckpt.save(job_id, r.model, r.optim, r.done_iters)
@@ -119,8 +134,8 @@ class TorchBackend:
         for p in plan.places:
             if p.migrate_from is None and p.resize_from is None:
                 continue
-            if p.job_id in preempted:
-                continue  # already checkpointed above
+            if p.job_id in preempted or p.job_id in bcast:
+                continue  # already checkpointed / broadcast path
             r = self.runners.get(p.job_id)
             if r is None:
                 continue
@@ -139,15 +154,42 @@ class TorchBackend:
                     torch.cuda.empty_cache()
 
     def _apply_phase_b(self, plan: Plan, meta: Dict[str, tuple]) -> None:
+        bcast = self._bcast_resizes(plan)
         for p in plan.places:
             if self.rank not in p.gpus:
-                # if we host a stale resident copy (migrated away), drop
+                # if we host a stale resident copy (migrated away or
+                # shrunk off this rank), drop it
                 r = self.runners.pop(p.job_id, None)
-                if r is not None:
+                if r is not None and not RUNNER_POOL.put(r):
                     r.release()
                 continue
             model_name, batch, gpus, done_iters = meta[p.job_id]
             runner = self.runners.get(p.job_id)
+            if p.job_id in bcast:
+                # overlap resize: swap gang in place on survivors,
+                # fresh-build (or adopt) on joiners, then collectively
+                # broadcast the full training state from a survivor.
+                from ..parallel.ddp import broadcast_state
+                group = self.gangs.get(gpus)
+                key = (model_name, batch, tuple(gpus))
+                old = p.resize_from_gpus
+                if runner is None:
+                    pooled = RUNNER_POOL.get(key)
+                    if pooled is not None:
+                        pooled.adopt(p.job_id)
+                        runner = pooled
+                    else:
+                        runner = self._build_runner(
+                            p.job_id, model_name, batch, group, gpus,
+                            0, None, key)
+                    self.runners[p.job_id] = runner
+                # every gang member (survivor, joiner, pool-adopted)
+                # rebinds to the SAME new group object
+                runner.regang(group, gpus.index(self.rank), len(gpus))
+                src = next(g for g in gpus if g in old)
+                broadcast_state(runner.model, runner.optim, group, src)
+                runner.done_iters = done_iters
+                continue
             if runner is not None:
                 continue  # resume-in-place: state resident in HBM
             state = None

@@ -201,6 +201,24 @@ class JobRunner:
         if crng is not None and torch.cuda.is_available():
             torch.cuda.set_rng_state(crng, self.device)
 
+    def regang(self, group, gang_rank: int, gang_world: int) -> None:
+        """Rebind a SURVIVING runner to its job's resized gang: swap
+        the collective group + reducer in place (model/optimizer state
+        tensors stay — that is the point: no checkpoint round-trip).
+        The hipGraph captured for the old gang width is dropped; width
+        1 recaptures after one eager iteration (allocator/MIOpen are
+        already warm)."""
+        self.reducer.detach()
+        self.gang_group = group
+        self.gang_rank = gang_rank
+        self.gang_world = gang_world
+        self.reducer = GradReducer(self.model, group)
+        self._graph = None
+        self._static = None
+        self.use_graph = (str(self.device) != "cpu" and gang_world == 1
+                          and not os.environ.get("GSAMD_NO_GRAPHS"))
+        self._graph_warmup = 1
+
     @torch.no_grad()
     def adopt(self, job_id: str, done_iters: int = 0) -> None:
         """Rebind this (pooled) runner to a new job: copy the family's

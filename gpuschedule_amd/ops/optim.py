@@ -55,6 +55,49 @@ class _MasterStateMixin:
         if isinstance(mt, tuple):   # AdamW: (..., stepbuf, params)
             mt[4].zero_()
 
+    @torch.no_grad()
+    def init_state(self):
+        """Materialize state for every trainable param (the step()
+        lazy-init requires grads).  Elastic joiners call this so the
+        state tensors exist as broadcast destinations."""
+        for group in self.param_groups:
+            for p in group["params"]:
+                if not p.requires_grad or self.state.get(p):
+                    continue
+                st = self.state[p]
+                if isinstance(self, FusedAdamW):
+                    st["step"] = 0
+                st["p32"] = p.detach().float().clone()
+                if isinstance(self, FusedAdamW):
+                    st["m"] = torch.zeros_like(st["p32"])
+                    st["v"] = torch.zeros_like(st["p32"])
+                else:
+                    st["buf"] = torch.zeros_like(st["p32"])
+
+    def true_step(self) -> int:
+        """Current optimizer step count, reading the device counter
+        when the multi-tensor path (possibly under graph replay) is
+        active."""
+        mt = getattr(self, "_mt", None)
+        if isinstance(mt, tuple):
+            return int(mt[4][0].item())
+        for group in self.param_groups:
+            for p in group["params"]:
+                st = self.state.get(p)
+                if st and "step" in st:
+                    return st["step"]
+        return 0
+
+    def set_step(self, step: int) -> None:
+        for group in self.param_groups:
+            for p in group["params"]:
+                st = self.state.get(p)
+                if st and "step" in st:
+                    st["step"] = step
+        mt = getattr(self, "_mt", None)
+        if isinstance(mt, tuple):
+            mt[4][0] = float(step)
+
 
 def _mt_eligible(p) -> bool:
     """Param qualifies for the flat multi-tensor kernels: any DENSE

@@ -99,9 +99,47 @@ class GradReducer:
 def broadcast_module(model: torch.nn.Module,
                      group: Optional[dist.ProcessGroup], src_rank: int):
     """Broadcast params+buffers from global rank ``src_rank`` (joiners
-    in an elastic resize sync from an existing replica)."""
+    in an elastic resize sync from an existing replica).  Iterates the
+    state_dict, NOT parameters()+buffers(): non-persistent buffers
+    (e.g. lazily-built RoPE tables) may exist on trained ranks but not
+    on joiners, which would desynchronize the collective sequence."""
     if group is None or dist.get_world_size(group) <= 1:
         return
     with torch.no_grad():
-        for t in list(model.parameters()) + list(model.buffers()):
-            dist.broadcast(t.data, src=src_rank, group=group)
+        for t in model.state_dict().values():
+            if isinstance(t, torch.Tensor) and t.numel():
+                dist.broadcast(t.data, src=src_rank, group=group)
+
+
+def broadcast_state(model: torch.nn.Module, optim,
+                    group: Optional[dist.ProcessGroup], src_rank: int):
+    """Move a job's full training state to elastic joiners over the
+    gang's collective backend (RCCL over xGMI on GPU): params+buffers,
+    optimizer state (fp32 masters/moments), and the step counter.  All
+    ranks of the new gang must call this collectively; joiners first
+    materialize empty state tensors as broadcast destinations
+    (optim.init_state), so the tensor sequence is identical on every
+    rank.  Replaces the /dev/shm checkpoint round-trip for pure-grow
+    (and overlap) resizes."""
+    if group is None or dist.get_world_size(group) <= 1:
+        return
+    broadcast_module(model, group, src_rank)
+    if optim is None:
+        return
+    with torch.no_grad():
+        optim.init_state()
+        for g in optim.param_groups:
+            for p in g["params"]:
+                st = optim.state.get(p)
+                if not st:
+                    continue
+                for k in sorted(st.keys()):
+                    v = st[k]
+                    if isinstance(v, torch.Tensor):
+                        dist.broadcast(v.data, src=src_rank, group=group)
+        if hasattr(optim, "true_step"):
+            dev = next(model.parameters()).device
+            step = torch.tensor([optim.true_step()], dtype=torch.long,
+                                device=dev)
+            dist.broadcast(step, src=src_rank, group=group)
+            optim.set_step(int(step.item()))

@@ -34,6 +34,10 @@ class Placement:
     from_evicted: bool = False            # state must come from host ckpt
     migrate_from: Optional[List[int]] = None
     resize_from: Optional[int] = None
+    # old placement of a resize; when it overlaps the new gpus, the
+    # backend moves state by RCCL broadcast from a surviving rank
+    # instead of a host-checkpoint round-trip
+    resize_from_gpus: Optional[List[int]] = None
 
 
 @dataclass
@@ -338,6 +342,7 @@ class Scheduler:
             if had_state and job.placement and gpus != job.placement:
                 if len(gpus) != len(job.placement):
                     p.resize_from = len(job.placement)
+                    p.resize_from_gpus = list(job.placement)
                 else:
                     p.migrate_from = list(job.placement)
             plan.places.append(p)

@@ -27,6 +27,7 @@ COLD_START_S = 0.5
 RESUME_HBM_S = 0.05
 D2H_BYTES_PER_S = 48e9   # pinned hipMemcpyAsync, side stream
 H2D_BYTES_PER_S = 48e9
+XGMI_BYTES_PER_S = 150e9  # per-link broadcast (overlap-resize path)
 
 
 class Simulator:
@@ -45,6 +46,11 @@ class Simulator:
         mem = self.sched.cluster.job_mem(p.job_id) or \
             catalog.job_mem_bytes(job.spec)
         ckpt = mem / 3  # checkpoint = weights+optimizer, not activations
+        if (p.resize_from_gpus and not p.from_evicted
+                and set(p.resize_from_gpus) & set(p.gpus)):
+            # overlap resize: state broadcast over xGMI from a
+            # surviving rank (exec/backend.py), no host round-trip
+            return ckpt / XGMI_BYTES_PER_S
         if p.migrate_from is not None or p.resize_from is not None:
             return ckpt / D2H_BYTES_PER_S + ckpt / H2D_BYTES_PER_S
         if p.from_evicted:

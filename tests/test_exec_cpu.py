@@ -229,14 +229,74 @@ def _worker(rank, world, port, scenario, out_file):
             pol = "elastic"
         ck = f"/tmp/gpuschedule_test_ck_{port}"
         be = make_backend(specs, pol, rank=rank, world=world, ckpt_root=ck)
+        saves = []
+        orig_save = be.ckpt.save
+        be.ckpt.save = lambda jid, *a, **kw: (saves.append(jid),
+                                              orig_save(jid, *a, **kw))[1]
         log = be.run(max_seconds=300)
         if rank == 0:
             import json
             with open(out_file, "w") as f:
                 json.dump({"summary": log.summary(),
-                           "events": log.events}, f)
+                           "events": log.events, "saves": saves}, f)
     finally:
         dist.destroy_process_group()
+
+
+def _bcast_worker(rank, world, port, out_file):
+    import torch.distributed as dist
+    from gpuschedule_amd.parallel.ddp import broadcast_state
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        r = JobRunner("el", "llama3-tiny", 2, "cpu", torch.float32,
+                      None, 0, 1)
+        if rank == 0:  # the pre-grow survivor trained solo
+            r.step(3)
+        group = dist.group.WORLD
+        broadcast_state(r.model, r.optim, group, src_rank=0)
+        if rank == 1:
+            # joiner must now hold the survivor's exact state
+            sums = sorted((n, float(p.double().sum()))
+                          for n, p in r.model.named_parameters())
+            st = [float(v.double().sum())
+                  for g in r.optim.param_groups for p in g["params"]
+                  for k, v in sorted(r.optim.state[p].items())
+                  if isinstance(v, torch.Tensor)]
+            steps = [r.optim.state[p].get("step")
+                     for g in r.optim.param_groups for p in g["params"]]
+            import json
+            with open(out_file + ".r1", "w") as f:
+                json.dump({"sums": sums, "st": st, "steps": steps}, f)
+        else:
+            sums = sorted((n, float(p.double().sum()))
+                          for n, p in r.model.named_parameters())
+            st = [float(v.double().sum())
+                  for g in r.optim.param_groups for p in g["params"]
+                  for k, v in sorted(r.optim.state[p].items())
+                  if isinstance(v, torch.Tensor)]
+            steps = [r.optim.state[p].get("step")
+                     for g in r.optim.param_groups for p in g["params"]]
+            import json
+            with open(out_file + ".r0", "w") as f:
+                json.dump({"sums": sums, "st": st, "steps": steps}, f)
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_broadcast_state_syncs_joiner(tmp_path):
+    import torch.multiprocessing as mp
+    import json
+    out = str(tmp_path / "b")
+    mp.spawn(_bcast_worker, args=(2, 29320, out), nprocs=2, join=True)
+    with open(out + ".r0") as f:
+        a = json.load(f)
+    with open(out + ".r1") as f:
+        b = json.load(f)
+    assert a == b, "joiner state differs from survivor after broadcast"
+    assert a["steps"][0] == 3
 
 
 @pytest.mark.parametrize("scenario", ["gang", "mix", "elastic"])
@@ -255,3 +315,7 @@ def test_world2_scenarios(tmp_path, scenario):
         assert any(e["event"] == "resize" for e in res["events"]) or \
             any(e["event"] == "start" and len(e.get("gpus", [])) == 2
                 for e in res["events"])
+        # grow resizes move state by RCCL broadcast from a surviving
+        # rank — the /dev/shm checkpoint round-trip must NOT happen
+        if any(e["event"] == "resize" for e in res["events"]):
+            assert "el" not in res["saves"], res["saves"]
