@@ -14,10 +14,12 @@ def J(jid, t=0.0, n=1, iters=4, model="llama3-tiny", batch=2, **kw):
                    model=model, batch_size=batch, **kw)
 
 
-def make_backend(specs, policy_name="fifo", rank=0, world=1, **kw):
+def make_backend(specs, policy_name="fifo", rank=0, world=1,
+                 policy_kw=None, **kw):
     sched = None
     if rank == 0:
-        sched = Scheduler(Cluster(world, world), make_policy(policy_name))
+        sched = Scheduler(Cluster(world, world),
+                          make_policy(policy_name, **(policy_kw or {})))
         sched.submit(specs)
     return TorchBackend(sched, rank=rank, world=world, device_type="cpu",
                         iters_per_tick=2, **kw)
@@ -210,9 +212,11 @@ def test_packing_never_overcommits_memory():
 
 def _worker(rank, world, port, scenario, out_file):
     import torch.distributed as dist
+    torch.set_num_threads(max(1, (os.cpu_count() or 4) // world))
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
+    pol_kw = None
     try:
         if scenario == "gang":
             specs = [J("gang", n=2, iters=4)]
@@ -221,6 +225,33 @@ def _worker(rank, world, port, scenario, out_file):
             specs = [J("g", n=2, iters=3), J("s1", t=0.2, iters=2),
                      J("s2", t=0.2, iters=2)]
             pol = "srtf"
+        elif scenario == "las4":
+            # LAS gangs on 4 ranks: a 4-gang, a 2-gang and singles
+            # contend; LAS preempts long-served jobs for arrivals
+            specs = [J("g4", n=4, iters=6), J("g2", t=0.1, n=2, iters=4),
+                     J("s1", t=0.1, iters=3), J("s2", t=0.2, iters=3)]
+            pol = "las"
+            pol_kw = {"thresholds": (0.05, 0.4), "starve_limit": 2.0}
+        elif scenario == "gandiva4":
+            # quantum rotation + migration across 4 ranks; the 2-gang
+            # forces defrag when singles fragment the window
+            specs = ([J(f"s{i}", t=0.05 * i, iters=4) for i in range(5)]
+                     + [J("g2", t=0.3, n=2, iters=4)])
+            pol = "gandiva"
+            pol_kw = {"quantum": 0.2}
+        elif scenario == "elastic4":
+            specs = [JobSpec(job_id="el", submit_time=0.0, n_gpus=1,
+                             total_iters=10, model="llama3-tiny",
+                             batch_size=2, elastic=True, min_gpus=1,
+                             max_gpus=4)]
+            pol = "elastic"
+        elif scenario == "elastic8":
+            # BASELINE config-5 shape: one elastic job growing 1->8
+            specs = [JobSpec(job_id="el", submit_time=0.0, n_gpus=1,
+                             total_iters=12, model="llama3-tiny",
+                             batch_size=2, elastic=True, min_gpus=1,
+                             max_gpus=8)]
+            pol = "elastic"
         else:  # elastic grow
             specs = [JobSpec(job_id="el", submit_time=0.0, n_gpus=1,
                              total_iters=6, model="llama3-tiny",
@@ -228,7 +259,8 @@ def _worker(rank, world, port, scenario, out_file):
                              max_gpus=2)]
             pol = "elastic"
         ck = f"/tmp/gpuschedule_test_ck_{port}"
-        be = make_backend(specs, pol, rank=rank, world=world, ckpt_root=ck)
+        be = make_backend(specs, pol, rank=rank, world=world,
+                          policy_kw=pol_kw, ckpt_root=ck)
         saves = []
         orig_save = be.ckpt.save
         be.ckpt.save = lambda jid, *a, **kw: (saves.append(jid),
@@ -299,23 +331,38 @@ def test_broadcast_state_syncs_joiner(tmp_path):
     assert a["steps"][0] == 3
 
 
-@pytest.mark.parametrize("scenario", ["gang", "mix", "elastic"])
-def test_world2_scenarios(tmp_path, scenario):
+_SCENARIO_WORLD = {"gang": 2, "mix": 2, "elastic": 2,
+                   "las4": 4, "gandiva4": 4, "elastic4": 4,
+                   "elastic8": 8}
+
+
+@pytest.mark.parametrize("scenario", list(_SCENARIO_WORLD))
+def test_multiproc_scenarios(tmp_path, scenario):
     import torch.multiprocessing as mp
     import json
-    port = 29310 + {"gang": 0, "mix": 1, "elastic": 2}[scenario]
+    world = _SCENARIO_WORLD[scenario]
+    port = 29310 + list(_SCENARIO_WORLD).index(scenario)
     out = str(tmp_path / "out.json")
-    mp.spawn(_worker, args=(2, port, scenario, out), nprocs=2,
+    mp.spawn(_worker, args=(world, port, scenario, out), nprocs=world,
              join=True)
     with open(out) as f:
         res = json.load(f)
     s = res["summary"]
     assert s["n_complete"] == s["n_jobs"], res["events"]
-    if scenario == "elastic":
-        assert any(e["event"] == "resize" for e in res["events"]) or \
-            any(e["event"] == "start" and len(e.get("gpus", [])) == 2
-                for e in res["events"])
+    events = res["events"]
+    if scenario.startswith("elastic"):
+        grew = [e for e in events if e["event"] == "resize"
+                and e["new_gpus"] > e["old_gpus"]]
+        widest = max((len(e.get("gpus", [])) for e in events
+                      if e["event"] in ("start", "resume")), default=0)
+        assert grew or widest == world, events
         # grow resizes move state by RCCL broadcast from a surviving
         # rank — the /dev/shm checkpoint round-trip must NOT happen
-        if any(e["event"] == "resize" for e in res["events"]):
+        if grew:
             assert "el" not in res["saves"], res["saves"]
+        if scenario == "elastic8":
+            assert widest == 8, f"never reached width 8: {events}"
+    if scenario == "las4":
+        assert s["preemptions"] >= 1, events
+    if scenario == "gandiva4":
+        assert s["preemptions"] >= 1, events
