@@ -154,10 +154,16 @@ class JobRunner:
         self.last_loss = float("nan")
         # hipGraph capture of the whole training step (fwd+bwd+fused
         # optimizer): per-job static shapes make the step replayable;
-        # launch-bound models (small CNNs) gain the most.  Single-GPU
-        # jobs only (gang collectives stay out of capture).
-        self.use_graph = (str(device) != "cpu" and gang_world == 1
-                          and not os.environ.get("GSAMD_NO_GRAPHS"))
+        # launch-bound models (small CNNs) gain the most.  Gang steps
+        # (RCCL collectives are hipGraph-capturable on ROCm) are gated
+        # behind GSAMD_GANG_GRAPHS=1: the path cannot be validated on a
+        # 1-GPU lease — RCCL refuses same-device 2-rank communicators
+        # ("Duplicate GPU detected", scripts/rccl_1gpu_probe.py) — so
+        # it stays off by default until a multi-GPU run proves it.
+        self.use_graph = (str(device) != "cpu"
+                          and not os.environ.get("GSAMD_NO_GRAPHS")
+                          and (gang_world == 1
+                               or bool(os.environ.get("GSAMD_GANG_GRAPHS"))))
         self._graph = None
         self._graph_warmup = 3  # eager iters before capture (lazy opt
         #                         state init + allocator warmup)
@@ -215,8 +221,10 @@ class JobRunner:
         self.reducer = GradReducer(self.model, group)
         self._graph = None
         self._static = None
-        self.use_graph = (str(self.device) != "cpu" and gang_world == 1
-                          and not os.environ.get("GSAMD_NO_GRAPHS"))
+        self.use_graph = (str(self.device) != "cpu"
+                          and not os.environ.get("GSAMD_NO_GRAPHS")
+                          and (gang_world == 1
+                               or bool(os.environ.get("GSAMD_GANG_GRAPHS"))))
         self._graph_warmup = 1
 
     @torch.no_grad()

@@ -31,9 +31,20 @@ constexpr int BM = 128, BN = 128, BK = 64;
 constexpr int NTHREADS = 256;
 
 // LDS row stride in bytes (BK bf16) and the bank swizzle.
+//
+// g-swizzle (conflict-free for the fragment ds_read_b128 pattern):
+// slot' = slot ^ g(row), g(row) = 2*bit1(row) + 4*bit3(row).  A b128
+// is serviced in four fixed 16-lane groups that pair {f0-3,f12-15} of
+// one k-slot with {f4-11} of the next; with this g every group's 16
+// addresses land on the 16 bank-16-units exactly once (checked against
+// the gfx950 lane-group table).  The previous (row&7)<<4 XOR left
+// 2-way conflicts: rows r and r+8 share both bank parity and XOR.
+__device__ __forceinline__ int gsw(int row) {
+  return ((row >> 1) & 1) * 2 + ((row >> 3) & 1) * 4;
+}
 constexpr int ROWB = BK * 2;  // 128 B
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
-  return row * ROWB + (byte_in_row ^ ((row & 7) << 4));
+  return row * ROWB + (byte_in_row ^ (gsw(row) << 4));
 }
 
 // Tile staging via global_load_lds (direct HBM->LDS DMA, 16 B/lane —
@@ -50,10 +61,12 @@ __device__ __forceinline__ void glds_stage_tile(
     const short* __restrict__ src, long ld, int row0, int rows_max,
     int k0, char* lds_buf, int wid, int lane) {
   const int rl = lane >> 3;                       // row within stripe
-  const int gcolb = ((lane & 7) * 16) ^ (rl << 4);  // swizzle-inv col
 #pragma unroll
   for (int p = 0; p < 4; ++p) {
     int rp = p * 32 + wid * 8;
+    // g-swizzle inverse on the SOURCE column (LDS row = rp + rl)
+    int gcolb = (((lane & 7) ^
+                  (((rl >> 1) & 1) * 2 + ((rp >> 3) & 1) * 4)) << 4);
     int gr = row0 + rp + rl;
     if (gr >= rows_max) gr = rows_max - 1;
     const void* g = src + (long)gr * ld + k0 + gcolb / 2;
@@ -327,10 +340,11 @@ __device__ __forceinline__ void glds_stage_half(
     const short* __restrict__ src, long ld, int row0, int rows_max,
     int k0, char* lds_img, int half, int wid, int lane) {
   const int rl = lane >> 3;
-  const int gcolb = ((lane & 7) * 16) ^ (rl << 4);
 #pragma unroll
   for (int p = 0; p < 2; ++p) {
     int rp = half * 128 + p * 64 + wid * 8;
+    int gcolb = (((lane & 7) ^
+                  (((rl >> 1) & 1) * 2 + ((rp >> 3) & 1) * 4)) << 4);
     int gr = row0 + rp + rl;
     if (gr >= rows_max) gr = rows_max - 1;
     const void* g = src + (long)gr * ld + k0 + gcolb / 2;
@@ -509,7 +523,9 @@ __device__ __forceinline__ void glds_cs_piece(
     const short* __restrict__ src, long ld, int row0, int rows_max,
     int k0, char* lds_img, int rp_base, int lane) {
   const int rl = lane >> 3;
-  const int gcolb = ((lane & 7) * 16) ^ (rl << 4);
+  const int gcolb = (((lane & 7) ^
+                      (((rl >> 1) & 1) * 2 +
+                       ((rp_base >> 3) & 1) * 4)) << 4);
   int gr = row0 + rp_base + rl;
   if (gr >= rows_max) gr = rows_max - 1;
   const void* g = src + (long)gr * ld + k0 + gcolb / 2;
@@ -623,6 +639,414 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_cs_kernel(
   const bool mtail = bm0 + BM2 > M;
   const bool ntail = bn0 + BN2 > N;
   __syncthreads();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
+// ==== 256x256 / BK=64 consumption-pipelined GEMM (p8) ===============
+// Two LDS tile buffers, BK=64, 8 waves (2Mx4N).  Every wave stages,
+// per tile, EXACTLY the four 16-row pieces that it (and the waves
+// sharing those rows) consume, issued in consumption order during the
+// PREVIOUS tile:   ph0: A-jr0   ph1: B-jc0   ph2: B-jc1   ph3: A-jr1.
+// Consumption (quadrant phases (jr,jc) = 00,01,10,11):
+//   ph0 reads A-jr0 + B-jc0 (the FIFO's oldest 4 instructions),
+//   ph1 reads B-jc1, ph2 reads A-jr1, ph3 reads NOTHING (af held from
+//   ph2, bf1 held from ph1).
+// So every phase's wait is a counted s_waitcnt vmcnt(4) — the DMA
+// queue NEVER drains in the main loop (the round-1 kernel's
+// __syncthreads drained vmcnt(0) every tile; MfmaUtil 40%).  The wait
+// sits BEFORE the phase barrier, so after the barrier every wave's
+// needed pieces — including ones staged by its row/column partners,
+// who run the identical wait schedule — are in LDS.  One barrier per
+// phase is also what makes staging race-free: stage targets buf(t+1),
+// whose last reader (tile t-1, which reads nothing at its ph3) is
+// provably past its reads once t.ph0's barrier passed.
+__device__ __forceinline__ void glds_piece16(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int k0, char* lds_img, int rp, int lane) {
+  const int rl = lane >> 3;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int rpp = rp + p * 8;
+    int gcolb = (((lane & 7) ^
+                  (((rl >> 1) & 1) * 2 + ((rpp >> 3) & 1) * 4)) << 4);
+    int gr = row0 + rpp + rl;
+    if (gr >= rows_max) gr = rows_max - 1;
+    const void* g = src + (long)gr * ld + k0 + gcolb / 2;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (lds_img + rpp * ROWB),
+        16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  // this wave's staged-piece row bases (16 rows each)
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_rp = (wid & 3) * 64 + (wid >> 2) * 16;   // + jc*32
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK2;
+  // prologue: tile 0's pieces in FIFO order (ph0's vmcnt(4) + barrier
+  // is the only sync the first tile needs)
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp + 32, lane);
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp + 64, lane);
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
+    char* Bct = Act + BM2 * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM2 * ROWB);
+    char* Bn = An + BM2 * ROWB;
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK2;
+    bf16x8 af[4][2], bf0[2][2], bf1[2][2];
+
+    // ---- ph0 (jr0, jc0): needs the FIFO's oldest 4 ----------------
+    __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf0[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
+    }
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph1 (jr0, jc1): needs B-jc1 ------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf1[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + 32 + fj * 16 + frow, kk * 64 + kb));
+    if (pre) glds_piece16(B, K, bn0, N, k1, Bn, b_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph2 (jr1, jc0): needs A-jr1 ------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
+    if (pre) glds_piece16(B, K, bn0, N, k1, Bn, b_rp + 32, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[4 + fi][fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph3 (jr1, jc1): everything already in registers ----------
+    __builtin_amdgcn_s_barrier();
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 64, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[4 + fi][2 + fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);   // drain before LDS reuse
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
+// ==== 256x256 / BK=32 three-buffer pipelined GEMM (v2) ==============
+// The round-1 256-tile kernel drains vmcnt(0) at EVERY K-tile boundary
+// (its __syncthreads carries the wait while LDS-DMA is in flight) and
+// runs one barrier per phase; MfmaUtil measured 40%.  This kernel is
+// built to the verified 8-phase-template structure instead:
+//  * THREE LDS tile buffers (BK=32 so 3x(A+B) = 96 KiB fits): during
+//    tile t the wave stages tile t+2, so at the t->t+1 boundary the
+//    counted wait vmcnt(4) proves tile t+1 landed while t+2's four
+//    DMAs stay in flight.  No vmcnt(0) anywhere in the main loop.
+//  * TWO raw barriers per phase isolating the MFMA cluster (reads +
+//    stage issue | barrier; lgkmcnt(0); setprio(1); 16 MFMA;
+//    setprio(0) | barrier): on each SIMD one wave's MFMA segment
+//    overlaps its partner's load segment.
+//  * Conflict-free LDS k-slot permutation: LDS(row, slot') holds
+//    k-chunk c at slot' = c ^ h(q), q = (row>>2)&3, h = [0,2,3,1].
+//    ds_read_b128 services a wave in four fixed 16-lane groups that
+//    MIX k-slots ({f0-3,f12-15}@s + {f4-11}@s+1); with this h every
+//    group's 16 addresses hit the 16 bank-16-units exactly once
+//    (checked against the gfx950 b128 lane-group table) — the plain
+//    XOR swizzles leave 2-4-way conflicts on this pattern.  The
+//    permutation commutes with the lane-linear glds write (source
+//    column of lane l = ((l&3) ^ h((l>>4)&3))*16B) and each 4-lane row
+//    group still covers one contiguous 64-B global segment.
+constexpr int BK3 = 32;
+constexpr int ROW3 = BK3 * 2;          // 64 B per LDS row
+constexpr int TILE3 = 256 * ROW3;      // one operand tile = 16 KiB
+
+__device__ __forceinline__ int hq3(int q) {       // [0,2,3,1]
+  return (0x1320 >> (q * 4)) & 3;
+}
+__device__ __forceinline__ int rot3(int row, int slot) {
+  return row * ROW3 + ((slot ^ hq3((row >> 2) & 3)) << 4);
+}
+
+// Stage one operand K-tile (256 rows x 32 bf16) into lds_img: 2 glds
+// per wave (16 rows each x 8 waves = 256 rows per pass... 2 passes).
+// Lane l of the stage covers (row r0 + (l>>2), slot l&3); the global
+// column is the rotation-inverse so the LDS image equals rot3().
+__device__ __forceinline__ void glds_stage32(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int k0, char* lds_img, int wid, int lane) {
+  const int rl = lane >> 2;                      // row within 16-group
+  const int gcolb = (((lane & 3) ^ hq3((lane >> 4) & 3)) << 4);
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int rp = p * 128 + wid * 16;
+    int gr = row0 + rp + rl;
+    if (gr >= rows_max) gr = rows_max - 1;
+    const void* g = src + (long)gr * ld + k0 + gcolb / 2;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (lds_img + rp * ROW3),
+        16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_v2_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  // 3 x (A,B) staging buffers (96 KiB) inside the 128 KiB block that
+  // the epilogue reuses as the C bounce image.  ONE __shared__ object
+  // only (a second one makes hipcc emit vmcnt(0) before every ds_read
+  // of a glds pipeline).
+  __shared__ __attribute__((aligned(16))) char lds[2 * BM2 * BN2];
+  // buffer b occupies [b*2*TILE3, b*2*TILE3 + 2*TILE3): A then B
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;   // 2(M) x 4(N) wave grid
+  const int wc = (wid & 3) * 64;
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK3;
+  // prologue: stage tiles 0 and 1; prove tile 0 landed (4 younger
+  // DMAs stay in flight), then one full barrier.
+  glds_stage32(A, K, bm0, M, 0, lds, wid, lane);
+  glds_stage32(B, K, bn0, N, 0, lds + TILE3, wid, lane);
+  if (nk > 1) {
+    glds_stage32(A, K, bm0, M, BK3, lds + 2 * TILE3, wid, lane);
+    glds_stage32(B, K, bn0, N, BK3, lds + 3 * TILE3, wid, lane);
+  }
+  // vmcnt(4): tile 0 landed, tile 1's four DMAs stay in flight
+  if (nk > 1) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+  else __builtin_amdgcn_s_waitcnt(0xf70);
+  __builtin_amdgcn_s_barrier();
+
+  const int frow = lane & 15;        // fragment row within 16
+  const int slot = lane >> 4;        // k slot (16 B)
+
+  for (int t = 0; t < nk; ++t) {
+    const char* Act = lds + (t % 3) * (2 * TILE3);
+    const char* Bct = Act + TILE3;
+    char* nxt = lds + ((t + 2) % 3) * (2 * TILE3);
+    const bool pre = t + 2 < nk;
+    const int k2 = (t + 2) * BK3;
+    bf16x8 bfr[4];
+#pragma unroll
+    for (int ph = 0; ph < 2; ++ph) {   // jr half of the wave's rows
+      bf16x8 af[4];
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        int r = wr + ph * 64 + fi * 16 + frow;
+        af[fi] = *reinterpret_cast<const bf16x8*>(Act + rot3(r, slot));
+      }
+      if (ph == 0) {
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj) {
+          int r = wc + fj * 16 + frow;
+          bfr[fj] = *reinterpret_cast<const bf16x8*>(Bct + rot3(r, slot));
+        }
+        if (pre)
+          glds_stage32(A, K, bm0, M, k2, nxt, wid, lane);
+      } else if (pre) {
+        glds_stage32(B, K, bn0, N, k2, nxt + TILE3, wid, lane);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj)
+          acc[ph * 4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi], bfr[fj], acc[ph * 4 + fi][fj], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (ph == 1 && t + 1 < nk) {
+        // tile boundary: prove tile t+1 landed; t+2's 4 DMAs (just
+        // issued this tile) stay in flight across the barrier.
+        if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+        else __builtin_amdgcn_s_waitcnt(0xf70);
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  // epilogue: LDS bounce (identical to gemm_nt256_kernel)
+  __builtin_amdgcn_s_waitcnt(0xf70);   // drain any stray DMA
+  __builtin_amdgcn_s_barrier();
   short* c_lds = (short*)lds;  // [BM2][BN2]
   {
     const int lrow0 = wr + (lane >> 4) * 4;
@@ -837,6 +1261,29 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
   long blocks256 = (long)((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
   if (M >= 256 && N >= 256 && blocks256 >= 200) {
     int grid = (int)blocks256;
+    // variant select: 0 = round-1 kernel (vmcnt(0) drain per tile),
+    // 1 = v2 (BK=32, 3 buffers), 2 = p8 (BK=64, consumption-staged
+    // counted-vmcnt pipeline) — the measured default.
+    static const int var = [] {
+      const char* e = getenv("GSAMD_GEMM_VAR");
+      return e ? atoi(e) : 2;
+    }();
+    if (var == 2) {
+      hipLaunchKernelGGL(gemm_nt256_p8_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
+    if (var == 1 && K % BK3 == 0) {
+      hipLaunchKernelGGL(gemm_nt256_v2_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
     // consumption-scheduled staging variant (counted vmcnt waits, no
     // boundary drain) — A/B gate while it is being qualified
     static const bool use_cs = [] {
