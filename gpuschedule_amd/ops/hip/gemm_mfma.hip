@@ -724,9 +724,26 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
     int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
     wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
   }
+  // GROUP_M rasterization: within a band of GM m-blocks, consecutive
+  // ids walk the M dimension first, so the ~8 consecutive blocks an
+  // XCD chunk runs back-to-back share ONE B column panel (4 MiB = the
+  // XCD's L2) instead of each streaming a different one; B's chip-wide
+  // re-read factor drops from nbm to nbm/GM.  A panels of the band
+  // (GM x 4 MiB) cycle through L3.  (gm, not GM, in the tail band.)
   const int nbn = (N + BN2 - 1) / BN2;
-  const int bm0 = (wg / nbn) * BM2;
-  const int bn0 = (wg % nbn) * BN2;
+  const int nbm = (M + BM2 - 1) / BM2;
+  int bm, bn;
+  {
+    const int GM = 8;
+    int per_band = GM * nbn;
+    int band = wg / per_band;
+    int rem = wg - band * per_band;
+    int gm = min(GM, nbm - band * GM);
+    bn = rem / gm;
+    bm = band * GM + rem - bn * gm;
+  }
+  const int bm0 = bm * BM2;
+  const int bn0 = bn * BN2;
 
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
 
@@ -889,6 +906,556 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
     }
   }
 }
+
+// ==== p9: p8 with region-recycled deep staging ======================
+// Same consumption-ordered pieces as p8, but every piece of tile u is
+// issued ~4 phases (one full tile) before its read instead of 1-3:
+// each buffer REGION is recycled immediately after its single read
+// (A-jr0 of buf t is only read at t.ph0, so t+2's A-jr0 can be staged
+// into it at t.ph1 — the phase barrier proves all waves are past the
+// read).  Every steady-state wait becomes vmcnt(10) — five pieces
+// younger than the needed one stay in flight — so the waits
+// essentially never block.
+//   stage(u).A-jr0 @ (u-2).ph1     stage(u).B-jc0 @ (u-2).ph2
+//   stage(u).B-jc1 @ (u-2).ph3     stage(u).A-jr1 @ (u-1).ph0
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_p9_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  // this wave's staged-piece row bases (16 rows each)
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_rp = (wid & 3) * 64 + (wid >> 2) * 16;   // + jc*32
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK2;
+  // prologue: tile 0's pieces in FIFO order (ph0's vmcnt(4) + barrier
+  // is the only sync the first tile needs)
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp + 32, lane);
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp + 64, lane);
+  if (nk > 1) {
+    char* A1 = lds + 2 * BM2 * ROWB;
+    glds_piece16(A, K, bm0, M, BK2, A1, a_rp, lane);
+    glds_piece16(B, K, bn0, N, BK2, A1 + BM2 * ROWB, b_rp, lane);
+    glds_piece16(B, K, bn0, N, BK2, A1 + BM2 * ROWB, b_rp + 32, lane);
+  }
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
+    char* Bct = Act + BM2 * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM2 * ROWB);
+    const bool pre = t + 1 < nk;
+    const bool pre2 = t + 2 < nk;
+    const int k1 = (t + 1) * BK2;
+    const int k2 = (t + 2) * BK2;
+    bf16x8 af[4][2], bf0[2][2], bf1[2][2];
+
+    // ---- ph0 (jr0, jc0): needs the FIFO's oldest 4 ----------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 10);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf0[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
+    }
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 64, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph1 (jr0, jc1): needs B-jc1 ------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 10);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf1[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + 32 + fj * 16 + frow, kk * 64 + kb));
+    if (pre2) glds_piece16(A, K, bm0, M, k2, Act, a_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph2 (jr1, jc0): needs A-jr1 ------------------------------
+    if (pre2) __builtin_amdgcn_s_waitcnt(0xf70 | 10);
+    else if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 8);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
+    if (pre2) glds_piece16(B, K, bn0, N, k2, Bct, b_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[4 + fi][fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph3 (jr1, jc1): everything already in registers ----------
+    __builtin_amdgcn_s_barrier();
+    if (pre2) glds_piece16(B, K, bn0, N, k2, Bct, b_rp + 32, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[4 + fi][2 + fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);   // drain before LDS reuse
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
+
+// ==== p11: p9 merged to TWO megaphases per tile (one vmcnt wait +
+// two barriers per tile; 32-MFMA clusters) =========================
+void gemm_nt256_p11_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  // this wave's staged-piece row bases (16 rows each)
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_rp = (wid & 3) * 64 + (wid >> 2) * 16;   // + jc*32
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK2;
+  // prologue: tile 0's pieces in FIFO order (ph0's vmcnt(4) + barrier
+  // is the only sync the first tile needs)
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp + 32, lane);
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp + 64, lane);
+  if (nk > 1) {
+    char* A1 = lds + 2 * BM2 * ROWB;
+    glds_piece16(A, K, bm0, M, BK2, A1, a_rp, lane);
+    glds_piece16(B, K, bn0, N, BK2, A1 + BM2 * ROWB, b_rp, lane);
+    glds_piece16(B, K, bn0, N, BK2, A1 + BM2 * ROWB, b_rp + 32, lane);
+  }
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
+    char* Bct = Act + BM2 * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM2 * ROWB);
+    const bool pre = t + 1 < nk;
+    const bool pre2 = t + 2 < nk;
+    const int k1 = (t + 1) * BK2;
+    const int k2 = (t + 2) * BK2;
+    bf16x8 af[4][2], bf0[2][2], bf1[2][2];
+
+    // ---- megaphase 0 (jr0 x both jc): one deep wait proves ALL of
+    // tile t landed (each piece is >= 3 megaphases old); 16 fragment
+    // reads; 32 MFMAs in one cluster.
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 6);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj) {
+        bf0[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
+        bf1[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + 32 + fj * 16 + frow, kk * 64 + kb));
+      }
+    }
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 64, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
+          acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
+        }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- megaphase 1 (jr1 x both jc): every region of tile t's
+    // buffers has been read, so all three of tile t+2's early pieces
+    // stage here (into the just-freed regions); af reloads in place.
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
+    if (pre2) {
+      glds_piece16(A, K, bm0, M, k2, Act, a_rp, lane);
+      glds_piece16(B, K, bn0, N, k2, Bct, b_rp, lane);
+      glds_piece16(B, K, bn0, N, k2, Bct, b_rp + 32, lane);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[4 + fi][fj], 0, 0, 0);
+          acc[4 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[4 + fi][2 + fj], 0, 0, 0);
+        }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);   // drain before LDS reuse
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
+
+
+// ==== p10: p9 + fragment reads pipelined into the previous
+// phase's MFMA cluster (single deep wait per tile) ================
+void gemm_nt256_p10_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int bm0 = (wg / nbn) * BM2;
+  const int bn0 = (wg % nbn) * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  // this wave's staged-piece row bases (16 rows each)
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_rp = (wid & 3) * 64 + (wid >> 2) * 16;   // + jc*32
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK2;
+  // prologue: tile 0's pieces in FIFO order (ph0's vmcnt(4) + barrier
+  // is the only sync the first tile needs)
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp + 32, lane);
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp + 64, lane);
+  if (nk > 1) {
+    char* A1 = lds + 2 * BM2 * ROWB;
+    glds_piece16(A, K, bm0, M, BK2, A1, a_rp, lane);
+    glds_piece16(B, K, bn0, N, BK2, A1 + BM2 * ROWB, b_rp, lane);
+    glds_piece16(B, K, bn0, N, BK2, A1 + BM2 * ROWB, b_rp + 32, lane);
+  }
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
+    char* Bct = Act + BM2 * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM2 * ROWB);
+    const bool pre = t + 1 < nk;
+    const bool pre2 = t + 2 < nk;
+    const int k1 = (t + 1) * BK2;
+    const int k2 = (t + 2) * BK2;
+    bf16x8 af[4][2], bf0[2][2], bf1[2][2];
+
+    // ---- ph0: ONE deep wait proves ALL of tile t landed (pieces are
+    // a full tile old); later phases then read EARLY — each phase's
+    // fragments are loaded during the PREVIOUS phase's MFMA cluster,
+    // so the reads' LDS latency hides under matrix work.
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 6);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf0[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
+    }
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 64, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)   // ph1's fragments, early
+        bf1[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + 32 + fj * 16 + frow, kk * 64 + kb));
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph1 (jr0, jc1): bf1 already read during ph0 ---------------
+    __builtin_amdgcn_s_barrier();
+    if (pre2) glds_piece16(A, K, bm0, M, k2, Act, a_rp, lane);
+    bf16x8 af2[4][2];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)   // ph2's fragments, early
+        af2[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph2 (jr1, jc0): af2 already read during ph1 ---------------
+    __builtin_amdgcn_s_barrier();
+    if (pre2) glds_piece16(B, K, bn0, N, k2, Bct, b_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af2[fi][kk], bf0[fj][kk], acc[4 + fi][fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph3 (jr1, jc1): everything already in registers ----------
+    __builtin_amdgcn_s_barrier();
+    if (pre2) glds_piece16(B, K, bn0, N, k2, Bct, b_rp + 32, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af2[fi][kk], bf1[fj][kk], acc[4 + fi][2 + fj], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);   // drain before LDS reuse
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
+
 
 // ==== 256x256 / BK=32 three-buffer pipelined GEMM (v2) ==============
 // The round-1 256-tile kernel drains vmcnt(0) at EVERY K-tile boundary
@@ -1266,8 +1833,33 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
     // counted-vmcnt pipeline) — the measured default.
     static const int var = [] {
       const char* e = getenv("GSAMD_GEMM_VAR");
-      return e ? atoi(e) : 2;
+      return e ? atoi(e) : 2;   // p8: the measured winner (see
+                                // profiles/README.md GEMM ladder)
     }();
+    if (var == 5) {
+      hipLaunchKernelGGL(gemm_nt256_p11_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
+    if (var == 4) {
+      hipLaunchKernelGGL(gemm_nt256_p10_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
+    if (var == 3) {
+      hipLaunchKernelGGL(gemm_nt256_p9_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
     if (var == 2) {
       hipLaunchKernelGGL(gemm_nt256_p8_kernel, dim3(grid), dim3(NT2), 0,
                          stream.stream(), (const short*)a.data_ptr(),

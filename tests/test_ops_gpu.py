@@ -491,3 +491,17 @@ def test_fused_embedding_module_autograd():
     torch.nn.functional.embedding(idx, wf).float().pow(2).mean().backward()
     scale = wf.grad.abs().max().item() + 1e-4
     assert mx(m.weight.grad, wf.grad) / scale < 2e-2
+
+
+def test_gemm_nt_large_tile_path():
+    """The 256-tile pipelined kernel (p8) serves grids >= 200 blocks —
+    exercise it at a driver-relevant shape with bias+relu."""
+    torch.manual_seed(7)
+    M, N, K = 4096, 4096, 4096
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    bias = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+    c = ext().gemm_nt(a, b, bias, True)
+    ref = (a.float() @ b.float().t() + bias.float()).relu()
+    scale = ref.abs().max().item() + 1e-3
+    assert mx(c, ref) / scale < 1.5e-2
