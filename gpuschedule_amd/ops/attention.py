@@ -98,10 +98,9 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if q.is_cuda:
         ext = get_ext(required=True)
         if hasattr(ext, "attn_fwd"):
-            if k.shape[1] != q.shape[1]:  # GQA: expand KV heads
-                rep = q.shape[1] // k.shape[1]
-                k = k.repeat_interleave(rep, dim=1)
-                v = v.repeat_interleave(rep, dim=1)
+            # GQA is kernel-native: k/v keep their (fewer) heads; the
+            # kernels read kv head h/(H/HK) directly and the backward
+            # group-accumulates dK/dV over each head's query group.
             return _FlashAttnFn.apply(q.contiguous(), k.contiguous(),
                                       v.contiguous(), causal)
         raise RuntimeError("attn_fwd missing from HIP extension")

@@ -66,8 +66,11 @@ template <int D, bool CAUSAL, int NW, bool STRIDED>
 __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ Vt, short* __restrict__ O,
-    float* __restrict__ LSE, int B, int H, int S, float scale,
+    float* __restrict__ LSE, int B, int H, int HK, int S, float scale,
     long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
+  // HK = kv heads (GQA: q head h reads kv head h/(H/HK) — stride-0
+  // style sharing, no host-side K/V expansion).  The STRIDED
+  // (packed-QKV) path requires HK == H (host-checked).
   // !STRIDED folds the row strides to the compile-time D (measured:
   // runtime-stride address math cost 8% on the 8-wave long-seq path)
   const long qsS = STRIDED ? qsS_ : (long)D;
@@ -102,7 +105,9 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
 
   const long qb = (long)b * qsB + (long)h * qsH;
   const long ob = (long)b * osB + (long)h * osH;
-  const long voff = ((long)(b * H + h) * D) * S;  // Vt is [D][S]
+  const int kh = h / (H / HK);                    // GQA kv head
+  const long kb = STRIDED ? qb : ((long)(b * HK + kh) * S) * (long)D;
+  const long voff = ((long)(b * HK + kh) * D) * S;  // Vt is [D][S]
 
   // ---- Q into registers (A-fragment layout), pre-scaled ----------
   bf16x8 qreg[KC];
@@ -162,7 +167,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(
       int kvr = kv0 + r;
       long roff = (kvr < S) ? kv_add + (long)r * qsS
                             : (long)(S - 1) * qsS;
-      const void* g = K + qb + roff + cbs / 2;
+      const void* g = K + kb + roff + cbs / 2;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)g,
           (__attribute__((address_space(3))) unsigned int*)
@@ -381,7 +386,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     const short* __restrict__ Qt, const short* __restrict__ dOt,
     const float* __restrict__ LSE, const float* __restrict__ DR,
     short* __restrict__ dK, short* __restrict__ dV,
-    int B, int H, int S, float scale,
+    int B, int H, int HK, int S, float scale,
     long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
   const long qsS = STRIDED ? qsS_ : (long)D;
   const long osS = STRIDED ? osS_ : (long)D;
@@ -407,17 +412,20 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   char* my_p = p_base + wid * 16 * KVB * 2;
   char* my_ds = p_base + (4 + wid) * 16 * KVB * 2;
 
+  // grid is over (b, KV head, kv tile); for GQA the block loops the
+  // rep = H/HK query heads that share this KV head, accumulating
+  // dK/dV across the whole group IN REGISTERS — grouped (not atomic)
+  // accumulation: deterministic, K/V loaded once per group.
   const int n_kvt = (S + KVB - 1) / KVB;
   const int bh = blockIdx.x / n_kvt;
   const int kvt = blockIdx.x % n_kvt;
-  const int b = bh / H, h = bh % H;
+  const int b = bh / HK, kvh = bh % HK;
+  const int rep = H / HK;
   const int kv0 = kvt * KVB;
   const int krow = kv0 + wid * 16 + (lane & 15);
 
-  const long qvb = (long)b * qsB + (long)h * qsH;   // Q/K/V + d* base
-  const long dob = (long)b * osB + (long)h * osH;   // dO base
-  const long offT = ((long)(b * H + h) * D) * S;
-  const long roff = (long)(b * H + h) * S;
+  const long kvb = STRIDED ? ((long)b * qsB + (long)kvh * qsH)
+      : ((long)(b * HK + kvh) * S) * (long)D;   // K/V + dK/dV base
 
   // K (scaled) and V fragments in registers
   bf16x8 kreg[KC], vreg[KC];
@@ -427,13 +435,13 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     for (int c = 0; c < KC; ++c) {
       int d0 = c * 32 + (lane >> 4) * 8;
       bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
-          K + qvb + (long)row * qsS + d0);
+          K + kvb + (long)row * qsS + d0);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         kv8[j] = (__bf16)((float)kv8[j] * scale);
       kreg[c] = kv8;
       vreg[c] = *reinterpret_cast<const bf16x8*>(
-          V + qvb + (long)row * qsS + d0);
+          V + kvb + (long)row * qsS + d0);
     }
   }
 
@@ -446,6 +454,12 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
 
   const int qt_start = CAUSAL ? kv0 / 64 : 0;
   const int n_qt = (S + 63) / 64;
+  for (int g = 0; g < rep; ++g) {
+  const int h = kvh * rep + g;
+  const long qvb = (long)b * qsB + (long)h * qsH;   // Q base
+  const long dob = (long)b * osB + (long)h * osH;   // dO base
+  const long offT = ((long)(b * H + h) * D) * S;
+  const long roff = (long)(b * H + h) * S;
   for (int qt = qt_start; qt < n_qt; ++qt) {
     const int q0 = qt * 64;
     glds_tile<KVB, RB>(Q + qvb, qsS, q0, S, 0, D, q_lds, wid, lane);
@@ -526,14 +540,15 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
     }
     __syncthreads();
   }
+  }  // q-head group
 
   const int orow0 = kv0 + wid * 16 + (lane >> 4) * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int row = orow0 + r;
     if (row >= S) continue;
-    short* dkr = dK + qvb + (long)row * qsS;
-    short* dvr = dV + qvb + (long)row * qsS;
+    short* dkr = dK + kvb + (long)row * qsS;
+    short* dvr = dV + kvb + (long)row * qsS;
 #pragma unroll
     for (int f = 0; f < DF; ++f) {
       dkr[f * 16 + (lane & 15)] = f32_to_bf16(dk_acc[f][r]);
@@ -548,7 +563,7 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
     const short* __restrict__ V, const short* __restrict__ dO,
     const short* __restrict__ Kt, const float* __restrict__ LSE,
     const float* __restrict__ DR, short* __restrict__ dQ,
-    int B, int H, int S, float scale,
+    int B, int H, int HK, int S, float scale,
     long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
   const long qsS = STRIDED ? qsS_ : (long)D;
   const long osS = STRIDED ? osS_ : (long)D;
@@ -576,7 +591,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
 
   const long qvb = (long)b * qsB + (long)h * qsH;
   const long dob = (long)b * osB + (long)h * osH;
-  const long offT = ((long)(b * H + h) * D) * S;
+  const int kh = h / (H / HK);                    // GQA kv head
+  const long kvb = STRIDED ? qvb
+      : ((long)(b * HK + kh) * S) * (long)D;      // K/V base
+  const long offT = ((long)(b * HK + kh) * D) * S;  // Kt
   const long roff = (long)(b * H + h) * S;
 
   // Q (scaled) and dO fragments
@@ -619,8 +637,8 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
       : (S + KVB - 1) / KVB;
   for (int t = 0; t < n_kv; ++t) {
     const int kv0 = t * KVB;
-    glds_tile<KVB, RB>(K + qvb, qsS, kv0, S, 0, D, k_lds, wid, lane);
-    glds_tile<KVB, RB>(V + qvb, qsS, kv0, S, 0, D, v_lds, wid, lane);
+    glds_tile<KVB, RB>(K + kvb, qsS, kv0, S, 0, D, k_lds, wid, lane);
+    glds_tile<KVB, RB>(V + kvb, qsS, kv0, S, 0, D, v_lds, wid, lane);
     glds_tile<D, KVB * 2>(Kt + offT, S, 0, D, kv0, S, kt_lds, wid, lane);
     __syncthreads();
 
@@ -715,18 +733,22 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.dim() == 4, "attn_fwd: [B,H,S,D]");
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
-  TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes(),
-              "attn_fwd: expand KV heads before calling (GQA)");
+  const int HK = k.size(1);
+  TORCH_CHECK(k.sizes() == v.sizes() && k.size(0) == B &&
+              k.size(2) == S && k.size(3) == D && H % HK == 0,
+              "attn_fwd: kv heads must divide q heads (GQA)");
   TORCH_CHECK(D == 64 || D == 128, "attn_fwd: D must be 64 or 128");
   TORCH_CHECK(S % 8 == 0, "attn_fwd: S must be a multiple of 8");
   const bool cont = q.is_contiguous() && k.is_contiguous() &&
                     v.is_contiguous();
+  TORCH_CHECK(cont || HK == H,
+              "attn_fwd: GQA (kv heads < heads) needs contiguous q/k/v");
   if (!cont) check_qkv_strides(q, k, v);
   // pre-transpose V: [B,H,S,D] -> [B,H,D,S] (one memory-bound pass on
   // the batched LDS-tiled transpose; lets the PV B-fragments read
   // 8 bf16 contiguous along kv)
-  auto vt = cont ? transpose2d(v.reshape({(long)B * H, S, D}))
-                       .reshape({B, H, D, S})
+  auto vt = cont ? transpose2d(v.reshape({(long)B * HK, S, D}))
+                       .reshape({B, HK, D, S})
                  : transpose2d_bhsd(v);
   // strided (packed-QKV) path: O goes out as a [B,S,H,D] buffer viewed
   // [B,H,S,D], so the caller's head-merge reshape is a free view
@@ -758,7 +780,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                      (const short*)k.data_ptr(),                         \
                      (const short*)vt.data_ptr(),                        \
                      (short*)o.data_ptr(), lse.data_ptr<float>(),        \
-                     B, H, S, scale, q.stride(0), q.stride(1),           \
+                     B, H, HK, S, scale, q.stride(0), q.stride(1),       \
                      q.stride(2), o.stride(0), o.stride(1), o.stride(2))
 #define LAUNCH_D(DD, SR)                                                 \
   do {                                                                   \
@@ -790,8 +812,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor dout,
                                     torch::Tensor lse, bool causal) {
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int HK = k.size(1);
   const bool cont = q.is_contiguous() && k.is_contiguous() &&
                     v.is_contiguous();
+  TORCH_CHECK(cont || HK == H,
+              "attn_bwd: GQA (kv heads < heads) needs contiguous q/k/v");
   if (!cont) check_qkv_strides(q, k, v);
   TORCH_CHECK(o.stride(3) == 1 && dout.strides() == o.strides(),
               "attn_bwd: dout must match o's layout");
@@ -799,7 +824,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   // pre-transposes (batched LDS-tiled kernel) + Drow
   auto qt = cont ? transpose2d(q.reshape({(long)B * H, S, D}))
                  : transpose2d_bhsd(q);
-  auto kt = cont ? transpose2d(k.reshape({(long)B * H, S, D}))
+  auto kt = cont ? transpose2d(k.reshape({(long)B * HK, S, D}))
                  : transpose2d_bhsd(k);
   auto dot = dout.is_contiguous()
                  ? transpose2d(dout.reshape({(long)B * H, S, D}))
@@ -815,8 +840,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   torch::Tensor dq, dk, dv, dqkv;
   if (cont) {
     dq = torch::empty({B, H, S, D}, q.options());
-    dk = torch::empty({B, H, S, D}, q.options());
-    dv = torch::empty({B, H, S, D}, q.options());
+    dk = torch::empty({B, HK, S, D}, q.options());
+    dv = torch::empty({B, HK, S, D}, q.options());
   } else {
     dqkv = torch::empty({B, S, 3, H, D}, q.options());
     dq = dqkv.select(2, 0).permute({0, 2, 1, 3});
@@ -826,7 +851,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                 "attn_bwd: packed grads must mirror the input layout");
   }
   float scale = 1.0f / sqrtf((float)D);
-  int grid_kv = B * H * ((S + KVB - 1) / KVB);
+  int grid_kv = B * HK * ((S + KVB - 1) / KVB);
   int grid_q = B * H * ((S + 63) / 64);
 #define LAUNCH_BWD(DD, CC)                                              \
   do {                                                                  \
@@ -840,7 +865,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        (const short*)dot.data_ptr(),                    \
                        lse.data_ptr<float>(), drow.data_ptr<float>(),   \
                        (short*)dk.data_ptr(), (short*)dv.data_ptr(),    \
-                       B, H, S, scale, q.stride(0), q.stride(1),        \
+                       B, H, HK, S, scale, q.stride(0), q.stride(1),    \
                        q.stride(2), o.stride(0), o.stride(1),           \
                        o.stride(2));                                    \
     hipLaunchKernelGGL((attn_bwd_q_kernel<DD, CC, SR>), dim3(grid_q),   \
@@ -851,7 +876,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        (const short*)dout.data_ptr(),                   \
                        (const short*)kt.data_ptr(),                     \
                        lse.data_ptr<float>(), drow.data_ptr<float>(),   \
-                       (short*)dq.data_ptr(), B, H, S, scale,           \
+                       (short*)dq.data_ptr(), B, H, HK, S, scale,       \
                        q.stride(0), q.stride(1), q.stride(2),           \
                        o.stride(0), o.stride(1), o.stride(2));          \
   } while (0)

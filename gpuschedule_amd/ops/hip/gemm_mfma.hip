@@ -1106,7 +1106,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p9_kernel(
 
 // ==== p11: p9 merged to TWO megaphases per tile (one vmcnt wait +
 // two barriers per tile; 32-MFMA clusters) =========================
-void gemm_nt256_p11_kernel(
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_p11_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ C, const short* __restrict__ bias,
     int M, int N, int K, int relu) {
@@ -1273,7 +1273,7 @@ void gemm_nt256_p11_kernel(
 
 // ==== p10: p9 + fragment reads pipelined into the previous
 // phase's MFMA cluster (single deep wait per tile) ================
-void gemm_nt256_p10_kernel(
+__global__ __launch_bounds__(NT2, 2) void gemm_nt256_p10_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ C, const short* __restrict__ bias,
     int M, int N, int K, int relu) {

@@ -177,3 +177,44 @@ def test_attn_bwd_kernels(shape, causal):
         scale = want.abs().max().item() + 1e-4
         err = (got.float() - want).abs().max().item() / scale
         assert err < 2.5e-2, f"{name} rel err {err:.4f}"
+
+
+# ---- kernel-native GQA (kv_heads < heads, NO host expansion) --------
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("shape", [
+    (2, 8, 2, 256, 128),    # Llama-3-8B-like ratio (4 q per kv)
+    (1, 16, 8, 512, 128),
+    (2, 4, 1, 128, 64),     # MQA extreme: one kv head
+])
+def test_attn_gqa_native(shape, causal):
+    """GQA fwd/bwd straight through the kernels: k/v keep their fewer
+    heads (no repeat_interleave anywhere on the GPU path); dK/dV come
+    back with kv-head shape via grouped accumulation."""
+    from gpuschedule_amd.ops.attention import flash_attention
+    B, H, HK, S, D = shape
+    torch.manual_seed(11)
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, HK, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, HK, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = flash_attention(q, k, v, causal=causal)
+    o.float().square().mean().backward()
+    assert k.grad.shape == (B, HK, S, D)
+
+    rep = H // HK
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    ke = kf.repeat_interleave(rep, dim=1)
+    ve = vf.repeat_interleave(rep, dim=1)
+    ref = ref_attn(qf, ke, ve, causal)
+    ref.square().mean().backward()
+    scale_o = ref.abs().max().item() + 1e-3
+    assert (o.float() - ref).abs().max().item() / scale_o < 1.5e-2
+    for got, want in ((q.grad, qf.grad), (k.grad, kf.grad),
+                      (v.grad, vf.grad)):
+        s = want.abs().max().item() + 1e-6
+        assert (got.float() - want).abs().max().item() / s < 0.06
