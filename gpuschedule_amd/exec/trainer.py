@@ -32,6 +32,13 @@ from ..ops.optim import FusedAdamW, FusedSGD
 from ..parallel.ddp import GradReducer
 
 
+# MIOpen solver selection for the library conv path (ResNet): let
+# find-mode benchmark solvers per problem instead of immediate-mode
+# heuristics (A/B'd: gate to disable).
+if os.environ.get("GSAMD_MIOPEN_BENCHMARK", "1") != "0":
+    torch.backends.cudnn.benchmark = True
+
+
 def _job_seed(job_id: str) -> int:
     return zlib.crc32(job_id.encode()) & 0x7FFFFFFF
 
