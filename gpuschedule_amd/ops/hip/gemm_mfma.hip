@@ -714,6 +714,7 @@ __device__ __forceinline__ void glds_piece16(
   }
 }
 
+template <bool SPRIO>
 __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ C, const short* __restrict__ bias,
@@ -772,6 +773,11 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
   const int frow = lane & 15;
   const int kb = (lane >> 4) * 16;
 
+  // SPRIO: static arbitration boost for the second-dispatched half
+  // (waves 4-7) instead of per-cluster setprio flips — the younger
+  // half is the issue-arbitration loser on every segment.
+  if (SPRIO && wid >= 4) __builtin_amdgcn_s_setprio(1);
+
   for (int t = 0; t < nk; ++t) {
     char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
     char* Bct = Act + BM2 * ROWB;
@@ -796,7 +802,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
             Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
     }
     if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp, lane);
-    __builtin_amdgcn_s_setprio(1);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -805,7 +811,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
         for (int fj = 0; fj < 2; ++fj)
           acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(0);
 
     // ---- ph1 (jr0, jc1): needs B-jc1 ------------------------------
     if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
@@ -818,7 +824,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
         bf1[fj][kk] = *reinterpret_cast<const bf16x8*>(
             Bct + swz(wc + 32 + fj * 16 + frow, kk * 64 + kb));
     if (pre) glds_piece16(B, K, bn0, N, k1, Bn, b_rp, lane);
-    __builtin_amdgcn_s_setprio(1);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -827,7 +833,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
         for (int fj = 0; fj < 2; ++fj)
           acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(0);
 
     // ---- ph2 (jr1, jc0): needs A-jr1 ------------------------------
     if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
@@ -840,7 +846,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
         af[fi][kk] = *reinterpret_cast<const bf16x8*>(
             Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
     if (pre) glds_piece16(B, K, bn0, N, k1, Bn, b_rp + 32, lane);
-    __builtin_amdgcn_s_setprio(1);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -849,12 +855,12 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
         for (int fj = 0; fj < 2; ++fj)
           acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fi][kk], bf0[fj][kk], acc[4 + fi][fj], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(0);
 
     // ---- ph3 (jr1, jc1): everything already in registers ----------
     __builtin_amdgcn_s_barrier();
     if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 64, lane);
-    __builtin_amdgcn_s_setprio(1);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -863,7 +869,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
         for (int fj = 0; fj < 2; ++fj)
           acc[4 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fi][kk], bf1[fj][kk], acc[4 + fi][2 + fj], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+    if (!SPRIO) __builtin_amdgcn_s_setprio(0);
   }
 
   const bool mtail = bm0 + BM2 > M;
@@ -885,6 +891,214 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
           float v = acc[i][j][r] + bv;
           if (relu) v = fmaxf(v, 0.f);
           c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
+// ==== p12: p8 on v_mfma_f32_32x32x16_bf16 ==========================
+// Same consumption-staged counted-vmcnt schedule as p8, but the MFMA
+// is 32x32x16 (back-to-back issue 32 cyc/SIMD for 16384 MACs vs
+// 17 cyc for 16x16x32's 8192 — ~6% fewer issue cycles and a 2495 vs
+// 2075 TF ubench ceiling).  The 32-row fragment pattern needs its own
+// conflict-free LDS permutation: a b128 lane group reads 16 DISTINCT
+// rows of {0-3,12-15,20-27} (+32-offset) at ONE k-slot, so
+// slot' = slot ^ ((row>>1)&7) makes each group's 16 addresses hit the
+// 16 bank-16-units exactly once (checked against the lane-group
+// table; the p8 g-swizzle collides here: rows 0 and 20 alias).
+__device__ __forceinline__ int swz32(int row, int byte_in_row) {
+  return row * ROWB + (byte_in_row ^ (((row >> 1) & 7) << 4));
+}
+
+__device__ __forceinline__ void glds_piece16_32(
+    const short* __restrict__ src, long ld, int row0, int rows_max,
+    int k0, char* lds_img, int rp, int lane) {
+  const int rl = lane >> 3;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int rpp = rp + p * 8;
+    int gcolb = (((lane & 7) ^
+                  ((((rpp >> 3) & 1) * 4 + (rl >> 1)) & 7)) << 4);
+    int gr = row0 + rpp + rl;
+    if (gr >= rows_max) gr = rows_max - 1;
+    const void* g = src + (long)gr * ld + k0 + gcolb / 2;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (lds_img + rpp * ROWB),
+        16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256_p12_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int nbm = (M + BM2 - 1) / BM2;
+  int bm, bn;
+  {
+    const int GM = 8;
+    int per_band = GM * nbn;
+    int band = wg / per_band;
+    int rem = wg - band * per_band;
+    int gm = min(GM, nbm - band * GM);
+    bn = rem / gm;
+    bm = band * GM + rem - bn * gm;
+  }
+  const int bm0 = bm * BM2;
+  const int bn0 = bn * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_rp = (wid & 3) * 64 + (wid >> 2) * 16;   // + jc*32
+
+  float16v acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+
+  const int nk = K / BK2;
+  glds_piece16_32(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16_32(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp, lane);
+  glds_piece16_32(B, K, bn0, N, 0, lds + BM2 * ROWB, b_rp + 32, lane);
+  glds_piece16_32(A, K, bm0, M, 0, lds, a_rp + 64, lane);
+
+  const int frow = lane & 31;          // fragment row within 32
+  const int kb = (lane >> 5) * 16;     // k-half slot (16 B)
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
+    char* Bct = Act + BM2 * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM2 * ROWB);
+    char* Bn = An + BM2 * ROWB;
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK2;
+    bf16x8 af[2][4], bf0[4], bf1[4];
+
+    // ---- ph0 (jr0, jc0) -------------------------------------------
+    __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz32(wr + fi * 32 + frow, kk * 32 + kb));
+      bf0[kk] = *reinterpret_cast<const bf16x8*>(
+          Bct + swz32(wc + frow, kk * 32 + kb));
+    }
+    if (pre) glds_piece16_32(A, K, bm0, M, k1, An, a_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        acc[fi][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            af[fi][kk], bf0[kk], acc[fi][0], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph1 (jr0, jc1) -------------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      bf1[kk] = *reinterpret_cast<const bf16x8*>(
+          Bct + swz32(wc + 32 + frow, kk * 32 + kb));
+    if (pre) glds_piece16_32(B, K, bn0, N, k1, Bn, b_rp, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        acc[fi][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            af[fi][kk], bf1[kk], acc[fi][1], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph2 (jr1, jc0) -------------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz32(wr + 64 + fi * 32 + frow, kk * 32 + kb));
+    if (pre) glds_piece16_32(B, K, bn0, N, k1, Bn, b_rp + 32, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        acc[2 + fi][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            af[fi][kk], bf0[kk], acc[2 + fi][0], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- ph3 (jr1, jc1) -------------------------------------------
+    __builtin_amdgcn_s_barrier();
+    if (pre) glds_piece16_32(A, K, bm0, M, k1, An, a_rp + 64, lane);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        acc[2 + fi][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            af[fi][kk], bf1[kk], acc[2 + fi][1], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int col = wc + j * 32 + (lane & 31);
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = wr + i * 32 + (r & 3) + 8 * (r >> 2) +
+                    4 * (lane >> 5);
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[row * BN2 + col] = f32_to_bf16(v);
         }
       }
   }
@@ -1833,9 +2047,17 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
     // counted-vmcnt pipeline) — the measured default.
     static const int var = [] {
       const char* e = getenv("GSAMD_GEMM_VAR");
-      return e ? atoi(e) : 2;   // p8: the measured winner (see
-                                // profiles/README.md GEMM ladder)
+      return e ? atoi(e) : 7;   // p8 + static wave priority: the
+                                // measured winner (profiles/README.md)
     }();
+    if (var == 6) {
+      hipLaunchKernelGGL(gemm_nt256_p12_kernel, dim3(grid), dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
     if (var == 5) {
       hipLaunchKernelGGL(gemm_nt256_p11_kernel, dim3(grid), dim3(NT2), 0,
                          stream.stream(), (const short*)a.data_ptr(),
@@ -1860,8 +2082,18 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
       HIP_CHECK_LAST();
       return c;
     }
+    if (var == 7) {
+      hipLaunchKernelGGL((gemm_nt256_p8_kernel<true>), dim3(grid),
+                         dim3(NT2), 0,
+                         stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
     if (var == 2) {
-      hipLaunchKernelGGL(gemm_nt256_p8_kernel, dim3(grid), dim3(NT2), 0,
+      hipLaunchKernelGGL((gemm_nt256_p8_kernel<false>), dim3(grid),
+                         dim3(NT2), 0,
                          stream.stream(), (const short*)a.data_ptr(),
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
