@@ -13,6 +13,8 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                                     c10::optional<torch::Tensor> dadd);
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> bias, bool relu);
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor transpose2d(torch::Tensor x);
 torch::Tensor transpose2d_bhsd(torch::Tensor x);
 std::vector<torch::Tensor> transpose2d_colsum(torch::Tensor x);
@@ -71,6 +73,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("gamma"), py::arg("rms"),
         py::arg("dadd") = c10::optional<torch::Tensor>());
   m.def("gemm_nt", &gemm_nt, "MFMA bf16 GEMM C = A @ B^T (+bias, +relu)");
+  m.def("gemm_nn", &gemm_nn,
+        "MFMA bf16 GEMM C = A @ B, B red-major (tr_b16 fragment reads)");
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");
   m.def("transpose2d_bhsd", &transpose2d_bhsd,
         "strided [B,H,S,D] -> contiguous [B,H,D,S] bf16 transpose");

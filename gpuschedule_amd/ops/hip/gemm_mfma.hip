@@ -913,6 +913,266 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
   }
 }
 
+// ==== NN GEMM: C[M,N] = A[M,R] . B[R,N] (B stored red-major) ========
+// The backward data GEMM dX = dY . W consumes W exactly as stored, so
+// this kernel removes the explicit W^T transpose pass (round 1:
+// transpose kernels = 5.3% of the Llama-8B job).  A side: identical
+// to p8 (consumption-staged counted-vmcnt pipeline, g-swizzle).
+// B side: each K-tile's [64 red][256 out] slab is stored as 128-B
+// subtiles [4 red][16 out] and fragments are read with
+// ds_read_b64_tr_b16 — gfx950's hardware 4x4-bf16 transpose.
+// Semantics (probed on hardware, scripts/tr16_probe.py): each 16-lane
+// group gathers 16 lane-addressed 8-B chunks as a [4 red][16 out]
+// tile — lane 16g+j ADDRESSES row j>>2, chunk j&3 and RECEIVES column
+// j (4 red values of out-col j).  Subtile order perm(r4) (swap bits
+// 0<->1 within each group of 4; an involution) makes the two tiles of
+// each 32-lane bank half 256 B contiguous -> conflict-free.
+__device__ __forceinline__ int perm_nn(int r4) {
+  return (r4 & ~3) | ((r4 & 1) << 1) | ((r4 >> 1) & 1);
+}
+
+// Stage one 16-out-col band (all 64 red rows = 16 subtiles = 2 KiB)
+// of the B slab: 2 glds instrs of 8 subtiles.  tile red rows must be
+// full (red tails are the caller's padding job: red is the REDUCTION
+// dim, clamped rows would corrupt the product).
+__device__ __forceinline__ void glds_nn_piece(
+    const short* __restrict__ src, long ld, int red0, int c16,
+    int tile_out0, int outs_max, char* img, int lane) {
+  const int red_in = (lane & 7) >> 1;
+  const int out8 = (lane & 1) * 8;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int pidx = p * 8 + (lane >> 3);
+    const int r4 = perm_nn(pidx);          // involution
+    const int red = red0 + r4 * 4 + red_in;
+    int out = tile_out0 + c16 * 16 + out8;
+    if (out + 7 >= outs_max) out = max(outs_max - 8, 0);
+    const void* g = src + (long)red * ld + out;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)
+            (img + (c16 * 16 + p * 8) * 128 + (lane & 7) * 16),
+        16, 0, 0);
+  }
+}
+
+// B fragment (16 out cols at fc0, red = kk*32 + (lane>>4)*8 + 0..7)
+// via two hardware-transpose reads.  NO wait here: the caller issues
+// every fragment's reads, then one s_waitcnt lgkmcnt(0).
+__device__ __forceinline__ void bfrag_nn_issue(
+    const char* img, int fc0, int kk, int lane,
+    short4v* lo, short4v* hi) {
+  const int g = lane >> 4;
+  const int j = lane & 15;
+  const int c16 = fc0 >> 4;
+  const int r4a = kk * 8 + g * 2;
+  const int woff = (j >> 2) * 32 + (j & 3) * 8;
+  unsigned a0 = (unsigned)(size_t)(img) +
+                (unsigned)((c16 * 16 + perm_nn(r4a)) * 128 + woff);
+  unsigned a1 = (unsigned)(size_t)(img) +
+                (unsigned)((c16 * 16 + perm_nn(r4a + 1)) * 128 + woff);
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+               : "=v"(*lo) : "v"(a0));
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+               : "=v"(*hi) : "v"(a1));
+}
+
+__device__ __forceinline__ bf16x8 bfrag_nn_pack(short4v lo, short4v hi) {
+  short8 s8 = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+  return *reinterpret_cast<bf16x8*>(&s8);
+}
+
+__global__ __launch_bounds__(NT2, 1) void gemm_nn_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int R, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN2 - 1) / BN2;
+  const int nbm = (M + BM2 - 1) / BM2;
+  int bm, bn;
+  {
+    const int GM = 8;
+    int per_band = GM * nbn;
+    int band = wg / per_band;
+    int rem = wg - band * per_band;
+    int gm = min(GM, nbm - band * GM);
+    bn = rem / gm;
+    bm = band * GM + rem - bn * gm;
+  }
+  const int bm0 = bm * BM2;
+  const int bn0 = bn * BN2;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM2 * BK2 * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;
+  const int wc = (wid & 3) * 64;
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_c16 = (wid & 3) * 4 + (wid >> 2);        // + jc*2
+
+  if (true && wid >= 4) __builtin_amdgcn_s_setprio(1);
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = R / BK2;
+  glds_piece16(A, R, bm0, M, 0, lds, a_rp, lane);
+  glds_nn_piece(B, N, 0, b_c16, bn0, N, lds + BM2 * ROWB, lane);
+  glds_nn_piece(B, N, 0, b_c16 + 2, bn0, N, lds + BM2 * ROWB, lane);
+  glds_piece16(A, R, bm0, M, 0, lds, a_rp + 64, lane);
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM2 * ROWB);
+    char* Bct = Act + BM2 * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM2 * ROWB);
+    char* Bn = An + BM2 * ROWB;
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK2;
+    bf16x8 af[4][2], bf0[2][2], bf1[2][2];
+    short4v tlo[2][2], thi[2][2];
+
+    // ---- ph0 (jr0, jc0) -------------------------------------------
+    __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bfrag_nn_issue(Bct, wc + fj * 16, kk, lane,
+                       &tlo[fj][kk], &thi[fj][kk]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)");
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf0[fj][kk] = bfrag_nn_pack(tlo[fj][kk], thi[fj][kk]);
+    if (pre) glds_piece16(A, R, bm0, M, k1, An, a_rp, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
+
+    // ---- ph1 (jr0, jc1) -------------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bfrag_nn_issue(Bct, wc + 32 + fj * 16, kk, lane,
+                       &tlo[fj][kk], &thi[fj][kk]);
+    asm volatile("s_waitcnt lgkmcnt(0)");
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf1[fj][kk] = bfrag_nn_pack(tlo[fj][kk], thi[fj][kk]);
+    if (pre) glds_nn_piece(B, N, k1, b_c16, bn0, N, Bn, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
+
+    // ---- ph2 (jr1, jc0) -------------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
+    if (pre) glds_nn_piece(B, N, k1, b_c16 + 2, bn0, N, Bn, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[4 + fi][fj], 0, 0, 0);
+
+    // ---- ph3 (jr1, jc1) -------------------------------------------
+    __builtin_amdgcn_s_barrier();
+    if (pre) glds_piece16(A, R, bm0, M, k1, An, a_rp + 64, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[4 + fi][2 + fj], 0, 0, 0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BN2 > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM2][BN2]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 5;
+    int colb = (idx & 31) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN2 + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN2 + colb);
+    }
+  }
+}
+
 // ==== p12: p8 on v_mfma_f32_32x32x16_bf16 ==========================
 // Same consumption-staged counted-vmcnt schedule as p8, but the MFMA
 // is 32x32x16 (back-to-back issue 32 cyc/SIMD for 16384 MACs vs
@@ -2146,6 +2406,37 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          bias_p, M, N, K, relu ? 1 : 0);
     }
   }
+  HIP_CHECK_LAST();
+  return c;
+}
+
+// C[M,N] = A[M,R] . B[R,N] with B stored red-major (dX = dY . W with
+// W exactly as stored — no transpose pass).  256-tile only: the caller
+// gates small grids to the transpose+NT path.  R must be %64 (it is
+// the reduction dim; clamping would corrupt the product).
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> bias, bool relu) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+              b.scalar_type() == torch::kBFloat16, "gemm_nn: bf16 only");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm_nn: 2-D only");
+  const int M = a.size(0), R = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == R, "gemm_nn: reduction dim mismatch");
+  TORCH_CHECK(R % BK2 == 0, "gemm_nn: R must be a multiple of ", BK2);
+  TORCH_CHECK(N % 8 == 0, "gemm_nn: N must be a multiple of 8");
+  auto c = torch::empty({M, N}, a.options());
+  const short* bias_p = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kBFloat16 &&
+                bias->numel() == N);
+    bias_p = (const short*)bias->data_ptr();
+  }
+  long grid = (long)((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gemm_nn_kernel, dim3((int)grid), dim3(NT2), 0,
+                     stream.stream(), (const short*)a.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                     bias_p, M, N, R, relu ? 1 : 0);
   HIP_CHECK_LAST();
   return c;
 }

@@ -59,8 +59,24 @@ class _MFMALinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         if ctx.relu:
             dy = dy * (y > 0).to(dy.dtype)
-        wt = ext.transpose2d(weight)            # [K, N]
-        dx = _gemm_nt(ext, dy, wt)              # [M, K]
+        M, N = dy.shape
+        K = weight.shape[1]
+        # dx = dy @ W: gemm_nn consumes W exactly as stored ([N,K] =
+        # red-major) via ds_read_b64_tr_b16 fragment reads — no W^T
+        # pass.  MEASURED: ties the transpose+NT composite (the
+        # transpose pass was ~2-4%, and the tr-read path gives the
+        # same back in fragment-read overhead), so the proven
+        # composite stays default; GSAMD_GEMM_NN=1 enables the native
+        # path (numerics-tested either way).
+        import os
+        blocks = ((M + 255) // 256) * ((K + 255) // 256)
+        if (os.environ.get("GSAMD_GEMM_NN") == "1"
+                and N % 64 == 0 and K % 8 == 0 and M >= 256 and K >= 256
+                and blocks >= 200):
+            dx = ext.gemm_nn(dy, weight, None, False)
+        else:
+            wt = ext.transpose2d(weight)        # [K, N]
+            dx = _gemm_nt(ext, dy, wt)          # [M, K]
         dyp = _pad_rows64(dy)
         xp = _pad_rows64(x2d)
         if ctx.has_bias:

@@ -505,3 +505,15 @@ def test_gemm_nt_large_tile_path():
     ref = (a.float() @ b.float().t() + bias.float()).relu()
     scale = ref.abs().max().item() + 1e-3
     assert mx(c, ref) / scale < 1.5e-2
+
+
+def test_gemm_nn_red_major():
+    """gemm_nn: C = A @ B with B stored red-major (the dX = dY @ W
+    form) via hardware-transpose (ds_read_b64_tr_b16) fragment reads."""
+    torch.manual_seed(9)
+    M, R, N = 4096, 4096, 4096
+    a = torch.randn(M, R, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(R, N, device="cuda", dtype=torch.bfloat16)
+    c = ext().gemm_nn(a, b, None, False)
+    ref = a.float() @ b.float()
+    assert mx(c, ref) / (ref.abs().max().item() + 1e-3) < 1.5e-2
