@@ -14,6 +14,7 @@ import torch.nn.functional as F
 
 from ..ops.attention import flash_attention_qkv
 from ..ops.ce import fused_cross_entropy
+from ..ops.elementwise import fused_gelu
 from ..ops.embedding import FusedEmbedding
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedLayerNorm
@@ -50,7 +51,7 @@ class BertLayer(nn.Module):
         att = flash_attention_qkv(qkv, causal=False)
         att = att.transpose(1, 2).reshape(B, S, Hd)
         x = self.ln1(self.proj(att), residual=x)  # fused add+LN
-        h = F.gelu(self.fc1(x), approximate="tanh")
+        h = fused_gelu(self.fc1(x))
         x = self.ln2(self.fc2(h), residual=x)
         return x
 

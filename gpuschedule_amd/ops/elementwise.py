@@ -52,3 +52,26 @@ def silu_mul(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
     if g.is_cuda and g.dtype == torch.bfloat16:
         return _SiluMulFn.apply(g.contiguous(), u.contiguous())
     return torch.nn.functional.silu(g) * u
+
+
+class _GeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = get_ext(required=True)
+        ctx.save_for_backward(x)
+        return ext.gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        (x,) = ctx.saved_tensors
+        return ext.gelu_bwd(dy.contiguous(), x)
+
+
+def fused_gelu(x: torch.Tensor) -> torch.Tensor:
+    """tanh-approximate GELU: one fused bf16 pass each way on GPU
+    (torch's Gelu + GeluBackward elementwise pair otherwise); fp32
+    torch reference on CPU."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _GeluFn.apply(x.contiguous())
+    return torch.nn.functional.gelu(x, approximate="tanh")

@@ -49,6 +49,8 @@ void sgd_multi(torch::Tensor ptrs, torch::Tensor sizes,
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosb,
                          torch::Tensor sinb, bool backward);
 torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
+torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor g,
                                         torch::Tensor u);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits,
@@ -86,6 +88,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, "flash-attention backward (FA2 two-kernel)");
   m.def("rope_apply", &rope_apply, "fused rotate-half RoPE fwd/bwd");
   m.def("silu_mul_fwd", &silu_mul_fwd, "fused SwiGLU gate fwd");
+  m.def("gelu_fwd", &gelu_fwd, "fused tanh-GELU fwd (bf16)");
+  m.def("gelu_bwd", &gelu_bwd, "fused tanh-GELU bwd (bf16)");
   m.def("silu_mul_bwd", &silu_mul_bwd, "fused SwiGLU gate bwd");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward");

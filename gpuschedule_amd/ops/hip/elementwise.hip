@@ -98,6 +98,51 @@ __global__ __launch_bounds__(NT) void silu_mul_bwd_kernel(
   }
 }
 
+
+// tanh-approximate GELU (matches torch F.gelu(approximate="tanh")):
+// y = 0.5 x (1 + tanh(c(x + a x^3))), c = sqrt(2/pi), a = 0.044715.
+// One fused pass each way — replaces torch's separate Gelu{,Backward}
+// elementwise kernels in the BERT FFN.
+__global__ __launch_bounds__(NT) void gelu_fwd_kernel(
+    const short* __restrict__ x, short* __restrict__ y, long n) {
+  constexpr float C = 0.7978845608028654f, A = 0.044715f;
+  long i = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * NT * 8;
+  for (; i + 7 < n; i += stride) {
+    short8 xv = *reinterpret_cast<const short8*>(x + i);
+    short8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float v = bf16_to_f32(xv[j]);
+      float t = tanhf(C * (v + A * v * v * v));
+      out[j] = f32_to_bf16(0.5f * v * (1.f + t));
+    }
+    *reinterpret_cast<short8*>(y + i) = out;
+  }
+}
+
+__global__ __launch_bounds__(NT) void gelu_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    short* __restrict__ dx, long n) {
+  constexpr float C = 0.7978845608028654f, A = 0.044715f;
+  long i = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * NT * 8;
+  for (; i + 7 < n; i += stride) {
+    short8 dv = *reinterpret_cast<const short8*>(dy + i);
+    short8 xv = *reinterpret_cast<const short8*>(x + i);
+    short8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float v = bf16_to_f32(xv[j]);
+      float t = tanhf(C * (v + A * v * v * v));
+      float du = C * (1.f + 3.f * A * v * v);
+      float g = 0.5f * (1.f + t) + 0.5f * v * (1.f - t * t) * du;
+      out[j] = f32_to_bf16(bf16_to_f32(dv[j]) * g);
+    }
+    *reinterpret_cast<short8*>(dx + i) = out;
+  }
+}
+
 int grid_for_n(long items) {
   return (int)std::min<long>((items + NT - 1) / NT, 2048);
 }
@@ -164,4 +209,29 @@ std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor g,
                      (short*)du.data_ptr(), n);
   HIP_CHECK_LAST();
   return {dg, du};
+}
+
+torch::Tensor gelu_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16);
+  long n = x.numel();
+  TORCH_CHECK(n % 8 == 0);
+  auto y = torch::empty_like(x);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gelu_fwd_kernel, dim3(grid_for_n(n / 8)), dim3(NT),
+                     0, stream.stream(), (const short*)x.data_ptr(),
+                     (short*)y.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x) {
+  long n = x.numel();
+  auto dx = torch::empty_like(x);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gelu_bwd_kernel, dim3(grid_for_n(n / 8)), dim3(NT),
+                     0, stream.stream(), (const short*)dy.data_ptr(),
+                     (const short*)x.data_ptr(), (short*)dx.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return dx;
 }

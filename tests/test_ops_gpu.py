@@ -517,3 +517,18 @@ def test_gemm_nn_red_major():
     c = ext().gemm_nn(a, b, None, False)
     ref = a.float() @ b.float()
     assert mx(c, ref) / (ref.abs().max().item() + 1e-3) < 1.5e-2
+
+
+def test_fused_gelu_fwd_bwd():
+    from gpuschedule_amd.ops.elementwise import fused_gelu
+    torch.manual_seed(12)
+    x = torch.randn(4096, 3072, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = fused_gelu(x)
+    y.float().square().mean().backward()
+    xf = x.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.gelu(xf, approximate="tanh")
+    ref.square().mean().backward()
+    assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 8e-3
+    s = xf.grad.abs().max().item() + 1e-6
+    assert (x.grad.float() - xf.grad).abs().max().item() / s < 2e-2
