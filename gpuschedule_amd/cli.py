@@ -72,9 +72,15 @@ def cmd_replay(args):
     rank, world = init_distributed("gloo")
     n_gpus = max(cfg.get("gpus", args.gpus), world)
     device = "cuda" if torch.cuda.is_available() else "cpu"
+    # multi-node: ranks 0..world-1 span nodes of `gpus_per_node` GPUs
+    # (torchrun --nnodes N); the cluster model bounds gang windows to
+    # one node, so RCCL gangs never cross nodes and the gloo control
+    # plane is the only cross-node traffic.
+    gpn = cfg.get("gpus_per_node", getattr(args, "gpus_per_node", 0)) \
+        or max(n_gpus, 8)
     sched = None
     if rank == 0:
-        sched = Scheduler(Cluster(n_gpus, max(n_gpus, 8)),
+        sched = Scheduler(Cluster(n_gpus, gpn),
                           _policy_from_args(args, cfg))
         sched.submit(_specs_from_args(args, cfg))
     be = TorchBackend(sched, rank=rank, world=world, device_type=device,
@@ -138,6 +144,9 @@ def main(argv=None):
     p.set_defaults(fn=cmd_simulate)
     p = sub.add_parser("replay", parents=[common])
     p.add_argument("--max-seconds", type=float, default=None)
+    p.add_argument("--gpus-per-node", type=int, default=0,
+                   help="GPUs per node for multi-node torchrun replays "
+                        "(0 = single node); gangs never cross nodes")
     p.set_defaults(fn=cmd_replay)
     p = sub.add_parser("gen-trace")
     p.add_argument("--jobs", type=int, default=64)

@@ -366,3 +366,17 @@ def test_multiproc_scenarios(tmp_path, scenario):
         assert s["preemptions"] >= 1, events
     if scenario == "gandiva4":
         assert s["preemptions"] >= 1, events
+
+
+def test_multinode_gangs_stay_node_local(tmp_path):
+    """world 4 modeled as 2 nodes x 2 GPUs: a 2-wide gang must land on
+    GPUs {0,1} or {2,3}, never {1,2} (xGMI/RCCL is per node)."""
+    from gpuschedule_amd.sched.policies.fifo import FIFOPolicy
+    sched = Scheduler(Cluster(4, gpus_per_node=2), FIFOPolicy())
+    sched.submit([J("a", n=2, iters=4), J("b", n=2, iters=4),
+                  J("c", n=2, iters=4)])
+    plan = sched.tick(0.0)
+    placed = {p.job_id: p.gpus for p in plan.places}
+    for jid, gpus in placed.items():
+        assert gpus in ([0, 1], [2, 3]), f"{jid} crossed nodes: {gpus}"
+    assert len(placed) == 2  # third job queues: no node window free
