@@ -70,7 +70,15 @@ def cmd_replay(args):
     from .exec.backend import init_distributed
     cfg = _load_config(args.config) if args.config else {}
     rank, world = init_distributed("gloo")
-    n_gpus = max(cfg.get("gpus", args.gpus), world)
+    # the REAL backend needs one rank per GPU: the cluster is sized by
+    # the launched world, not the config (configs size the SIMULATOR;
+    # under torchrun the two should agree).
+    n_gpus = world
+    cfg_gpus = cfg.get("gpus", args.gpus)
+    if rank == 0 and cfg_gpus != world:
+        import sys
+        print(f"replay: world={world} ranks; config wants {cfg_gpus} "
+              f"GPUs — running on {world}", file=sys.stderr)
     device = "cuda" if torch.cuda.is_available() else "cpu"
     # multi-node: ranks 0..world-1 span nodes of `gpus_per_node` GPUs
     # (torchrun --nnodes N); the cluster model bounds gang windows to
