@@ -913,6 +913,192 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
   }
 }
 
+// ==== 128x128 / BK=64 consumption-pipelined GEMM (p8-128) ==========
+// The p8 discipline (per-wave consumption-ordered pieces, counted
+// vmcnt(4) at every phase, one barrier per phase, g-swizzle) on a
+// 128x128 tile with 4 waves: serves the small-N GEMMs (BERT hidden
+// 768) whose grids cannot fill the chip with 256x256 blocks.  64 KiB
+// LDS -> 2 blocks/CU, so two blocks' phase schedules interleave on
+// each CU.  Replaces the round-1 gemm_nt_kernel (vmcnt(0) drain per
+// tile, MfmaUtil 19%) as the small-grid default.
+__global__ __launch_bounds__(NTHREADS, 2) void gemm_nt128_p8_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BN - 1) / BN;
+  const int nbm = (M + BM - 1) / BM;
+  int bm, bn;
+  {
+    const int GM = 8;
+    int per_band = GM * nbn;
+    int band = wg / per_band;
+    int rem = wg - band * per_band;
+    int gm = min(GM, nbm - band * GM);
+    bn = rem / gm;
+    bm = band * GM + rem - bn * gm;
+  }
+  const int bm0 = bm * BM;
+  const int bn0 = bn * BN;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * BM * BK * 2];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 1) * 64;    // 2(M) x 2(N) wave grid
+  const int wc = (wid & 1) * 64;
+  const int a_rp = (wid >> 1) * 64 + (wid & 1) * 16;  // + jr*32
+  const int b_rp = (wid & 1) * 64 + (wid >> 1) * 16;  // + jc*32
+
+  if (wid >= 2) __builtin_amdgcn_s_setprio(1);  // younger-half boost
+
+  float4v acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK;
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM * ROWB, b_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + BM * ROWB, b_rp + 32, lane);
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp + 32, lane);
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (2 * BM * ROWB);
+    char* Bct = Act + BM * ROWB;
+    char* An = lds + ((t + 1) & 1) * (2 * BM * ROWB);
+    char* Bn = An + BM * ROWB;
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK;
+    bf16x8 af[2][2], bf0[2][2], bf1[2][2];
+
+    // ---- ph0 (jr0, jc0) -------------------------------------------
+    __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf0[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
+    }
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[fi][fj], 0, 0, 0);
+
+    // ---- ph1 (jr0, jc1) -------------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf1[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + 32 + fj * 16 + frow, kk * 64 + kb));
+    if (pre) glds_piece16(B, K, bn0, N, k1, Bn, b_rp, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[fi][2 + fj], 0, 0, 0);
+
+    // ---- ph2 (jr1, jc0) -------------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 32 + fi * 16 + frow, kk * 64 + kb));
+    if (pre) glds_piece16(B, K, bn0, N, k1, Bn, b_rp + 32, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[2 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf0[fj][kk], acc[2 + fi][fj], 0, 0, 0);
+
+    // ---- ph3 (jr1, jc1) -------------------------------------------
+    __builtin_amdgcn_s_barrier();
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 32, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[2 + fi][2 + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf1[fj][kk], acc[2 + fi][2 + fj], 0, 0, 0);
+  }
+
+  const bool mtail = bm0 + BM > M;
+  const bool ntail = bn0 + BN > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [BM][BN]
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BN + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int idx = p * NTHREADS + (int)threadIdx.x;
+    int row = idx >> 4;
+    int colb = (idx & 15) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BN + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BN + colb);
+    }
+  }
+}
+
 // ==== NN GEMM: C[M,N] = A[M,R] . B[R,N] (B stored red-major) ========
 // The backward data GEMM dX = dY . W consumes W exactly as stored, so
 // this kernel removes the explicit W^T transpose pass (round 1:
@@ -2386,14 +2572,25 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          bias_p, M, N, K, relu ? 1 : 0);
     }
   } else {
-    // A/B result: the 32x32 MFMA variant ties at 4k/8k and loses on
-    // small grids (the structure is stage/barrier-bound, not
-    // MFMA-issue-bound), so 16x16 stays the default.
+    // small-grid default: the p8-discipline 128-tile kernel (2
+    // blocks/CU); GSAMD_GEMM128=0 falls back to the round-1 kernel.
+    static const int use128p8 = [] {
+      const char* e = getenv("GSAMD_GEMM128");
+      return !e || e[0] != '0';
+    }();
     static const bool use32 = [] {
       const char* e = getenv("GSAMD_GEMM32");
       return e && e[0] == '1';
     }();
     int grid = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+    if (use128p8 && !use32) {
+      hipLaunchKernelGGL(gemm_nt128_p8_kernel, dim3(grid), dim3(NTHREADS),
+                         0, stream.stream(), (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
     if (use32) {
       hipLaunchKernelGGL(gemm_nt32_kernel, dim3(grid), dim3(NTHREADS), 0,
                          stream.stream(), (const short*)a.data_ptr(),
