@@ -380,3 +380,21 @@ def test_multinode_gangs_stay_node_local(tmp_path):
     for jid, gpus in placed.items():
         assert gpus in ([0, 1], [2, 3]), f"{jid} crossed nodes: {gpus}"
     assert len(placed) == 2  # third job queues: no node window free
+
+
+def test_checkpoint_restores_rng_state(tmp_path):
+    """The documented checkpoint format carries torch RNG state; a
+    restore must resume the stochastic stream where the save left it."""
+    ck = CheckpointStore(str(tmp_path / "ck"))
+    r = JobRunner("j", "llama3-tiny", 2, "cpu", torch.float32, None, 0, 1)
+    r.step(2)
+    torch.manual_seed(1234)
+    torch.rand(7)  # advance the global stream to a known point
+    ck.save("j", r.model, r.optim, r.done_iters)
+    want = torch.rand(4)  # what the stream yields after the save point
+
+    torch.manual_seed(999)  # scramble
+    r2 = JobRunner("j", "llama3-tiny", 2, "cpu", torch.float32, None, 0, 1,
+                   state=ck.load("j"))
+    got = torch.rand(4)
+    assert torch.equal(got, want), "RNG stream not restored"

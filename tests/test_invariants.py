@@ -125,3 +125,18 @@ def test_elastic_resize_events_consistent():
         assert e["job"] in ids
         assert e["new_gpus"] >= 1 and e["old_gpus"] >= 1
         assert e["new_gpus"] != e["old_gpus"]
+
+
+def test_chrome_trace_evict_marker():
+    """The standalone `evict` event renders as an instant marker in the
+    Chrome/Perfetto export (the old dead preempt-evicted branch)."""
+    from gpuschedule_amd.sched.events import EventLog
+    log = EventLog()
+    log.emit(0.0, "submit", "a", n_gpus=1, model="m", iters=2, ideal_s=1)
+    log.emit(0.1, "start", "a", gpus=[0], first=True)
+    log.emit(0.5, "preempt", "a", reason="policy")
+    log.emit(0.5, "evict", "a")
+    log.emit(1.0, "complete", "a", jct=1.0, iters=2)
+    tr = log.to_chrome_trace()
+    names = [e.get("name") for e in tr["traceEvents"]]
+    assert "evict a" in names
