@@ -1099,6 +1099,166 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt128_p8_kernel(
   }
 }
 
+// ==== 256x128 / BK=64 consumption-pipelined GEMM (p8h) ==============
+// Half-width p8 for grids where 256x256 blocks cannot fill the chip
+// (e.g. Llama-3-8B at batch 1: M = 2048): same consumption-ordered
+// counted-vmcnt discipline, 8 waves in a 2(M) x 4(N) grid with
+// per-wave 128x32 output, TWO phases per K-tile (jr halves; both fj
+// columns in one 16-MFMA cluster).  LDS = 2 x (A 32K + B 16K) = 96 K
+// -> 1 block/CU; blocks = (M/256)*(N/128).
+__global__ __launch_bounds__(NT2, 1) void gemm_nt256h_p8_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, const short* __restrict__ bias,
+    int M, int N, int K, int relu) {
+  constexpr int BNH = 128;
+  constexpr int ATILE = 256 * ROWB;      // 32 KiB
+  constexpr int BTILE = BNH * ROWB;      // 16 KiB
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int nbn = (N + BNH - 1) / BNH;
+  const int nbm = (M + BM2 - 1) / BM2;
+  int bm, bn;
+  {
+    const int GM = 8;
+    int per_band = GM * nbn;
+    int band = wg / per_band;
+    int rem = wg - band * per_band;
+    int gm = min(GM, nbm - band * GM);
+    bn = rem / gm;
+    bm = band * GM + rem - bn * gm;
+  }
+  const int bm0 = bm * BM2;
+  const int bn0 = bn * BNH;
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * (ATILE + BTILE)];
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 2) * 128;   // 2(M) x 4(N)
+  const int wc = (wid & 3) * 32;
+  const int a_rp = (wid >> 2) * 128 + (wid & 3) * 16;  // + jr*64
+  const int b_rp = wid * 16;                           // B: 128 rows/8
+
+  if (wid >= 4) __builtin_amdgcn_s_setprio(1);
+
+  float4v acc[8][2];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK2;
+  // prologue FIFO per tile: [A-jr0(2), B(2), A-jr1(2)]
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp, lane);
+  glds_piece16(B, K, bn0, N, 0, lds + ATILE, b_rp, lane);
+  glds_piece16(A, K, bm0, M, 0, lds, a_rp + 64, lane);
+
+  const int frow = lane & 15;
+  const int kb = (lane >> 4) * 16;
+
+  for (int t = 0; t < nk; ++t) {
+    char* Act = lds + (t & 1) * (ATILE + BTILE);
+    char* Bct = Act + ATILE;
+    char* An = lds + ((t + 1) & 1) * (ATILE + BTILE);
+    char* Bn = An + ATILE;
+    const bool pre = t + 1 < nk;
+    const int k1 = (t + 1) * BK2;
+    bf16x8 af[4][2], bf[2][2];
+
+    // ---- ph0 (jr0, both fj): needs A-jr0 + B = the oldest 4 --------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    else __builtin_amdgcn_s_waitcnt(0xf70 | 2);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + fi * 16 + frow, kk * 64 + kb));
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        bf[fj][kk] = *reinterpret_cast<const bf16x8*>(
+            Bct + swz(wc + fj * 16 + frow, kk * 64 + kb));
+    }
+    if (pre) {
+      glds_piece16(A, K, bm0, M, k1, An, a_rp, lane);
+      glds_piece16(B, K, bn0, N, k1, Bn, b_rp, lane);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf[fj][kk], acc[fi][fj], 0, 0, 0);
+
+    // ---- ph1 (jr1): needs A-jr1 ------------------------------------
+    if (pre) __builtin_amdgcn_s_waitcnt(0xf70 | 4);
+    else __builtin_amdgcn_s_waitcnt(0xf70);
+    __builtin_amdgcn_s_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+        af[fi][kk] = *reinterpret_cast<const bf16x8*>(
+            Act + swz(wr + 64 + fi * 16 + frow, kk * 64 + kb));
+    if (pre) glds_piece16(A, K, bm0, M, k1, An, a_rp + 64, lane);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[4 + fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi][kk], bf[fj][kk], acc[4 + fi][fj], 0, 0, 0);
+  }
+
+  const bool mtail = bm0 + BM2 > M;
+  const bool ntail = bn0 + BNH > N;
+  __builtin_amdgcn_s_waitcnt(0xf70);
+  __builtin_amdgcn_s_barrier();
+  short* c_lds = (short*)lds;  // [256][BNH] = 64 KiB
+  {
+    const int lrow0 = wr + (lane >> 4) * 4;
+    const int lcol0 = wc + (lane & 15);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int col = lcol0 + j * 16;
+        float bv = bias ? bf16_to_f32(bias[min(bn0 + col, N - 1)]) : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[i][j][r] + bv;
+          if (relu) v = fmaxf(v, 0.f);
+          c_lds[(lrow0 + i * 16 + r) * BNH + col] = f32_to_bf16(v);
+        }
+      }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int idx = p * NT2 + (int)threadIdx.x;
+    int row = idx >> 4;
+    int colb = (idx & 15) * 8;
+    int grow = bm0 + row;
+    int gcol = bn0 + colb;
+    if (mtail && grow >= M) continue;
+    if (ntail && gcol + 7 >= N) {
+      for (int j = 0; j < 8 && gcol + j < N; ++j)
+        C[(long)grow * N + gcol + j] = c_lds[row * BNH + colb + j];
+    } else if (!ntail || gcol < N) {
+      *reinterpret_cast<short8*>(C + (long)grow * N + gcol) =
+          *reinterpret_cast<const short8*>(c_lds + row * BNH + colb);
+    }
+  }
+}
+
 // ==== NN GEMM: C[M,N] = A[M,R] . B[R,N] (B stored red-major) ========
 // The backward data GEMM dX = dY . W consumes W exactly as stored, so
 // this kernel removes the explicit W^T transpose pass (round 1:
@@ -2486,6 +2646,14 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
   // when its grid fills most of the 256 CUs; otherwise the 128-tile
   // kernel's 4x block count wins.
   long blocks256 = (long)((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+  static const bool dbg = [] {
+    const char* e = getenv("GSAMD_GEMM_DEBUG");
+    return e && e[0] == '1';
+  }();
+  if (dbg)
+    fprintf(stderr, "gemm_nt M=%d N=%d K=%d blocks256=%ld path=%s\n",
+            M, N, K, blocks256,
+            (M >= 256 && N >= 256 && blocks256 >= 200) ? "256" : "128");
   if (M >= 256 && N >= 256 && blocks256 >= 200) {
     int grid = (int)blocks256;
     // variant select: 0 = round-1 kernel (vmcnt(0) drain per tile),
@@ -2572,6 +2740,25 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          bias_p, M, N, K, relu ? 1 : 0);
     }
   } else {
+    // half-width 256x128 p8 for mid grids (llama batch-1: M=2048)
+    // MEASURED: ties the 128-tile path at big-K M=2048 shapes (+0-3%)
+    // and LOSES ~15% at small-K (BERT, K=768: 1 block/CU cannot beat
+    // the 128-tile's 2 blocks/CU when the prologue dominates) -> off
+    // by default, kept as a gated experiment.
+    static const bool use_h = [] {
+      const char* e = getenv("GSAMD_GEMM_H");
+      return e && e[0] == '1';
+    }();
+    long blocksH = (long)((M + BM2 - 1) / BM2) * ((N + 127) / 128);
+    if (use_h && M >= 256 && N >= 128 && blocksH >= 200) {
+      hipLaunchKernelGGL(gemm_nt256h_p8_kernel, dim3((int)blocksH),
+                         dim3(NT2), 0, stream.stream(),
+                         (const short*)a.data_ptr(),
+                         (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                         bias_p, M, N, K, relu ? 1 : 0);
+      HIP_CHECK_LAST();
+      return c;
+    }
     // small-grid default: the p8-discipline 128-tile kernel (2
     // blocks/CU); GSAMD_GEMM128=0 falls back to the round-1 kernel.
     static const int use128p8 = [] {
