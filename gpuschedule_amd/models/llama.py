@@ -16,7 +16,7 @@ import torch.nn as nn
 
 from ..ops.attention import flash_attention
 from ..ops.ce import fused_cross_entropy
-from ..ops.elementwise import fused_rope, silu_mul
+from ..ops.elementwise import fused_rope, swiglu_packed
 from ..ops.embedding import FusedEmbedding
 from ..ops.linear import MFMALinear
 from ..ops.norms import FusedRMSNorm
@@ -84,9 +84,7 @@ class LlamaLayer(nn.Module):
         att = att.transpose(1, 2).reshape(B, S, cfg.hidden)
         # fused residual-add + RMSNorm; summed stream comes back out
         n, x = self.norm2(self.o(att), residual=x, return_sum=True)
-        gu = self.gate_up(n)
-        g, u = gu.chunk(2, dim=-1)
-        x = x + self.down(silu_mul(g, u))
+        x = x + self.down(swiglu_packed(self.gate_up(n)))
         return x
 
 

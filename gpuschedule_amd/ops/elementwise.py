@@ -75,3 +75,28 @@ def fused_gelu(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda and x.dtype == torch.bfloat16:
         return _GeluFn.apply(x.contiguous())
     return torch.nn.functional.gelu(x, approximate="tanh")
+
+
+class _SwigluPackedFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        ext = get_ext(required=True)
+        ctx.save_for_backward(gu)
+        return ext.swiglu_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_ext(required=True)
+        (gu,) = ctx.saved_tensors
+        return ext.swiglu_bwd(dy.contiguous(), gu)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+    """SwiGLU on the PACKED [.., 2F] gate_up projection: the kernels
+    read both halves through the row stride and the backward writes one
+    packed grad buffer — no contiguous() copies of the halves and no
+    chunk-backward cat.  CPU: fp32 torch reference."""
+    if gu.is_cuda and gu.dtype == torch.bfloat16:
+        return _SwigluPackedFn.apply(gu.contiguous())
+    g, u = gu.chunk(2, dim=-1)
+    return torch.nn.functional.silu(g) * u

@@ -50,6 +50,8 @@ torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosb,
                          torch::Tensor sinb, bool backward);
 torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
 torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor swiglu_fwd(torch::Tensor gu);
+torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu);
 torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor g,
                                         torch::Tensor u);
@@ -89,6 +91,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_apply", &rope_apply, "fused rotate-half RoPE fwd/bwd");
   m.def("silu_mul_fwd", &silu_mul_fwd, "fused SwiGLU gate fwd");
   m.def("gelu_fwd", &gelu_fwd, "fused tanh-GELU fwd (bf16)");
+  m.def("swiglu_fwd", &swiglu_fwd,
+        "packed SwiGLU fwd: y = silu(gu[..,:F]) * gu[..,F:]");
+  m.def("swiglu_bwd", &swiglu_bwd,
+        "packed SwiGLU bwd: one [rows,2F] grad buffer, no cat");
   m.def("gelu_bwd", &gelu_bwd, "fused tanh-GELU bwd (bf16)");
   m.def("silu_mul_bwd", &silu_mul_bwd, "fused SwiGLU gate bwd");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (bf16 logits)");

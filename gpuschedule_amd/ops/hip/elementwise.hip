@@ -143,6 +143,66 @@ __global__ __launch_bounds__(NT) void gelu_bwd_kernel(
   }
 }
 
+
+// Packed SwiGLU: gu = [rows][2F] (one fused gate_up projection);
+// y = silu(gu[:, :F]) * gu[:, F:].  The packed forms read the
+// projection's halves through the row stride and the backward writes
+// BOTH grads into one [rows][2F] buffer — no contiguous() copies in,
+// no chunk-backward cat out.
+__global__ __launch_bounds__(NT) void swiglu_fwd_kernel(
+    const short* __restrict__ gu, short* __restrict__ y,
+    long rows, int F) {
+  const long per_row = F / 8;
+  const long total = rows * per_row;
+  long idx = (long)blockIdx.x * NT + threadIdx.x;
+  const long stride = (long)gridDim.x * NT;
+  for (; idx < total; idx += stride) {
+    const long r = idx / per_row;
+    const int c = (int)(idx % per_row) * 8;
+    const short* base = gu + r * 2 * F;
+    short8 gv = *reinterpret_cast<const short8*>(base + c);
+    short8 uv = *reinterpret_cast<const short8*>(base + F + c);
+    short8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gg = bf16_to_f32(gv[j]);
+      float sig = 1.f / (1.f + __expf(-gg));
+      out[j] = f32_to_bf16(gg * sig * bf16_to_f32(uv[j]));
+    }
+    *reinterpret_cast<short8*>(y + r * F + c) = out;
+  }
+}
+
+__global__ __launch_bounds__(NT) void swiglu_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ gu,
+    short* __restrict__ dgu, long rows, int F) {
+  const long per_row = F / 8;
+  const long total = rows * per_row;
+  long idx = (long)blockIdx.x * NT + threadIdx.x;
+  const long stride = (long)gridDim.x * NT;
+  for (; idx < total; idx += stride) {
+    const long r = idx / per_row;
+    const int c = (int)(idx % per_row) * 8;
+    const short* base = gu + r * 2 * F;
+    short8 dv = *reinterpret_cast<const short8*>(dy + r * F + c);
+    short8 gv = *reinterpret_cast<const short8*>(base + c);
+    short8 uv = *reinterpret_cast<const short8*>(base + F + c);
+    short8 og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float d = bf16_to_f32(dv[j]);
+      float gg = bf16_to_f32(gv[j]);
+      float sig = 1.f / (1.f + __expf(-gg));
+      ou[j] = f32_to_bf16(d * gg * sig);
+      og[j] = f32_to_bf16(d * bf16_to_f32(uv[j]) *
+                          (sig * (1.f + gg * (1.f - sig))));
+    }
+    short* dbase = dgu + r * 2 * F;
+    *reinterpret_cast<short8*>(dbase + c) = og;
+    *reinterpret_cast<short8*>(dbase + F + c) = ou;
+  }
+}
+
 int grid_for_n(long items) {
   return (int)std::min<long>((items + NT - 1) / NT, 2048);
 }
@@ -234,4 +294,37 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x) {
                      (const short*)x.data_ptr(), (short*)dx.data_ptr(), n);
   HIP_CHECK_LAST();
   return dx;
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor gu) {
+  TORCH_CHECK(gu.is_cuda() && gu.is_contiguous() && gu.dim() >= 2 &&
+              gu.scalar_type() == torch::kBFloat16);
+  const int F2 = gu.size(-1);
+  TORCH_CHECK(F2 % 16 == 0, "swiglu: last dim must be 2F, F % 8 == 0");
+  const int F = F2 / 2;
+  long rows = gu.numel() / F2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = F;
+  auto y = torch::empty(sizes, gu.options());
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid_for_n(rows * (F / 8))),
+                     dim3(NT), 0, stream.stream(),
+                     (const short*)gu.data_ptr(), (short*)y.data_ptr(),
+                     rows, F);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu) {
+  const int F = gu.size(-1) / 2;
+  long rows = gu.numel() / (2 * F);
+  auto dgu = torch::empty_like(gu);
+  auto stream = c10::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid_for_n(rows * (F / 8))),
+                     dim3(NT), 0, stream.stream(),
+                     (const short*)dy.data_ptr(),
+                     (const short*)gu.data_ptr(),
+                     (short*)dgu.data_ptr(), rows, F);
+  HIP_CHECK_LAST();
+  return dgu;
 }

@@ -532,3 +532,19 @@ def test_fused_gelu_fwd_bwd():
     assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 8e-3
     s = xf.grad.abs().max().item() + 1e-6
     assert (x.grad.float() - xf.grad).abs().max().item() / s < 2e-2
+
+
+def test_swiglu_packed_fwd_bwd():
+    from gpuschedule_amd.ops.elementwise import swiglu_packed
+    torch.manual_seed(13)
+    gu = torch.randn(512, 2 * 1408, device="cuda", dtype=torch.bfloat16,
+                     requires_grad=True)
+    y = swiglu_packed(gu)
+    y.float().square().mean().backward()
+    guf = gu.detach().float().requires_grad_(True)
+    g, u = guf.chunk(2, dim=-1)
+    ref = torch.nn.functional.silu(g) * u
+    ref.square().mean().backward()
+    assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 1e-2
+    s = guf.grad.abs().max().item() + 1e-6
+    assert (gu.grad.float() - guf.grad).abs().max().item() / s < 2e-2
