@@ -84,7 +84,8 @@ class LlamaLayer(nn.Module):
         att = att.transpose(1, 2).reshape(B, S, cfg.hidden)
         # fused residual-add + RMSNorm; summed stream comes back out
         n, x = self.norm2(self.o(att), residual=x, return_sum=True)
-        x = x + self.down(swiglu_packed(self.gate_up(n)))
+        # residual add fused into the down-projection epilogue
+        x = self.down(swiglu_packed(self.gate_up(n)), add=x)
         return x
 
 

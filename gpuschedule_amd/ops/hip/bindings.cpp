@@ -12,7 +12,8 @@ std::vector<torch::Tensor> norm_bwd(torch::Tensor dy, torch::Tensor x,
                                     bool rms,
                                     c10::optional<torch::Tensor> dadd);
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
-                      c10::optional<torch::Tensor> bias, bool relu);
+                      c10::optional<torch::Tensor> bias, bool relu,
+                      c10::optional<torch::Tensor> add);
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor transpose2d(torch::Tensor x);
@@ -76,7 +77,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dy"), py::arg("x"), py::arg("mean"), py::arg("rstd"),
         py::arg("gamma"), py::arg("rms"),
         py::arg("dadd") = c10::optional<torch::Tensor>());
-  m.def("gemm_nt", &gemm_nt, "MFMA bf16 GEMM C = A @ B^T (+bias, +relu)");
+  m.def("gemm_nt", &gemm_nt,
+        "MFMA bf16 GEMM C = A @ B^T (+bias, +relu, +fused addend)",
+        py::arg("a"), py::arg("b"), py::arg("bias"), py::arg("relu"),
+        py::arg("add") = c10::optional<torch::Tensor>());
   m.def("gemm_nn", &gemm_nn,
         "MFMA bf16 GEMM C = A @ B, B red-major (tr_b16 fragment reads)");
   m.def("transpose2d", &transpose2d, "bf16 2-D transpose");

@@ -718,7 +718,7 @@ template <bool SPRIO>
 __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ C, const short* __restrict__ bias,
-    int M, int N, int K, int relu) {
+    const short* __restrict__ ADD, int M, int N, int K, int relu) {
   int nwg = gridDim.x;
   int wg = blockIdx.x;
   {
@@ -889,6 +889,10 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float v = acc[i][j][r] + bv;
+          if (ADD)  // fused elementwise addend (residual stream)
+            v += bf16_to_f32(ADD[(long)min(bm0 + lrow0 + i * 16 + r,
+                                           M - 1) * N +
+                                 min(bn0 + col, N - 1)]);
           if (relu) v = fmaxf(v, 0.f);
           c_lds[(lrow0 + i * 16 + r) * BN2 + col] = f32_to_bf16(v);
         }
@@ -924,7 +928,7 @@ __global__ __launch_bounds__(NT2, 1) void gemm_nt256_p8_kernel(
 __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt128_p8_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ C, const short* __restrict__ bias,
-    int M, int N, int K, int relu) {
+    const short* __restrict__ ADD, int M, int N, int K, int relu) {
   int nwg = gridDim.x;
   int wg = blockIdx.x;
   {
@@ -1075,6 +1079,10 @@ __global__ __launch_bounds__(NTHREADS, 2) void gemm_nt128_p8_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float v = acc[i][j][r] + bv;
+          if (ADD)
+            v += bf16_to_f32(ADD[(long)min(bm0 + lrow0 + i * 16 + r,
+                                           M - 1) * N +
+                                 min(bn0 + col, N - 1)]);
           if (relu) v = fmaxf(v, 0.f);
           c_lds[(lrow0 + i * 16 + r) * BN + col] = f32_to_bf16(v);
         }
@@ -2626,7 +2634,8 @@ __global__ void colsum_reduce_kernel(const float* __restrict__ slots,
 }  // namespace
 
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
-                      c10::optional<torch::Tensor> bias, bool relu) {
+                      c10::optional<torch::Tensor> bias, bool relu,
+                      c10::optional<torch::Tensor> add) {
   TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
   TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
               b.scalar_type() == torch::kBFloat16, "gemm_nt: bf16 only");
@@ -2641,6 +2650,19 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                 bias->numel() == N);
     bias_p = (const short*)bias->data_ptr();
   }
+  // optional fused elementwise addend (residual stream): supported by
+  // the p8 / p8-128 default kernels; gated experimental variants fall
+  // back to a separate add below.
+  const short* add_p = nullptr;
+  if (add.has_value() && add->defined()) {
+    TORCH_CHECK(add->scalar_type() == torch::kBFloat16 &&
+                add->is_contiguous() && add->sizes() == c.sizes());
+    add_p = (const short*)add->data_ptr();
+  }
+  auto finish = [&](torch::Tensor out) {  // fallback for variants
+    if (add_p) out.add_(*add);             // without the fused addend
+    return out;
+  };
   auto stream = c10::cuda::getCurrentCUDAStream();
   // 256-tile kernel runs 1 block/CU (128 KiB LDS): only profitable
   // when its grid fills most of the 256 CUs; otherwise the 128-tile
@@ -2670,7 +2692,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
-      return c;
+      return finish(c);
     }
     if (var == 5) {
       hipLaunchKernelGGL(gemm_nt256_p11_kernel, dim3(grid), dim3(NT2), 0,
@@ -2678,7 +2700,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
-      return c;
+      return finish(c);
     }
     if (var == 4) {
       hipLaunchKernelGGL(gemm_nt256_p10_kernel, dim3(grid), dim3(NT2), 0,
@@ -2686,7 +2708,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
-      return c;
+      return finish(c);
     }
     if (var == 3) {
       hipLaunchKernelGGL(gemm_nt256_p9_kernel, dim3(grid), dim3(NT2), 0,
@@ -2694,14 +2716,14 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
-      return c;
+      return finish(c);
     }
     if (var == 7) {
       hipLaunchKernelGGL((gemm_nt256_p8_kernel<true>), dim3(grid),
                          dim3(NT2), 0,
                          stream.stream(), (const short*)a.data_ptr(),
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
-                         bias_p, M, N, K, relu ? 1 : 0);
+                         bias_p, add_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
       return c;
     }
@@ -2710,7 +2732,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          dim3(NT2), 0,
                          stream.stream(), (const short*)a.data_ptr(),
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
-                         bias_p, M, N, K, relu ? 1 : 0);
+                         bias_p, add_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
       return c;
     }
@@ -2720,7 +2742,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
-      return c;
+      return finish(c);
     }
     // consumption-scheduled staging variant (counted vmcnt waits, no
     // boundary drain) — A/B gate while it is being qualified
@@ -2757,7 +2779,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
                          bias_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
-      return c;
+      return finish(c);
     }
     // small-grid default: the p8-discipline 128-tile kernel (2
     // blocks/CU); GSAMD_GEMM128=0 falls back to the round-1 kernel.
@@ -2774,7 +2796,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
       hipLaunchKernelGGL(gemm_nt128_p8_kernel, dim3(grid), dim3(NTHREADS),
                          0, stream.stream(), (const short*)a.data_ptr(),
                          (const short*)b.data_ptr(), (short*)c.data_ptr(),
-                         bias_p, M, N, K, relu ? 1 : 0);
+                         bias_p, add_p, M, N, K, relu ? 1 : 0);
       HIP_CHECK_LAST();
       return c;
     }
@@ -2791,7 +2813,7 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
     }
   }
   HIP_CHECK_LAST();
-  return c;
+  return finish(c);
 }
 
 // C[M,N] = A[M,R] . B[R,N] with B stored red-major (dX = dY . W with

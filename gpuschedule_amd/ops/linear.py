@@ -31,7 +31,7 @@ def _pad_rows64(t: torch.Tensor) -> torch.Tensor:
     return F.pad(t, (0, 0, 0, pad))
 
 
-def _gemm_nt(ext, a, b, bias=None, relu=False):
+def _gemm_nt(ext, a, b, bias=None, relu=False, add=None):
     """gemm_nt with zero-padding of the reduction dim when K % 64 != 0
     (zero columns contribute nothing to the product)."""
     k = a.shape[1]
@@ -39,17 +39,24 @@ def _gemm_nt(ext, a, b, bias=None, relu=False):
     if pad:
         a = F.pad(a, (0, pad))
         b = F.pad(b, (0, pad))
-    return ext.gemm_nt(a.contiguous(), b.contiguous(), bias, relu)
+    return ext.gemm_nt(a.contiguous(), b.contiguous(), bias, relu,
+                       add=add)
 
 
 class _MFMALinearFn(torch.autograd.Function):
+    """relu and the fused elementwise addend (residual stream) are
+    mutually exclusive here: the epilogue applies add BEFORE relu and
+    no model composes both.  d(add) = dy exactly (identity), so the
+    residual junction costs ZERO kernels in both directions."""
+
     @staticmethod
-    def forward(ctx, x2d, weight, bias, relu):
+    def forward(ctx, x2d, weight, bias, relu, add):
         ext = get_ext(required=True)
-        y = _gemm_nt(ext, x2d, weight, bias, relu)
+        y = _gemm_nt(ext, x2d, weight, bias, relu, add=add)
         ctx.save_for_backward(x2d, weight, y if relu else None)
         ctx.relu = relu
         ctx.has_bias = bias is not None
+        ctx.has_add = add is not None
         return y
 
     @staticmethod
@@ -86,7 +93,7 @@ class _MFMALinearFn(torch.autograd.Function):
             db = None
         xt = ext.transpose2d(xp)                # [K, Mp]
         dw = _gemm_nt(ext, dyt, xt)             # [N, K]
-        return dx, dw, db, None
+        return dx, dw, db, None, (dy if ctx.has_add else None)
 
 
 class MFMALinear(nn.Module):
@@ -106,16 +113,23 @@ class MFMALinear(nn.Module):
         self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
         nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
 
-    def forward(self, x):
+    def forward(self, x, add=None):
+        """``add``: optional residual stream fused into the GEMM
+        epilogue (y = x @ W^T + b + add); its grad is dy (identity)."""
         shape = x.shape
         x2d = x.reshape(-1, shape[-1])
         if x.is_cuda:
             x2d = x2d.contiguous()
-            y = _MFMALinearFn.apply(x2d, self.weight, self.bias, self.relu)
+            add2d = (add.reshape(-1, self.out_features).contiguous()
+                     if add is not None else None)
+            y = _MFMALinearFn.apply(x2d, self.weight, self.bias,
+                                    self.relu, add2d)
         else:
             y = F.linear(x2d, self.weight, self.bias)
             if self.relu:
                 y = F.relu(y)
+            if add is not None:
+                y = y + add.reshape(-1, self.out_features)
         return y.reshape(*shape[:-1], self.out_features)
 
     def extra_repr(self):

@@ -548,3 +548,25 @@ def test_swiglu_packed_fwd_bwd():
     assert mx(y, ref) / (ref.abs().max().item() + 1e-3) < 1e-2
     s = guf.grad.abs().max().item() + 1e-6
     assert (gu.grad.float() - guf.grad).abs().max().item() / s < 2e-2
+
+
+def test_gemm_fused_addend():
+    """y = x @ W^T + add (residual fused into the epilogue), with the
+    addend grad = dy exactly."""
+    from gpuschedule_amd.ops.linear import MFMALinear
+    torch.manual_seed(14)
+    for M, N, K in [(4096, 4096, 4096), (2048, 512, 512)]:
+        lin = MFMALinear(K, N, bias=False).to("cuda", torch.bfloat16)
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        add = torch.randn(M, N, device="cuda", dtype=torch.bfloat16,
+                          requires_grad=True)
+        y = lin(x, add=add)
+        y.float().square().mean().backward()
+        ref = x.detach().float() @ lin.weight.detach().float().t() \
+            + add.detach().float()
+        scale = ref.abs().max().item() + 1e-3
+        assert mx(y, ref) / scale < 1.5e-2
+        # d(add) must be exactly dy = 2*y/numel (identity pass-through)
+        dy = (2 * y.detach().float() / y.numel()).to(torch.bfloat16)
+        assert torch.equal(add.grad, dy)
