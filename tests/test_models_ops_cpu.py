@@ -93,3 +93,23 @@ def test_adamw_cpu_matches_torch_adamw():
             o.step()
     for pa, pb in zip(a.parameters(), b.parameters()):
         assert torch.allclose(pa, pb, atol=1e-4), "AdamW semantics drifted"
+
+
+def test_mfma_linear_fused_addend_cpu_fallback():
+    import torch
+    from gpuschedule_amd.ops.linear import MFMALinear
+    lin = MFMALinear(16, 24, bias=True)
+    x = torch.randn(4, 16)
+    add = torch.randn(4, 24)
+    y = lin(x, add=add)
+    ref = torch.nn.functional.linear(x, lin.weight, lin.bias) + add
+    assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_swiglu_packed_cpu_fallback():
+    import torch
+    from gpuschedule_amd.ops.elementwise import swiglu_packed
+    gu = torch.randn(5, 16)
+    g, u = gu.chunk(2, dim=-1)
+    ref = torch.nn.functional.silu(g) * u
+    assert torch.allclose(swiglu_packed(gu), ref, atol=1e-6)
