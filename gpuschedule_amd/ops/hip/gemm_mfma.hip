@@ -2657,6 +2657,9 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
   if (add.has_value() && add->defined()) {
     TORCH_CHECK(add->scalar_type() == torch::kBFloat16 &&
                 add->is_contiguous() && add->sizes() == c.sizes());
+    TORCH_CHECK(!relu, "gemm_nt: relu and fused addend are exclusive "
+                       "(the fused path applies add BEFORE relu, the "
+                       "fallback after — no caller composes them)");
     add_p = (const short*)add->data_ptr();
   }
   auto finish = [&](torch::Tensor out) {  // fallback for variants

@@ -116,6 +116,8 @@ class MFMALinear(nn.Module):
     def forward(self, x, add=None):
         """``add``: optional residual stream fused into the GEMM
         epilogue (y = x @ W^T + b + add); its grad is dy (identity)."""
+        assert not (self.relu and add is not None), \
+            "MFMALinear: relu and fused addend are exclusive"
         shape = x.shape
         x2d = x.reshape(-1, shape[-1])
         if x.is_cuda:
