@@ -184,8 +184,12 @@ class TorchBackend:
                             0, None, key)
                     self.runners[p.job_id] = runner
                 # every gang member (survivor, joiner, pool-adopted)
-                # rebinds to the SAME new group object
+                # rebinds to the SAME new group object; the pool key
+                # must follow (a later put/adopt under the OLD gang's
+                # key would hand out a runner whose reducer still
+                # all-reduces with the new gang — phantom partners)
                 runner.regang(group, gpus.index(self.rank), len(gpus))
+                runner.pool_key = key
                 src = next(g for g in gpus if g in old)
                 broadcast_state(runner.model, runner.optim, group, src)
                 runner.done_iters = done_iters

@@ -398,3 +398,24 @@ def test_checkpoint_restores_rng_state(tmp_path):
                    state=ck.load("j"))
     got = torch.rand(4)
     assert torch.equal(got, want), "RNG stream not restored"
+
+
+def test_regang_updates_pool_key(tmp_path):
+    """A runner rebound to a resized gang must be pooled under the NEW
+    gang's key — adoption under the old key would hand a single-GPU job
+    a reducer still bound to the multi-rank group."""
+    from gpuschedule_amd.sched.scheduler import Plan, Placement
+    specs = [J("el", iters=100)]
+    be = make_backend(specs, ckpt_root=str(tmp_path / "ck"))
+    be.tick_once()
+    r = be.runners["el"]
+    assert r.pool_key == ("llama3-tiny", 2, (0,))
+    # shrink-style overlap resize back onto [0] exercises the regang
+    plan = Plan(places=[Placement(job_id="el", gpus=[0], resume=True,
+                                  resize_from=2,
+                                  resize_from_gpus=[0, 1])])
+    meta = {"el": ("llama3-tiny", 2, [0], r.done_iters)}
+    be._apply_phase_a(plan, meta)
+    be._apply_phase_b(plan, meta)
+    assert be.runners["el"].pool_key == ("llama3-tiny", 2, (0,))
+    assert be.runners["el"].gang_world == 1
