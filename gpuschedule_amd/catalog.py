@@ -71,3 +71,12 @@ def sim_iter_time(spec, n_gpus: int) -> float:
     if n_gpus <= 1:
         return base
     return base * (1.0 + m.allreduce_frac * (n_gpus - 1) / n_gpus)
+
+
+def job_tokens_per_iter(spec, width: int = 1) -> int:
+    """Tokens (sequence models) or samples (image models) processed by
+    one training iteration of the whole gang: per-GPU batch x tokens
+    per sample x DP width.  Drives the per-job throughput metric in
+    the event log."""
+    from .models.registry import SAMPLE_TOKENS
+    return job_batch(spec) * SAMPLE_TOKENS.get(spec.model, 1) * width

@@ -16,7 +16,7 @@ Event kinds and extra fields:
 * ``resume``    — gpus
 * ``migrate``   — src, dst (lists)
 * ``resize``    — old_gpus, new_gpus (GPU counts before/after)
-* ``complete``  — jct, iters
+* ``complete``  — jct, iters, tokens, tokens_per_s (tokens for sequence models, samples for image models; whole-gang)
 * ``fail``      — error
 * ``tick``      — util, running, queued (cluster samples; optional)
 
@@ -168,6 +168,9 @@ class EventLog:
             "p95_jct": pct(0.95),
             "makespan": makespan,
             "avg_queue_delay": sum(qdelays) / len(qdelays) if qdelays else 0.0,
+            "tokens_per_s_sum": round(sum(
+                e.get("tokens_per_s", 0.0)
+                for e in completes.values()), 1),
             "preemptions": float(sum(1 for e in self.events
                                      if e["event"] == "preempt")),
             "migrations": float(sum(1 for e in self.events

@@ -396,6 +396,8 @@ class Scheduler:
         job = self.jobs[job_id]
         job.done_iters += iters
         job.attained_service += service
+        job.tokens_done += iters * catalog.job_tokens_per_iter(
+            job.spec, max(len(job.placement), 1))
         if iter_time:
             job.iter_time = iter_time
         if job.done_iters >= job.spec.total_iters:
@@ -404,7 +406,10 @@ class Scheduler:
             job.finish_t = now
             job.placement = []
             self.log.emit(now, "complete", job_id, jct=job.jct,
-                          iters=job.done_iters)
+                          iters=job.done_iters,
+                          tokens=int(job.tokens_done),
+                          tokens_per_s=round(
+                              job.tokens_done / max(job.jct, 1e-9), 1))
             return True
         return False
 

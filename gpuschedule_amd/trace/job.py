@@ -77,6 +77,7 @@ class Job:
     placement: List[int] = field(default_factory=list)  # global GPU ids
     done_iters: int = 0
     attained_service: float = 0.0   # GPU-seconds of service received
+    tokens_done: float = 0.0        # tokens/samples processed (all GPUs)
     queue_level: int = 0            # LAS/Tiresias MLFQ level
     submit_t: float = 0.0
     first_start_t: Optional[float] = None

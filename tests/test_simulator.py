@@ -126,3 +126,16 @@ def test_unsatisfiable_job_fails_at_admission():
     fails = [e for e in log.events if e["event"] == "fail"]
     assert len(fails) == 1 and fails[0]["job"] == "wide"
     assert int(s["n_complete"]) == 1
+
+
+def test_complete_event_reports_tokens():
+    from gpuschedule_amd import JobSpec, make_policy
+    from gpuschedule_amd.sim.simulator import simulate
+    specs = [JobSpec(job_id="a", submit_time=0.0, n_gpus=2,
+                     total_iters=10, model="bert-base", batch_size=4)]
+    log, s = simulate(specs, make_policy("fifo"), n_gpus=2)
+    done = [e for e in log.events if e["event"] == "complete"][0]
+    # 10 iters x batch 4 x 128 tokens x 2 GPUs
+    assert done["tokens"] == 10 * 4 * 128 * 2
+    assert done["tokens_per_s"] > 0
+    assert s["tokens_per_s_sum"] > 0
