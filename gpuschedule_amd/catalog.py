@@ -73,10 +73,16 @@ def sim_iter_time(spec, n_gpus: int) -> float:
     return base * (1.0 + m.allreduce_frac * (n_gpus - 1) / n_gpus)
 
 
+# Tokens processed per sample per iteration (sequence models) or
+# samples (image models count 1).  Torch-free here; models/registry
+# re-exports it.
+SAMPLE_TOKENS = {"resnet18": 1, "resnet50": 1, "bert-base": 128,
+                 "llama3-8b": 2048, "llama3-tiny": 256}
+
+
 def job_tokens_per_iter(spec, width: int = 1) -> int:
     """Tokens (sequence models) or samples (image models) processed by
     one training iteration of the whole gang: per-GPU batch x tokens
     per sample x DP width.  Drives the per-job throughput metric in
     the event log."""
-    from .models.registry import SAMPLE_TOKENS
     return job_batch(spec) * SAMPLE_TOKENS.get(spec.model, 1) * width

@@ -57,9 +57,8 @@ _REGISTRY = {
 }
 
 # Tokens processed per sample per iteration (for tokens/s metrics);
-# image models count images.
-SAMPLE_TOKENS = {"resnet18": 1, "resnet50": 1, "bert-base": 128,
-                 "llama3-8b": 2048, "llama3-tiny": 256}
+# image models count images.  Defined torch-free in catalog.py.
+SAMPLE_TOKENS = catalog.SAMPLE_TOKENS
 
 
 def build_model(name: str, device="cpu",
