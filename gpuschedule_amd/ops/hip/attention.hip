@@ -557,6 +557,234 @@ __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_kernel(
   }
 }
 
+// Pipelined dK/dV backward for D=128.  The base kernel's LDS image
+// (Q,dO,Qt,dOt + per-wave P AND dS) is 96.5 KiB -> ONE block/CU, ONE
+// wave/SIMD: nothing hides the per-iteration cold vmcnt(0) drain of
+// tiles staged immediately before the barrier.  This variant:
+//  * aliases dS onto the P buffer (write P -> consume P for dV ->
+//    write dS over it -> consume for dK; wave-private, lgkm-ordered),
+//  * double-buffers the staged set with the freed 16 KiB (145 KiB
+//    total, still 1 block/CU),
+//  * flattens the (q-head group x q-tile) loops and issues each
+//    iteration's DMA a FULL iteration early — the one __syncthreads
+//    per iteration then drains loads that are ~an iteration old.
+template <int D, bool CAUSAL, bool STRIDED>
+__global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_kv_pipe_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const short* __restrict__ Qt, const short* __restrict__ dOt,
+    const float* __restrict__ LSE, const float* __restrict__ DR,
+    short* __restrict__ dK, short* __restrict__ dV,
+    int B, int H, int HK, int S, float scale,
+    long qsB, long qsH, long qsS_, long osB, long osH, long osS_) {
+  const long qsS = STRIDED ? qsS_ : (long)D;
+  const long osS = STRIDED ? osS_ : (long)D;
+  constexpr int DF = D / 16;
+  constexpr int KC = D / 32;
+  constexpr int RB = D * 2;
+  constexpr int STG = 2 * KVB * D * 2 + 2 * D * KVB * 2 + 2 * KVB * 4;
+  __shared__ __attribute__((aligned(16))) char lds[
+      2 * STG + 4 * 16 * KVB * 2];
+  char* p_base = lds + 2 * STG;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  char* my_p = p_base + wid * 16 * KVB * 2;  // P, then reused for dS
+
+  const int n_kvt = (S + KVB - 1) / KVB;
+  const int bh = blockIdx.x / n_kvt;
+  const int kvt = blockIdx.x % n_kvt;
+  const int b = bh / HK, kvh = bh % HK;
+  const int rep = H / HK;
+  const int kv0 = kvt * KVB;
+  const int krow = kv0 + wid * 16 + (lane & 15);
+
+  const long kvb = STRIDED ? ((long)b * qsB + (long)kvh * qsH)
+      : ((long)(b * HK + kvh) * S) * (long)D;
+
+  bf16x8 kreg[KC], vreg[KC];
+  {
+    int row = krow < S ? krow : S - 1;
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      int d0 = c * 32 + (lane >> 4) * 8;
+      bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
+          K + kvb + (long)row * qsS + d0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        kv8[j] = (__bf16)((float)kv8[j] * scale);
+      kreg[c] = kv8;
+      vreg[c] = *reinterpret_cast<const bf16x8*>(
+          V + kvb + (long)row * qsS + d0);
+    }
+  }
+
+  float4v dv_acc[DF], dk_acc[DF];
+#pragma unroll
+  for (int f = 0; f < DF; ++f) {
+    dv_acc[f] = {0.f, 0.f, 0.f, 0.f};
+    dk_acc[f] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int qt_start = CAUSAL ? kv0 / 64 : 0;
+  const int n_qt = (S + 63) / 64;
+  const int span = n_qt - qt_start;
+  const int total = rep * span;
+
+  // stage iteration `it`'s tiles + lse/dr into buffer `buf`.  The
+  // lse/dr fill runs with NO glds outstanding (callers order it right
+  // after the barrier), so hipcc's ordinary-load waits stay counted.
+  auto fill_lse = [&](int it, char* buf) {
+    const int g = it / span;
+    const int qt = qt_start + it % span;
+    const int h = kvh * rep + g;
+    const long roff = (long)(b * H + h) * S;
+    const int q0 = qt * 64;
+    float* lse_b = (float*)(buf + 2 * KVB * D * 2 + 2 * D * KVB * 2);
+    float* dr_b = lse_b + KVB;
+    for (int i = tid; i < KVB; i += NTHREADS) {
+      int qr = q0 + i;
+      lse_b[i] = qr < S ? LSE[roff + qr] : 3e30f;
+      dr_b[i] = qr < S ? DR[roff + qr] : 0.f;
+    }
+  };
+  auto stage = [&](int it, char* buf) {
+    const int g = it / span;
+    const int qt = qt_start + it % span;
+    const int h = kvh * rep + g;
+    const long qvb = (long)b * qsB + (long)h * qsH;
+    const long dob = (long)b * osB + (long)h * osH;
+    const long offT = ((long)(b * H + h) * D) * S;
+    const int q0 = qt * 64;
+    char* q_b = buf;
+    char* do_b = q_b + KVB * D * 2;
+    char* qt_b = do_b + KVB * D * 2;
+    char* dot_b = qt_b + D * KVB * 2;
+    glds_tile<KVB, RB>(Q + qvb, qsS, q0, S, 0, D, q_b, wid, lane);
+    glds_tile<KVB, RB>(dO + dob, osS, q0, S, 0, D, do_b, wid, lane);
+    glds_tile<D, KVB * 2>(Qt + offT, S, 0, D, q0, S, qt_b, wid, lane);
+    glds_tile<D, KVB * 2>(dOt + offT, S, 0, D, q0, S, dot_b, wid, lane);
+  };
+
+  fill_lse(0, lds);
+  stage(0, lds);
+
+  for (int it = 0; it < total; ++it) {
+    char* cur = lds + (it & 1) * STG;
+    char* nxt = lds + ((it + 1) & 1) * STG;
+    // drains cur's DMA (issued a full iteration ago) + separates the
+    // previous iteration's reads of nxt's buffer from the overwrite
+    __syncthreads();
+    if (it + 1 < total) {
+      fill_lse(it + 1, nxt);
+      stage(it + 1, nxt);
+    }
+    char* q_lds = cur;
+    char* do_lds = q_lds + KVB * D * 2;
+    char* qt_lds = do_lds + KVB * D * 2;
+    char* dot_lds = qt_lds + D * KVB * 2;
+    float* lse_lds = (float*)(dot_lds + D * KVB * 2);
+    float* dr_lds = lse_lds + KVB;
+    const int g = it / span;
+    const int qt = qt_start + it % span;
+    const int q0 = qt * 64;
+    (void)g;
+
+    float4v st[4], dpt[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      st[f] = {0.f, 0.f, 0.f, 0.f};
+      dpt[f] = {0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int kk = 0; kk < KC; ++kk) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+            q_lds + swz(f * 16 + (lane & 15),
+                        kk * 64 + (lane >> 4) * 16, RB));
+        st[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            kreg[kk], qf, st[f], 0, 0, 0);
+        bf16x8 dof = *reinterpret_cast<const bf16x8*>(
+            do_lds + swz(f * 16 + (lane & 15),
+                         kk * 64 + (lane >> 4) * 16, RB));
+        dpt[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            vreg[kk], dof, dpt[f], 0, 0, 0);
+      }
+    }
+    const int krow0 = kv0 + wid * 16 + (lane >> 4) * 4;
+    // P^T -> wave-private LDS, dV += P^T dO
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int qcol = q0 + f * 16 + (lane & 15);
+      float lse_q = lse_lds[f * 16 + (lane & 15)];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        bool dead = qcol >= S || lse_q > 1e30f;
+        if (CAUSAL) dead |= qcol < krow0 + r;
+        float p = dead ? 0.f : __expf(st[f][r] - lse_q);
+        st[f][r] = p;  // keep P for the dS pass below
+        *reinterpret_cast<short*>(
+            my_p + swz((lane >> 4) * 4 + r, (f * 16 + (lane & 15)) * 2,
+                       KVB * 2)) = f32_to_bf16(p);
+      }
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 pf = *reinterpret_cast<const bf16x8*>(
+          my_p + swz(lane & 15, kk * 64 + (lane >> 4) * 16, KVB * 2));
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        bf16x8 dotf = *reinterpret_cast<const bf16x8*>(
+            dot_lds + swz(f * 16 + (lane & 15),
+                          kk * 64 + (lane >> 4) * 16, KVB * 2));
+        dv_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pf, dotf, dv_acc[f], 0, 0, 0);
+      }
+    }
+    // dS^T over the SAME wave-private buffer, dK += dS^T Q
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      float dr_q = dr_lds[f * 16 + (lane & 15)];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float ds = st[f][r] * (dpt[f][r] - dr_q) * scale;
+        *reinterpret_cast<short*>(
+            my_p + swz((lane >> 4) * 4 + r, (f * 16 + (lane & 15)) * 2,
+                       KVB * 2)) = f32_to_bf16(ds);
+      }
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 dsf = *reinterpret_cast<const bf16x8*>(
+          my_p + swz(lane & 15, kk * 64 + (lane >> 4) * 16, KVB * 2));
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        bf16x8 qtf = *reinterpret_cast<const bf16x8*>(
+            qt_lds + swz(f * 16 + (lane & 15),
+                         kk * 64 + (lane >> 4) * 16, KVB * 2));
+        dk_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf, qtf, dk_acc[f], 0, 0, 0);
+      }
+    }
+  }
+
+  const int orow0 = kv0 + wid * 16 + (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = orow0 + r;
+    if (row >= S) continue;
+    short* dkr = dK + kvb + (long)row * qsS;
+    short* dvr = dV + kvb + (long)row * qsS;
+#pragma unroll
+    for (int f = 0; f < DF; ++f) {
+      dkr[f * 16 + (lane & 15)] = f32_to_bf16(dk_acc[f][r]);
+      dvr[f * 16 + (lane & 15)] = f32_to_bf16(dv_acc[f][r]);
+    }
+  }
+}
+
 template <int D, bool CAUSAL, bool STRIDED>
 __global__ __launch_bounds__(NTHREADS, 2) void attn_bwd_q_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
@@ -853,21 +1081,50 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   float scale = 1.0f / sqrtf((float)D);
   int grid_kv = B * HK * ((S + KVB - 1) / KVB);
   int grid_q = B * H * ((S + 63) / 64);
+  static const bool pipe = [] {
+    // D=128 pipelined dK/dV variant (double-buffered staging, dS
+    // aliased onto P, one barrier/iter).  MEASURED: wash vs the base
+    // kernel (llama-8B 183.8 vs 183.1 ms/iter) — at 1 wave/SIMD the
+    // iteration is not drain-bound.  Kept gated.
+    const char* e = getenv("GSAMD_ATTN_PIPE");
+    return e && e[0] == '1';
+  }();
+#define LAUNCH_KV(DD, CC)                                               \
+  do {                                                                  \
+    if (DD == 128 && pipe)                                              \
+      hipLaunchKernelGGL((attn_bwd_kv_pipe_kernel<DD, CC, SR>),         \
+                         dim3(grid_kv), dim3(NTHREADS), 0,              \
+                         stream.stream(),                               \
+                         (const short*)q.data_ptr(),                    \
+                         (const short*)k.data_ptr(),                    \
+                         (const short*)v.data_ptr(),                    \
+                         (const short*)dout.data_ptr(),                 \
+                         (const short*)qt.data_ptr(),                   \
+                         (const short*)dot.data_ptr(),                  \
+                         lse.data_ptr<float>(), drow.data_ptr<float>(), \
+                         (short*)dk.data_ptr(), (short*)dv.data_ptr(),  \
+                         B, H, HK, S, scale, q.stride(0), q.stride(1),  \
+                         q.stride(2), o.stride(0), o.stride(1),         \
+                         o.stride(2));                                  \
+    else                                                                \
+      hipLaunchKernelGGL((attn_bwd_kv_kernel<DD, CC, SR>),              \
+                         dim3(grid_kv), dim3(NTHREADS), 0,              \
+                         stream.stream(),                               \
+                         (const short*)q.data_ptr(),                    \
+                         (const short*)k.data_ptr(),                    \
+                         (const short*)v.data_ptr(),                    \
+                         (const short*)dout.data_ptr(),                 \
+                         (const short*)qt.data_ptr(),                   \
+                         (const short*)dot.data_ptr(),                  \
+                         lse.data_ptr<float>(), drow.data_ptr<float>(), \
+                         (short*)dk.data_ptr(), (short*)dv.data_ptr(),  \
+                         B, H, HK, S, scale, q.stride(0), q.stride(1),  \
+                         q.stride(2), o.stride(0), o.stride(1),         \
+                         o.stride(2));                                  \
+  } while (0)
 #define LAUNCH_BWD(DD, CC)                                              \
   do {                                                                  \
-    hipLaunchKernelGGL((attn_bwd_kv_kernel<DD, CC, SR>), dim3(grid_kv), \
-                       dim3(NTHREADS), 0, stream.stream(),              \
-                       (const short*)q.data_ptr(),                      \
-                       (const short*)k.data_ptr(),                      \
-                       (const short*)v.data_ptr(),                      \
-                       (const short*)dout.data_ptr(),                   \
-                       (const short*)qt.data_ptr(),                     \
-                       (const short*)dot.data_ptr(),                    \
-                       lse.data_ptr<float>(), drow.data_ptr<float>(),   \
-                       (short*)dk.data_ptr(), (short*)dv.data_ptr(),    \
-                       B, H, HK, S, scale, q.stride(0), q.stride(1),    \
-                       q.stride(2), o.stride(0), o.stride(1),           \
-                       o.stride(2));                                    \
+    LAUNCH_KV(DD, CC);                                                  \
     hipLaunchKernelGGL((attn_bwd_q_kernel<DD, CC, SR>), dim3(grid_q),   \
                        dim3(NTHREADS), 0, stream.stream(),              \
                        (const short*)q.data_ptr(),                      \
