@@ -25,7 +25,7 @@ def fused_rope(x: torch.Tensor, cosb: torch.Tensor,
                sinb: torch.Tensor) -> torch.Tensor:
     """Rotate-half RoPE on [B, H, S, D] bf16 with fp32 [S, D/2] tables."""
     if x.is_cuda and x.dtype == torch.bfloat16:
-        return _RopeFn.apply(x.contiguous(), cosb, sinb)
+        return _RopeFn.apply(x, cosb, sinb)
     S = x.shape[-2]
     c = cosb[:S].to(x.dtype)[None, None]
     s = sinb[:S].to(x.dtype)[None, None]

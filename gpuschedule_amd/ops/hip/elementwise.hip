@@ -15,11 +15,16 @@ constexpr int NT = 256;
 
 // x [rows, D] where row r maps to sequence position pos = r % S
 // (rows are [B,H,S] flattened).  cos/sin: [S, D/2] fp32.
-template <bool BWD>
+// STRIDED: the input may be a [B,H,S,D] VIEW (e.g. a head slice of
+// the packed qkv projection transposed to BHSD) read through a
+// (b,h,s) stride triple — no .contiguous() materialization; the
+// OUTPUT is always written contiguous [B,H,S,D] (what the attention
+// kernels consume).
+template <bool BWD, bool STRIDED>
 __global__ __launch_bounds__(NT) void rope_kernel(
     const short* __restrict__ x, const float* __restrict__ cosb,
     const float* __restrict__ sinb, short* __restrict__ out,
-    long rows, int S, int D) {
+    long rows, int S, int D, int Hh, long sB, long sH, long sS) {
   const int HD = D / 2;
   const int chunks = HD / 8;  // 8-wide pieces per half-row
   long total = rows * chunks;
@@ -29,8 +34,15 @@ __global__ __launch_bounds__(NT) void rope_kernel(
     long row = i / chunks;
     int c8 = (int)(i % chunks) * 8;
     int pos = (int)(row % S);
-    const short* xr = x + row * D;
-    short* orow = out + row * D;
+    long xoff;
+    if (STRIDED) {
+      long bh = row / S;
+      xoff = (bh / Hh) * sB + (bh % Hh) * sH + (long)pos * sS;
+    } else {
+      xoff = row * D;
+    }
+    const short* xr = x + xoff;
+    short* orow = out + row * D;  // output always contiguous
     short8 x1 = *reinterpret_cast<const short8*>(xr + c8);
     short8 x2 = *reinterpret_cast<const short8*>(xr + HD + c8);
     const float* cr = cosb + (long)pos * HD + c8;
@@ -211,30 +223,44 @@ int grid_for_n(long items) {
 
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosb,
                          torch::Tensor sinb, bool backward) {
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
-              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(cosb.scalar_type() == torch::kFloat32);
   const int D = x.size(-1);
   const int S = x.size(-2);
   TORCH_CHECK(D % 16 == 0, "rope: D must be a multiple of 16");
   TORCH_CHECK(cosb.size(-1) == D / 2 && cosb.size(0) >= S);
   long rows = x.numel() / D;
-  auto out = torch::empty_like(x);
+  const bool cont = x.is_contiguous();
+  int Hh = 1;
+  long sB = 0, sH = 0, sS = 0;
+  if (!cont) {
+    TORCH_CHECK(x.dim() == 4 && x.stride(3) == 1 &&
+                x.stride(2) % 8 == 0,
+                "rope: strided input must be a [B,H,S,D] view with "
+                "contiguous 16B-aligned D rows");
+    Hh = x.size(1);
+    sB = x.stride(0);
+    sH = x.stride(1);
+    sS = x.stride(2);
+  }
+  auto out = cont ? torch::empty_like(x)
+                  : torch::empty(x.sizes(), x.options());
   long total = rows * (D / 16);
   auto stream = c10::cuda::getCurrentCUDAStream();
+#define ROPE_LAUNCH(BWD, SR)                                            \
+  hipLaunchKernelGGL((rope_kernel<BWD, SR>), dim3(grid_for_n(total)),   \
+                     dim3(NT), 0, stream.stream(),                      \
+                     (const short*)x.data_ptr(),                        \
+                     cosb.data_ptr<float>(), sinb.data_ptr<float>(),    \
+                     (short*)out.data_ptr(), rows, S, D, Hh, sB, sH, sS)
   if (backward) {
-    hipLaunchKernelGGL(rope_kernel<true>, dim3(grid_for_n(total)),
-                       dim3(NT), 0, stream.stream(),
-                       (const short*)x.data_ptr(),
-                       cosb.data_ptr<float>(), sinb.data_ptr<float>(),
-                       (short*)out.data_ptr(), rows, S, D);
+    if (cont) ROPE_LAUNCH(true, false);
+    else ROPE_LAUNCH(true, true);
   } else {
-    hipLaunchKernelGGL(rope_kernel<false>, dim3(grid_for_n(total)),
-                       dim3(NT), 0, stream.stream(),
-                       (const short*)x.data_ptr(),
-                       cosb.data_ptr<float>(), sinb.data_ptr<float>(),
-                       (short*)out.data_ptr(), rows, S, D);
+    if (cont) ROPE_LAUNCH(false, false);
+    else ROPE_LAUNCH(false, true);
   }
+#undef ROPE_LAUNCH
   HIP_CHECK_LAST();
   return out;
 }

@@ -570,3 +570,21 @@ def test_gemm_fused_addend():
         # d(add) must be exactly dy = 2*y/numel (identity pass-through)
         dy = (2 * y.detach().float() / y.numel()).to(torch.bfloat16)
         assert torch.equal(add.grad, dy)
+
+
+def test_rope_strided_view_matches_contiguous():
+    """RoPE reads the packed projection's transposed head-slice view
+    directly (no .contiguous() materialization)."""
+    from gpuschedule_amd.ops.elementwise import fused_rope
+    from gpuschedule_amd.models.llama import LLAMA3_TINY, rope_table
+    torch.manual_seed(15)
+    cfg = LLAMA3_TINY
+    cos, sin = rope_table(cfg, "cuda")
+    B, S, H, Dh = 2, 256, cfg.heads, cfg.head_dim
+    packed = torch.randn(B, S, H * Dh, device="cuda",
+                         dtype=torch.bfloat16)
+    view = packed.view(B, S, H, Dh).transpose(1, 2)   # strided BHSD
+    assert not view.is_contiguous()
+    y_str = fused_rope(view, cos, sin)
+    y_cont = fused_rope(view.contiguous(), cos, sin)
+    assert torch.equal(y_str, y_cont)
