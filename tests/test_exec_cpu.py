@@ -419,3 +419,59 @@ def test_regang_updates_pool_key(tmp_path):
     be._apply_phase_b(plan, meta)
     assert be.runners["el"].pool_key == ("llama3-tiny", 2, (0,))
     assert be.runners["el"].gang_world == 1
+
+
+def _gang_prewarm_worker(rank, world, port, out_file):
+    import json
+    import torch.distributed as dist
+    from gpuschedule_amd.exec.trainer import RUNNER_POOL
+    torch.set_num_threads(2)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        # batch unset (0) resolves to the model default, matching the
+        # prewarm key — exactly how bench.py's fill specs are built
+        specs = ([J("gang", n=2, iters=4, batch=0)]
+                 if rank == 0 else None)
+        be = make_backend(specs, "fifo", rank=rank, world=world,
+                          ckpt_root=f"/tmp/gsamd_gpw_{port}")
+        # what bench.py does before its warm ticks at N>=2 (the
+        # collective prewarm of the 2-GPU DDP gang fill runner)
+        be.prewarm_gang("llama3-tiny", [0, 1], iters=2)
+        key = ("llama3-tiny", 8, (0, 1))
+        pooled = RUNNER_POOL._pool.get(key)
+        assert pooled, f"gang runner not pooled on rank {rank}"
+        # the gang job scheduled next must ADOPT it
+        ran = {}
+        orig = be._apply_phase_b
+
+        def spy(plan, meta):
+            orig(plan, meta)
+            for p in plan.places:
+                if p.job_id in be.runners:
+                    ran[p.job_id] = id(be.runners[p.job_id])
+        be._apply_phase_b = spy
+        pooled_id = id(pooled[0])
+        log = be.run(max_seconds=120)
+        assert ran.get("gang") == pooled_id, "gang job did not adopt"
+        if rank == 0:
+            with open(out_file, "w") as f:
+                json.dump(log.summary(), f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_gang_prewarm_and_adopt(tmp_path):
+    """bench.py's N>=2 gang prewarm: collective build+step outside the
+    tick protocol, pooled symmetrically, adopted by the scheduled gang
+    job on every rank (batch=0 resolves to the model default = 8,
+    matching catalog.job_batch for a spec without batch_size)."""
+    import torch.multiprocessing as mp
+    import json
+    out = str(tmp_path / "o.json")
+    mp.spawn(_gang_prewarm_worker, args=(2, 29331, out), nprocs=2,
+             join=True)
+    with open(out) as f:
+        s = json.load(f)
+    assert s["n_complete"] == 1
