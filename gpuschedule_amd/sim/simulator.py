@@ -32,9 +32,17 @@ XGMI_BYTES_PER_S = 150e9  # per-link broadcast (overlap-resize path)
 
 class Simulator:
     def __init__(self, scheduler: Scheduler, reeval_s: float = 1.0,
-                 max_time: float = 10_000_000.0):
+                 max_time: float = 10_000_000.0,
+                 warm_pool: bool = False):
         self.sched = scheduler
         self.reeval_s = reeval_s
+        # warm_pool models the real backend's prewarmed runner pool
+        # (exec/trainer.py RunnerPool): a job whose model family has a
+        # pooled runner adopts it in-place instead of paying model
+        # construction + hipGraph capture + MIOpen find, so first
+        # placement costs ~RESUME_HBM_S, not COLD_START_S.  Measured:
+        # scripts/sim_fidelity.py (profiles/sim_fidelity.json).
+        self.warm_pool = warm_pool
         self.max_time = max_time
         self.now = 0.0
         self._frac: Dict[str, float] = {}      # fractional iterations
@@ -57,7 +65,7 @@ class Simulator:
             return ckpt / H2D_BYTES_PER_S
         if p.resume:
             return RESUME_HBM_S
-        return COLD_START_S
+        return RESUME_HBM_S if self.warm_pool else COLD_START_S
 
     def _apply_plan_overheads(self, plan: Plan) -> None:
         for p in plan.places:
@@ -146,11 +154,12 @@ class Simulator:
 
 
 def simulate(specs, policy, n_gpus: int = 8, gpus_per_node: int = 8,
-             reeval_s: float = 1.0, log: Optional[EventLog] = None):
+             reeval_s: float = 1.0, log: Optional[EventLog] = None,
+             warm_pool: bool = False):
     """Convenience: simulate a trace under a policy; returns (log, summary)."""
     from ..sched.cluster import Cluster
     sched = Scheduler(Cluster(n_gpus, gpus_per_node), policy, log=log)
     sched.submit(list(specs))
-    sim = Simulator(sched, reeval_s=reeval_s)
+    sim = Simulator(sched, reeval_s=reeval_s, warm_pool=warm_pool)
     out = sim.run()
     return out, out.summary()

@@ -139,3 +139,15 @@ def test_complete_event_reports_tokens():
     assert done["tokens"] == 10 * 4 * 128 * 2
     assert done["tokens_per_s"] > 0
     assert s["tokens_per_s_sum"] > 0
+
+
+def test_warm_pool_skips_cold_start():
+    # warm_pool models the backend's prewarmed RunnerPool: first
+    # placement costs RESUME_HBM_S (0.05s) instead of COLD_START_S
+    # (0.5s); a 10x1s job's JCT drops by ~0.45s and nothing else moves.
+    _, cold = simulate([J("a", iters=10)], make_policy("fifo"), n_gpus=1)
+    _, warm = simulate([J("a", iters=10)], make_policy("fifo"), n_gpus=1,
+                       warm_pool=True)
+    assert abs(cold["avg_jct"] - 10.5) < 0.3
+    assert abs(warm["avg_jct"] - 10.05) < 0.3
+    assert warm["avg_jct"] < cold["avg_jct"] - 0.3
