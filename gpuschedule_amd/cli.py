@@ -57,7 +57,8 @@ def cmd_simulate(args):
     pol = _policy_from_args(args, cfg)
     specs = _specs_from_args(args, cfg)
     n_gpus = cfg.get("gpus", args.gpus)
-    log, summary = simulate(specs, pol, n_gpus=n_gpus)
+    warm = cfg.get("warm_pool", getattr(args, "warm_pool", False))
+    log, summary = simulate(specs, pol, n_gpus=n_gpus, warm_pool=warm)
     if args.log:
         log.save(args.log)
     print(json.dumps({"policy": pol.name, "gpus": n_gpus,
@@ -149,6 +150,10 @@ def main(argv=None):
     common.add_argument("--log", default=None, help="event log JSONL out")
 
     p = sub.add_parser("simulate", parents=[common])
+    p.add_argument("--warm-pool", action="store_true",
+                   help="model the backend's prewarmed runner pool: "
+                        "first placements cost RESUME_HBM_S, not "
+                        "COLD_START_S (see sim/simulator.py)")
     p.set_defaults(fn=cmd_simulate)
     p = sub.add_parser("replay", parents=[common])
     p.add_argument("--max-seconds", type=float, default=None)

@@ -99,3 +99,15 @@ def test_bench_trace_identical_across_n():
     assert len(f1) == 1 and all(s.background for s in f1)
     # at N>=2 the first fill is a 2-GPU gang (exercises RCCL all-reduce)
     assert f8[0].n_gpus == 2 and sum(s.n_gpus for s in f8) == 8
+
+
+def test_simulate_warm_pool_flag():
+    # --warm-pool models the backend's prewarmed RunnerPool: cheaper
+    # first placements => strictly lower avg JCT on the same trace
+    cold = json.loads(run_cli("simulate", "--policy", "fifo",
+                              "--jobs", "6", "--gpus", "2").strip())
+    warm = json.loads(run_cli("simulate", "--policy", "fifo",
+                              "--jobs", "6", "--gpus", "2",
+                              "--warm-pool").strip())
+    assert warm["avg_jct"] < cold["avg_jct"]
+    assert warm["n_complete"] == cold["n_complete"]
