@@ -136,3 +136,33 @@ def test_gandiva_packing_on_gpu(tmp_path):
         elif e["event"] in ("complete", "preempt", "fail"):
             running.discard(e.get("job"))
     assert overlap
+
+
+def test_sim_predicts_real_backend(tmp_path):
+    """Simulator fidelity: the calibrated warm-pool simulator must
+    predict the real backend's avg JCT and makespan within 2x on a
+    small SRTF trace (measured agreement is ~1.1x on the 10-job trace,
+    profiles/sim_fidelity.json; the loose band absorbs box variance)."""
+    from gpuschedule_amd.sim.simulator import simulate
+
+    def trace():
+        return [JobSpec(job_id=f"f{i}", submit_time=0.3 * i, n_gpus=1,
+                        model=("resnet18" if i % 2 == 0 else "bert-base"),
+                        total_iters=(120 if i % 2 == 0 else 48))
+                for i in range(6)]
+
+    _, sim = simulate(trace(), make_policy("srtf"), n_gpus=1,
+                      reeval_s=0.05, warm_pool=True)
+    sched = Scheduler(Cluster(1, 1), make_policy("srtf"))
+    sched.submit(trace())
+    be = TorchBackend(sched, rank=0, world=1, device_type="cuda",
+                      iters_per_tick=10,
+                      ckpt_root=str(tmp_path / "ck"))
+    be.prewarm(["resnet18", "bert-base"], count=1)
+    torch.cuda.synchronize()
+    real = be.run(max_seconds=300).summary()
+    assert real["n_complete"] == 6
+    for k in ("avg_jct", "makespan"):
+        ratio = real[k] / sim[k]
+        assert 0.5 < ratio < 2.0, f"{k}: real {real[k]:.2f} vs " \
+                                  f"sim {sim[k]:.2f} (ratio {ratio:.2f})"
