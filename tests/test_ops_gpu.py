@@ -127,7 +127,10 @@ def test_transpose2d():
         assert torch.equal(ext().transpose2d(x), x.t().contiguous())
 
 
-def test_mfma_linear_autograd():
+def test_mfma_linear_autograd(monkeypatch):
+    # pin the hand-written MFMA path (production default is hipBLASLt
+    # for plain GEMMs, GSAMD_GEMM_LIB=1 — this test covers the kernel)
+    monkeypatch.setenv("GSAMD_GEMM_LIB", "0")
     from gpuschedule_amd.ops import MFMALinear
     torch.manual_seed(5)
     lin = MFMALinear(768, 512).cuda().bfloat16()
@@ -550,9 +553,10 @@ def test_swiglu_packed_fwd_bwd():
     assert (gu.grad.float() - guf.grad).abs().max().item() / s < 2e-2
 
 
-def test_gemm_fused_addend():
+def test_gemm_fused_addend(monkeypatch):
     """y = x @ W^T + add (residual fused into the epilogue), with the
     addend grad = dy exactly."""
+    monkeypatch.setenv("GSAMD_GEMM_LIB", "0")  # cover the kernel path
     from gpuschedule_amd.ops.linear import MFMALinear
     torch.manual_seed(14)
     for M, N, K in [(4096, 4096, 4096), (2048, 512, 512)]:
