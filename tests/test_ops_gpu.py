@@ -592,3 +592,24 @@ def test_rope_strided_view_matches_contiguous():
     y_str = fused_rope(view, cos, sin)
     y_cont = fused_rope(view.contiguous(), cos, sin)
     assert torch.equal(y_str, y_cont)
+
+
+def test_mfma_linear_lib_and_kernel_paths_agree(monkeypatch):
+    """The two MFMALinear dispatches (hand-written MFMA kernel vs
+    hipBLASLt, GSAMD_GEMM_LIB) must stay numerically interchangeable:
+    same bf16 inputs, outputs within bf16 rounding of each other."""
+    from gpuschedule_amd.ops.linear import MFMALinear
+    torch.manual_seed(21)
+    lin = MFMALinear(512, 384).to("cuda", torch.bfloat16)
+    x = torch.randn(640, 512, device="cuda", dtype=torch.bfloat16)
+    add = torch.randn(640, 384, device="cuda", dtype=torch.bfloat16)
+    outs = {}
+    for mode in ("0", "1"):
+        monkeypatch.setenv("GSAMD_GEMM_LIB", mode)
+        outs[mode] = lin(x, add=add).float()
+    ref = (x.float() @ lin.weight.float().t() + lin.bias.float()
+           + add.float())
+    scale = ref.abs().max().item()
+    assert (outs["0"] - outs["1"]).abs().max().item() / scale < 1.5e-2
+    for mode in outs:
+        assert (outs[mode] - ref).abs().max().item() / scale < 1.5e-2
