@@ -1,16 +1,20 @@
-"""MFMA-backed Linear layer.
+"""Linear layer with dual GEMM dispatch (see _use_lib):
 
-Forward: C = x @ W^T (+bias) on the hand-written bf16 MFMA GEMM
-(ops/hip/gemm_mfma.hip, NT layout).  Backward reuses the same NT kernel
-with fast LDS-tiled transposes:
+* GSAMD_GEMM_LIB=1 (default): plain projection GEMMs via hipBLASLt
+  (F.linear) — native NN/TN backward, measured 10-13% faster end to
+  end (profiles/gemm_mode_compare.json).
+* GSAMD_GEMM_LIB=0: the hand-written bf16 MFMA GEMM
+  (ops/hip/gemm_mfma.hip, NT layout) with the fused
+  bias+residual-addend epilogue.  Backward reuses the same NT kernel
+  with fast LDS-tiled transposes:
 
-    dx[M,K] = dy[M,N] @ W[N,K]        = gemm_nt(dy, W^T)
-    dW[N,K] = dy^T[N,M] @ x[M,K]      = gemm_nt(dy^T, x^T)
-    db      = dy.sum over rows
+      dx[M,K] = dy[M,N] @ W[N,K]        = gemm_nt(dy, W^T)
+      dW[N,K] = dy^T[N,M] @ x[M,K]      = gemm_nt(dy^T, x^T)
+      db      = dy.sum over rows
 
-The reduction dim of every gemm_nt call must be a multiple of 64; the
-wrapper zero-pads the token dim when needed (zero rows contribute
-nothing to dW / db).
+  The reduction dim of every gemm_nt call must be a multiple of 64;
+  the wrapper zero-pads the token dim when needed (zero rows
+  contribute nothing to dW / db).
 """
 from __future__ import annotations
 
