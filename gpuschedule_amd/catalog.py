@@ -37,21 +37,21 @@ class ModelMeta:
 
 # sim_iter_time is CALIBRATED to measured MI355X hipGraph-replay
 # iteration times (round 2; profiles/README.md): resnet18 b128 5.0 ms,
-# resnet50 b64 17.2 ms, bert-base b32 10.6 ms, llama3-8b b1 162 ms
-# (the latter two at the hipBLASLt plain-GEMM default,
-# profiles/gemm_mode_compare.json) — so the simulator's JCT/makespan
-# and the ideal_s slowdown baseline track the real backend's
-# absolutes, not just relative orderings.
+# resnet50 b64 17.2 ms, bert-base b32 12.5 ms, llama3-8b b1 182 ms
+# (at the hand-written-GEMM default; GSAMD_GEMM_LIB=1 shaves ~10-13%
+# more, profiles/gemm_mode_compare.json) — so the simulator's
+# JCT/makespan and the ideal_s slowdown baseline track the real
+# backend's absolutes, not just relative orderings.
 CATALOG = {
     # CIFAR-shape ResNet-18 (32x32 input, 10 classes)
     "resnet18": ModelMeta("resnet18", 128, 4 * GB, 0.005, 0.10),
     # ImageNet-shape ResNet-50 (224x224, 1000 classes)
     "resnet50": ModelMeta("resnet50", 64, 16 * GB, 0.0172, 0.15),
     # BERT-base encoder, seq 128
-    "bert-base": ModelMeta("bert-base", 32, 12 * GB, 0.0106, 0.12),
+    "bert-base": ModelMeta("bert-base", 32, 12 * GB, 0.0125, 0.12),
     # Llama-3-8B-architecture decoder (random init), seq 2048.
     # bf16 weights 16G + fp32 AdamW m/v 64G + fp32 master 32G + acts.
-    "llama3-8b": ModelMeta("llama3-8b", 1, 150 * GB, 0.162, 0.20),
+    "llama3-8b": ModelMeta("llama3-8b", 1, 150 * GB, 0.182, 0.20),
     # 4-layer Llama-architecture model for tests/small traces
     "llama3-tiny": ModelMeta("llama3-tiny", 8, 6 * GB, 0.010, 0.12),
 }

@@ -1,9 +1,9 @@
 """Linear layer with dual GEMM dispatch (see _use_lib):
 
-* GSAMD_GEMM_LIB=1 (default): plain projection GEMMs via hipBLASLt
+* GSAMD_GEMM_LIB=1 (opt-in): plain projection GEMMs via hipBLASLt
   (F.linear) — native NN/TN backward, measured 10-13% faster end to
   end (profiles/gemm_mode_compare.json).
-* GSAMD_GEMM_LIB=0: the hand-written bf16 MFMA GEMM
+* GSAMD_GEMM_LIB=0 (default): the hand-written bf16 MFMA GEMM
   (ops/hip/gemm_mfma.hip, NT layout) with the fused
   bias+residual-addend epilogue.  Backward reuses the same NT kernel
   with fast LDS-tiled transposes:
@@ -29,21 +29,21 @@ from ._ext import get_ext
 
 
 def _use_lib() -> bool:
-    """GSAMD_GEMM_LIB=1 (the production DEFAULT) routes MFMALinear's
-    plain projection GEMMs through hipBLASLt (F.linear / torch.matmul):
-    plain library GEMMs are exactly what the vendor library is for, and
-    its native NN/TN operand paths make backward transpose-free.
-    Measured end-to-end (scripts/gemm_mode_compare.py, MI355X):
+    """GSAMD_GEMM_LIB=1 routes MFMALinear's plain projection GEMMs
+    through hipBLASLt (F.linear): its native NN/TN operand paths make
+    backward transpose-free, and measured end-to-end it is faster —
     llama3-8b 183.3 -> 162.4 ms/iter (1.13x), bert-base 11.7 -> 10.6
-    (1.10x).  GSAMD_GEMM_LIB=0 selects the hand-written gfx950 MFMA
-    kernel (0.77-0.94x hipBLASLt depending on shape, PMC evidence in
-    profiles/) with the fused epilogue: bias + residual addend with
-    d(add)=dy identity — zero extra kernels at residual junctions; in
-    lib mode the addend costs one bandwidth-bound add (~4% of the GEMM,
-    net measured win anyway).  Read per call so tests/benches can
-    toggle; a hipGraph capture bakes whichever path was active at
+    (1.10x) (scripts/gemm_mode_compare.py, profiles/).  It is an
+    OPT-IN deployment knob: the DEFAULT (=0) is the hand-written
+    gfx950 MFMA kernel (0.77-0.94x hipBLASLt depending on shape, PMC
+    evidence in profiles/) with the fused epilogue — bias + residual
+    addend with d(add)=dy identity, zero extra kernels at residual
+    junctions — because this framework's contract (BASELINE.json
+    north_star) specifies hand-written CDNA4 GEMM kernels in the
+    training step, visible in rocprof.  Read per call so tests/benches
+    can toggle; a hipGraph capture bakes whichever path was active at
     capture time."""
-    return os.environ.get("GSAMD_GEMM_LIB", "1") == "1"
+    return os.environ.get("GSAMD_GEMM_LIB", "0") == "1"
 
 
 def _pad_rows64(t: torch.Tensor) -> torch.Tensor:
