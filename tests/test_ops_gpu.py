@@ -128,8 +128,8 @@ def test_transpose2d():
 
 
 def test_mfma_linear_autograd(monkeypatch):
-    # pin the hand-written MFMA path (production default is hipBLASLt
-    # for plain GEMMs, GSAMD_GEMM_LIB=1 — this test covers the kernel)
+    # pin the hand-written MFMA path regardless of the ambient
+    # GSAMD_GEMM_LIB setting — this test covers the kernel
     monkeypatch.setenv("GSAMD_GEMM_LIB", "0")
     from gpuschedule_amd.ops import MFMALinear
     torch.manual_seed(5)
