@@ -394,6 +394,13 @@ class Scheduler:
                     now: float, iter_time: Optional[float] = None) -> bool:
         """Record progress; returns True when the job completed."""
         job = self.jobs[job_id]
+        if not job.is_active:
+            # partial gang failure: a member's "fail" report can land
+            # before the leader's "ok" report for the same tick — the
+            # job is already FAILED (resources released); accepting
+            # progress now would double-release and emit complete
+            # after fail.
+            return False
         job.done_iters += iters
         job.attained_service += service
         job.tokens_done += iters * catalog.job_tokens_per_iter(

@@ -140,3 +140,28 @@ def test_chrome_trace_evict_marker():
     tr = log.to_chrome_trace()
     names = [e.get("name") for e in tr["traceEvents"]]
     assert "evict a" in names
+
+
+def test_partial_gang_failure_report_order():
+    """Partial gang failure: one member's fail report can be processed
+    BEFORE the leader's ok report for the same tick.  The job must end
+    FAILED (no resurrection to DONE, no complete-after-fail event, no
+    double release)."""
+    from gpuschedule_amd import Cluster, JobSpec, Scheduler, make_policy
+    sched = Scheduler(Cluster(2, 2), make_policy("fifo"))
+    sched.submit([JobSpec(job_id="g", submit_time=0.0, n_gpus=2,
+                          model="llama3-tiny", total_iters=4)])
+    sched.tick(0.0)
+    assert sched.jobs["g"].state.name == "RUNNING"
+    sched.on_fail("g", "RuntimeError: boom", 1.0)
+    # leader's ok report arrives after — enough iters to "complete"
+    fin = sched.on_progress("g", 4, 2.0, 1.0)
+    assert fin is False
+    assert sched.jobs["g"].state.name == "FAILED"
+    kinds = [e["event"] for e in sched.log.events if e.get("job") == "g"]
+    assert "complete" not in kinds and kinds.count("fail") == 1
+    # cluster fully freed exactly once: both GPUs placeable again
+    sched.submit([JobSpec(job_id="h", submit_time=1.0, n_gpus=2,
+                          model="llama3-tiny", total_iters=1)])
+    sched.tick(2.0)
+    assert sched.jobs["h"].state.name == "RUNNING"
