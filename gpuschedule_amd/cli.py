@@ -127,6 +127,14 @@ def cmd_compare(args):
               " ".join(f"{s[c]:14.2f}" for c in cols))
 
 
+def cmd_summarize(args):
+    """Print the JCT/makespan/slowdown summary of a saved event log."""
+    from .sched.events import EventLog
+    log = EventLog.load(args.log)
+    print(json.dumps({k: round(v, 3)
+                      for k, v in log.summary().items()}, indent=1))
+
+
 def cmd_export_trace(args):
     from .sched.events import EventLog
     log = EventLog.load(args.log)
@@ -173,6 +181,11 @@ def main(argv=None):
                             "(simulator) and print a metric table")
     p.add_argument("--policies", default="fifo,srtf,las,gandiva,elastic")
     p.set_defaults(fn=cmd_compare)
+    p = sub.add_parser(
+        "summarize",
+        help="print JCT/makespan/slowdown metrics of a saved event log")
+    p.add_argument("--log", required=True)
+    p.set_defaults(fn=cmd_summarize)
     p = sub.add_parser(
         "export-trace",
         help="convert an event-log JSONL to a Chrome/Perfetto trace")

@@ -111,3 +111,12 @@ def test_simulate_warm_pool_flag():
                               "--warm-pool").strip())
     assert warm["avg_jct"] < cold["avg_jct"]
     assert warm["n_complete"] == cold["n_complete"]
+
+
+def test_summarize_command(tmp_path):
+    log = str(tmp_path / "ev.jsonl")
+    run_cli("simulate", "--policy", "srtf", "--jobs", "5",
+            "--gpus", "2", "--log", log)
+    s = json.loads(run_cli("summarize", "--log", log))
+    assert s["n_complete"] > 0
+    assert s["makespan"] > 0 and s["avg_jct"] > 0
