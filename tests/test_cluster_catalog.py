@@ -83,3 +83,34 @@ def test_summary_percentiles():
     assert s["p50_jct"] == 3.0
     assert s["p95_jct"] == 100.0
     assert s["makespan"] == 100.0
+
+
+def test_chrome_trace_closes_open_runs_at_eof():
+    # a job still running when the log ends must get a slice closed at
+    # the last event time (not be dropped from the timeline)
+    from gpuschedule_amd import EventLog
+    log = EventLog()
+    log.emit(0.0, "submit", "a", n_gpus=1, model="resnet18", iters=10)
+    log.emit(1.0, "start", "a", gpus=[0], first=True)
+    log.emit(5.0, "tick", util=1.0, running=1, queued=0)
+    tr = log.to_chrome_trace()
+    slices = [e for e in tr["traceEvents"] if e.get("ph") == "X"]
+    assert len(slices) == 1
+    s = slices[0]
+    assert s["name"] == "a" and s["tid"] == 0
+    assert s["ts"] == 1.0e6 and s["dur"] == 4.0e6
+
+
+def test_job_tokens_per_iter_width_and_families():
+    from gpuschedule_amd.catalog import job_tokens_per_iter
+    from gpuschedule_amd.trace.job import JobSpec
+
+    def J(model, batch=0):
+        return JobSpec(job_id="x", submit_time=0.0, n_gpus=1,
+                       model=model, total_iters=1, batch_size=batch)
+    # image models count samples; sequence models count tokens
+    assert job_tokens_per_iter(J("resnet50", 64)) == 64
+    assert job_tokens_per_iter(J("bert-base", 32)) == 32 * 128
+    assert job_tokens_per_iter(J("llama3-8b", 1)) == 2048
+    # DP width multiplies whole-gang throughput
+    assert job_tokens_per_iter(J("bert-base", 32), width=4) == 4 * 32 * 128
