@@ -47,7 +47,9 @@ def _specs_from_args(args, cfg):
                                   total_iters=t.get("iters", 400))
     return make_trace(t.get("jobs", args.jobs), mix=t.get("mix", args.mix),
                       seed=t.get("seed", 0),
-                      mean_interarrival=t.get("mean_interarrival", 10.0),
+                      mean_interarrival=t.get(
+                          "mean_interarrival",
+                          getattr(args, "interarrival", None) or 10.0),
                       iters_range=tuple(t.get("iters_range", (50, 400))),
                       gang_sizes=t.get("gang_sizes"))
 
@@ -155,6 +157,11 @@ def main(argv=None):
     common.add_argument("--gpus", type=int, default=8)
     common.add_argument("--jobs", type=int, default=16)
     common.add_argument("--mix", default="mixed-cv-nlp")
+    common.add_argument("--interarrival", type=float, default=None,
+                        help="mean synthetic-trace interarrival seconds "
+                             "(default 10; lower it to create "
+                             "contention — iteration times are "
+                             "hardware-calibrated and fast)")
     common.add_argument("--log", default=None, help="event log JSONL out")
 
     p = sub.add_parser("simulate", parents=[common])
