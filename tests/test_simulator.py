@@ -151,3 +151,63 @@ def test_warm_pool_skips_cold_start():
     assert abs(cold["avg_jct"] - 10.5) < 0.3
     assert abs(warm["avg_jct"] - 10.05) < 0.3
     assert warm["avg_jct"] < cold["avg_jct"] - 0.3
+
+
+def test_placement_overhead_model():
+    """Unit-test the documented per-placement overhead model
+    (sim/simulator.py docstring): resume-in-HBM is cheapest, overlap
+    resizes ride xGMI broadcast (150 GB/s), migrations pay D2H+H2D
+    (48 GB/s each way), evicted restores pay H2D only."""
+    from gpuschedule_amd import Cluster, JobSpec, Scheduler
+    from gpuschedule_amd.sched.scheduler import Placement
+    from gpuschedule_amd.sim.simulator import (
+        COLD_START_S, D2H_BYTES_PER_S, H2D_BYTES_PER_S, RESUME_HBM_S,
+        XGMI_BYTES_PER_S, Simulator)
+
+    sched = Scheduler(Cluster(4, 4), make_policy("fifo"))
+    spec = JobSpec(job_id="a", submit_time=0.0, n_gpus=1,
+                   model="resnet50", total_iters=10)
+    sched.submit([spec])
+    sched.tick(0.0)
+    sim = Simulator(sched)
+    job = sched.jobs["a"]
+    from gpuschedule_amd import catalog
+    ckpt = catalog.job_mem_bytes(spec) / 3
+
+    def ov(**kw):
+        return sim._placement_overhead(Placement(job_id="a", gpus=[0],
+                                                 **kw), job)
+
+    assert ov() == COLD_START_S
+    assert ov(resume=True) == RESUME_HBM_S
+    assert abs(ov(resume=True, resize_from=2, resize_from_gpus=[0, 1])
+               - ckpt / XGMI_BYTES_PER_S) < 1e-9      # overlap: xGMI bcast
+    assert abs(ov(resume=True, migrate_from=[1])
+               - (ckpt / D2H_BYTES_PER_S + ckpt / H2D_BYTES_PER_S)) < 1e-9
+    assert abs(ov(resume=True, from_evicted=True)
+               - ckpt / H2D_BYTES_PER_S) < 1e-9
+    # disjoint resize (no surviving rank) = checkpoint round-trip
+    assert abs(ov(resume=True, resize_from=1, resize_from_gpus=[3])
+               - (ckpt / D2H_BYTES_PER_S + ckpt / H2D_BYTES_PER_S)) < 1e-9
+    # warm pool: fresh placements cost RESUME_HBM_S instead
+    assert Simulator(sched, warm_pool=True)._placement_overhead(
+        Placement(job_id="a", gpus=[0]), job) == RESUME_HBM_S
+
+
+def test_summary_counts_failures_and_tokens():
+    from gpuschedule_amd import EventLog
+    log = EventLog()
+    log.emit(0.0, "submit", "a", n_gpus=1, model="bert-base", iters=4,
+             ideal_s=2.0)
+    log.emit(0.0, "submit", "b", n_gpus=1, model="bert-base", iters=4,
+             ideal_s=2.0)
+    log.emit(0.1, "start", "a", gpus=[0], first=True)
+    log.emit(0.1, "start", "b", gpus=[1], first=True)
+    log.emit(1.0, "fail", "b", error="boom")
+    log.emit(4.0, "complete", "a", jct=4.0, iters=4, tokens=16384,
+             tokens_per_s=4096.0)
+    s = log.summary()
+    assert s["n_jobs"] == 2 and s["n_complete"] == 1 and s["n_failed"] == 1
+    assert s["tokens_per_s_sum"] == 4096.0
+    assert abs(s["avg_slowdown"] - 2.0) < 1e-9   # jct 4 / ideal 2
+    assert s["makespan"] == 4.0
