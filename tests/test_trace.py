@@ -108,3 +108,20 @@ def test_philly_log_schema_edge_cases(tmp_path):
     assert b.fail_at_iter > 0 and b.total_iters == b.fail_at_iter + 1
     # b ran 60 min total across two attempts
     assert b.submit_time == 600.0
+
+
+def test_philly_log_time_scale_proportional():
+    """time_scale compresses the trace uniformly: submit offsets scale
+    linearly and derived iteration counts scale with the scaled
+    duration (duration = iters x catalog iter_time is invariant)."""
+    from gpuschedule_amd.trace.philly_log import load_philly_log
+    a = load_philly_log("data/philly_sample.json", time_scale=1 / 3600,
+                        seed=4)
+    b = load_philly_log("data/philly_sample.json", time_scale=2 / 3600,
+                        seed=4)
+    assert [s.job_id for s in a] == [s.job_id for s in b]
+    for sa, sb in zip(a, b):
+        assert abs(sb.submit_time - 2 * sa.submit_time) < 1e-6
+        # iters halve... no: doubling time_scale doubles scaled duration
+        # -> doubles iters (within rounding of max(1, int(...)))
+        assert abs(sb.total_iters - 2 * sa.total_iters) <= 2
