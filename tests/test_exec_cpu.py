@@ -475,3 +475,45 @@ def test_gang_prewarm_and_adopt(tmp_path):
     with open(out) as f:
         s = json.load(f)
     assert s["n_complete"] == 1
+
+
+def test_runner_pool_budget_lru_eviction():
+    """RunnerPool unit semantics: byte-budgeted, LRU-evicting, get()
+    returns the most recently put runner for a key and maintains the
+    byte ledger; flush releases everything."""
+    from gpuschedule_amd.exec.trainer import RunnerPool
+
+    class FakeRunner:
+        def __init__(self, key, mem):
+            self.pool_key = key
+            self.measured_mem = mem
+            self.model = object()
+            self.poolable = True
+            self.released = False
+
+        def release(self):
+            self.released = True
+
+    MB = 1 << 20
+    pool = RunnerPool(max_bytes=5 * MB)
+    a = FakeRunner(("m", 1, (0,)), 2 * MB)
+    b = FakeRunner(("m", 1, (0,)), 2 * MB)
+    c = FakeRunner(("n", 2, (0,)), 2 * MB)
+    assert pool.put(a) and pool.put(b)
+    # c does not fit next to a+b: the LRU entry (a) is evicted+released
+    assert pool.put(c)
+    assert a.released and not b.released and not c.released
+    assert pool._bytes == 4 * MB
+    # get() pops by key; most-recent-put first
+    got = pool.get(("m", 1, (0,)))
+    assert got is b and pool._bytes == 2 * MB
+    assert pool.get(("m", 1, (0,))) is None   # a was evicted
+    # oversized runner is refused outright
+    huge = FakeRunner(("h", 1, (0,)), 6 * MB)
+    assert not pool.put(huge) and not huge.released
+    # non-poolable runner is refused
+    bad = FakeRunner(("m", 1, (0,)), MB)
+    bad.poolable = False
+    assert not pool.put(bad)
+    pool.flush()
+    assert c.released and pool._bytes == 0 and pool.get(("n", 2, (0,))) is None
