@@ -517,3 +517,33 @@ def test_runner_pool_budget_lru_eviction():
     assert not pool.put(bad)
     pool.flush()
     assert c.released and pool._bytes == 0 and pool.get(("n", 2, (0,))) is None
+
+
+def test_build_runner_oom_flushes_pool_and_retries(monkeypatch, tmp_path):
+    """Pooled runners hold HBM the scheduler does not account for; an
+    OOM during runner construction must flush the pool and retry once
+    (exec/backend.py:_build_runner)."""
+    import gpuschedule_amd.exec.backend as BK
+    from gpuschedule_amd.exec.trainer import RUNNER_POOL
+
+    calls = {"n": 0, "flushed_at": None}
+    real_flush = RUNNER_POOL.flush
+
+    def spy_flush():
+        calls["flushed_at"] = calls["n"]
+        real_flush()
+
+    class FlakyRunner:
+        def __init__(self, *a, **kw):
+            calls["n"] += 1
+            if calls["n"] == 1:
+                raise torch.cuda.OutOfMemoryError("fake OOM")
+            self.ok = True
+
+    monkeypatch.setattr(BK, "JobRunner", FlakyRunner)
+    monkeypatch.setattr(RUNNER_POOL, "flush", spy_flush)
+    be = make_backend([J("x", iters=1)])
+    r = be._build_runner("x", "llama3-tiny", 2, None, [0], 0, None,
+                         ("llama3-tiny", 2, (0,)))
+    assert r.ok and calls["n"] == 2
+    assert calls["flushed_at"] == 1  # flushed between the two attempts
