@@ -547,3 +547,15 @@ def test_build_runner_oom_flushes_pool_and_retries(monkeypatch, tmp_path):
                          ("llama3-tiny", 2, (0,)))
     assert r.ok and calls["n"] == 2
     assert calls["flushed_at"] == 1  # flushed between the two attempts
+
+
+def test_gang_groups_contiguity_contract():
+    """Width-1 needs no group; non-contiguous gangs are a scheduler
+    bug and must fail loudly (all pre-created groups are contiguous
+    rank ranges)."""
+    from gpuschedule_amd.exec.gangs import GangGroups
+    gg = GangGroups(4)  # dist not initialized -> no groups created
+    assert gg.get([2]) is None
+    assert gg.get([]) is None
+    with pytest.raises(AssertionError):
+        gg.get([0, 2])
